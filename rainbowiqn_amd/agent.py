@@ -38,6 +38,13 @@ class Agent:
         self.K = cfg.num_quantile_samples
         self.kappa = cfg.kappa
 
+        # bf16 compute via autocast on GPU (fp32 master weights + Adam state)
+        self.autocast_dtype = (
+            torch.bfloat16
+            if (cfg.dtype == "bf16" and self.device.type == "cuda")
+            else None
+        )
+
         self.online_net = RainbowIQN(cfg, action_space).to(self.device)
         self.target_net = RainbowIQN(cfg, action_space).to(self.device)
         if cfg.model:
@@ -60,8 +67,15 @@ class Agent:
             return int(self.online_net.q_values(x, self.K).argmax(dim=1).item())
 
     def act_batch(self, states: torch.Tensor) -> torch.Tensor:
-        with torch.no_grad():
+        with torch.no_grad(), self._autocast():
             return self.online_net.q_values(states.to(self.device), self.K).argmax(dim=1)
+
+    def _autocast(self):
+        if self.autocast_dtype is not None:
+            return torch.autocast("cuda", dtype=self.autocast_dtype)
+        import contextlib
+
+        return contextlib.nullcontext()
 
     def act_e_greedy(self, state: torch.Tensor, epsilon: float = 0.001) -> int:
         """Evaluation-time safety epsilon (reference eval behavior)."""
@@ -88,23 +102,25 @@ class Agent:
     def learn_from_batch(self, states, actions, returns, next_states, nonterminals, weights):
         b = states.shape[0]
         net = self.online_net
-        # online quantiles for the taken actions: theta_i  [B, N]
-        z, taus = net(states, self.N)  # [B, N, A], [B*N, 1]
-        theta = z.gather(2, actions.view(b, 1, 1).expand(b, self.N, 1)).squeeze(2)
-        taus = taus.view(b, self.N)
+        with self._autocast():
+            # online quantiles for the taken actions: theta_i  [B, N]
+            z, taus = net(states, self.N)  # [B, N, A], [B*N, 1]
+            theta = z.gather(2, actions.view(b, 1, 1).expand(b, self.N, 1)).squeeze(2)
+            taus = taus.view(b, self.N)
 
-        with torch.no_grad():
-            # double-Q action selection on s' via the ONLINE net (K samples)
-            a_star = net.q_values(next_states, self.K).argmax(dim=1)  # [B]
-            # target quantiles Z'_tau'(s', a*)  [B, N']
-            zp, _ = self.target_net(next_states, self.Np)
-            theta_p = zp.gather(2, a_star.view(b, 1, 1).expand(b, self.Np, 1)).squeeze(2)
-            t_theta = torch_ref.n_step_target(
-                theta_p, returns, nonterminals, self.discount, self.n
-            )
+            with torch.no_grad():
+                # double-Q action selection on s' via the ONLINE net (K samples)
+                a_star = net.q_values(next_states, self.K).argmax(dim=1)  # [B]
+                # target quantiles Z'_tau'(s', a*)  [B, N']
+                zp, _ = self.target_net(next_states, self.Np)
+                theta_p = zp.gather(2, a_star.view(b, 1, 1).expand(b, self.Np, 1)).squeeze(2)
+                t_theta = torch_ref.n_step_target(
+                    theta_p, returns, nonterminals, self.discount, self.n
+                )
 
+        # loss in fp32 for numerical fidelity of priorities/gradient scale
         loss, per_sample = torch_ref.quantile_huber_loss(
-            theta, t_theta, taus, weights, self.kappa
+            theta.float(), t_theta.float(), taus.float(), weights, self.kappa
         )
         self.optimiser.zero_grad(set_to_none=True)
         loss.backward()

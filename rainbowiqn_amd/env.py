@@ -139,6 +139,64 @@ class SyntheticAtariEnv:
         return state, reward, done
 
 
+class VectorSyntheticEnv:
+    """Batched synthetic env: E independent episode streams, numpy-vectorized.
+
+    Used by the Ape-X actor loop to amortize the per-step act() inference over
+    several env streams per actor GPU (capability superset of the reference's
+    one-env-per-actor-process; set n_envs=1 for reference behavior).
+    Frames are produced directly as uint8 [E,84,84]; the caller preprocesses
+    on-GPU (ops kernel K15).
+    """
+
+    def __init__(self, cfg, n_envs: int, seed_offset: int = 0):
+        self.cfg = cfg
+        self.n_envs = n_envs
+        self.n_actions = n_actions_for(cfg.game)
+        self.history = cfg.history_length
+        self.max_episode_length = cfg.max_episode_length
+        self.rng = np.random.default_rng(cfg.seed * 100003 + seed_offset)
+        self._t = np.zeros(n_envs, dtype=np.int64)
+        self._phase = self.rng.integers(0, 1 << 16, size=n_envs)
+        self._y = np.arange(FRAME_H, dtype=np.int64)[:, None]
+        self._x = np.arange(FRAME_W, dtype=np.int64)[None, :]
+        self.p_done = 4.0 / 512.0
+
+    def action_space(self) -> int:
+        return self.n_actions
+
+    def frames(self) -> np.ndarray:
+        """Current raw frames [E,84,84] uint8."""
+        t = ((self._t + self._phase) & 0xFFFF)[:, None, None]
+        img = ((self._x * 3 + self._y * 2 + t * 5) & 0xFF).astype(np.uint8)
+        cy = (FRAME_H / 2 + (FRAME_H / 2 - 6) * np.sin(t[:, 0, 0] * 0.11)).astype(np.int64)
+        cx = (FRAME_W / 2 + (FRAME_W / 2 - 6) * np.cos(t[:, 0, 0] * 0.07)).astype(np.int64)
+        for e in range(self.n_envs):  # sprite stamp (tiny loop, 8x8 region)
+            img[e, max(0, cy[e] - 4) : cy[e] + 4, max(0, cx[e] - 4) : cx[e] + 4] = 255
+        noise = self.rng.integers(0, 32, size=img.shape, dtype=np.uint8)
+        return img | noise
+
+    def reset_all(self) -> np.ndarray:
+        self._t[:] = 0
+        self._phase = self.rng.integers(0, 1 << 16, size=self.n_envs)
+        return self.frames()
+
+    def step(self, actions: np.ndarray):
+        """actions [E] -> (frames u8 [E,84,84], rewards f32 [E], dones bool [E])."""
+        self._t += 4
+        u = self.rng.random(self.n_envs)
+        rewards = np.where(
+            u < 0.02, np.where((actions + self._t) % 3 != 0, 1.0, -1.0), 0.0
+        ).astype(np.float32)
+        dones = self.rng.random(self.n_envs) < self.p_done
+        dones |= self._t >= self.max_episode_length
+        frames = self.frames()
+        if dones.any():
+            self._t[dones] = 0
+            self._phase[dones] = self.rng.integers(0, 1 << 16, size=int(dones.sum()))
+        return frames, rewards, dones
+
+
 def make_env(cfg, device: str | torch.device = "cpu"):
     """Return the real ALE wrapper when available, else the synthetic env."""
     try:  # pragma: no cover - no ALE wheel on this box
