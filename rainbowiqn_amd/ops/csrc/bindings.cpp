@@ -1,0 +1,144 @@
+// Python bindings for the gfx950 HIP kernels.  Torch headers stay in this
+// translation unit; device code lives in the *.hip files behind a C ABI
+// (riqn_kernels.h).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "riqn_kernels.h"
+
+namespace {
+
+void* cur_stream() { return (void*)at::cuda::getCurrentCUDAStream().stream(); }
+
+#define CHECK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+#define CHECK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+#define CHECK_T(t, ty) \
+    CHECK_DEV(t); CHECK_CONT(t); TORCH_CHECK((t).scalar_type() == ty, #t " dtype")
+
+void replay_append(torch::Tensor frames, torch::Tensor actions,
+                   torch::Tensor rewards, torch::Tensor terminals,
+                   torch::Tensor timesteps, torch::Tensor tree,
+                   int64_t tree_size, int64_t capacity,
+                   torch::Tensor ring_state, int64_t n_rings, int64_t ring_cap,
+                   torch::Tensor max_priority, torch::Tensor chunk_frames,
+                   torch::Tensor chunk_meta, int64_t ring_id) {
+    CHECK_T(frames, torch::kUInt8);
+    CHECK_T(tree, torch::kFloat32);
+    CHECK_T(chunk_frames, torch::kUInt8);
+    CHECK_T(chunk_meta, torch::kFloat32);
+    CHECK_T(ring_state, torch::kInt32);
+    int64_t c = chunk_frames.size(0);
+    TORCH_CHECK(chunk_meta.size(0) == c && chunk_meta.size(1) == 3);
+    TORCH_CHECK(c < ring_cap, "chunk longer than ring");
+    riqn_replay_append(
+        frames.data_ptr<uint8_t>(), actions.data_ptr<int32_t>(),
+        rewards.data_ptr<float>(), terminals.data_ptr<uint8_t>(),
+        timesteps.data_ptr<int32_t>(), tree.data_ptr<float>(), tree_size,
+        capacity, ring_state.data_ptr<int32_t>(), (int)n_rings, ring_cap,
+        max_priority.data_ptr<float>(), chunk_frames.data_ptr<uint8_t>(),
+        chunk_meta.data_ptr<float>(), c, (int)ring_id, cur_stream());
+}
+
+void replay_sample(torch::Tensor frames, torch::Tensor actions,
+                   torch::Tensor rewards, torch::Tensor terminals,
+                   torch::Tensor timesteps, torch::Tensor tree,
+                   int64_t tree_size, int64_t capacity,
+                   torch::Tensor ring_state, int64_t n_rings, int64_t ring_cap,
+                   int64_t history, int64_t n_step, double discount,
+                   torch::Tensor beta, torch::Tensor rng_state, int64_t batch,
+                   torch::Tensor out_idx, torch::Tensor out_states,
+                   torch::Tensor out_next_states, torch::Tensor out_actions,
+                   torch::Tensor out_returns, torch::Tensor out_nonterminals,
+                   torch::Tensor out_weights) {
+    CHECK_T(out_states, torch::kFloat32);
+    CHECK_T(out_actions, torch::kInt64);
+    CHECK_T(out_idx, torch::kInt32);
+    TORCH_CHECK(batch <= 512, "sample batch must be <= 512 (single block)");
+    riqn_replay_sample(
+        frames.data_ptr<uint8_t>(), actions.data_ptr<int32_t>(),
+        rewards.data_ptr<float>(), terminals.data_ptr<uint8_t>(),
+        timesteps.data_ptr<int32_t>(), tree.data_ptr<float>(), tree_size,
+        capacity, ring_state.data_ptr<int32_t>(), (int)n_rings, ring_cap,
+        (int)history, (int)n_step, (float)discount, beta.data_ptr<float>(),
+        (uint64_t*)rng_state.data_ptr<int64_t>(), (int)batch,
+        out_idx.data_ptr<int32_t>(), out_states.data_ptr<float>(),
+        out_next_states.data_ptr<float>(), out_actions.data_ptr<int64_t>(),
+        out_returns.data_ptr<float>(), out_nonterminals.data_ptr<float>(),
+        out_weights.data_ptr<float>(), cur_stream());
+}
+
+void replay_update_priorities(torch::Tensor tree, int64_t tree_size,
+                              torch::Tensor max_priority, torch::Tensor idxs,
+                              torch::Tensor priorities,
+                              double priority_exponent) {
+    CHECK_T(tree, torch::kFloat32);
+    CHECK_T(idxs, torch::kInt32);
+    CHECK_T(priorities, torch::kFloat32);
+    riqn_replay_update_priorities(
+        tree.data_ptr<float>(), tree_size, max_priority.data_ptr<float>(),
+        idxs.data_ptr<int32_t>(), priorities.data_ptr<float>(),
+        idxs.numel(), (float)priority_exponent, cur_stream());
+}
+
+void tree_rebuild(torch::Tensor tree, int64_t tree_size) {
+    CHECK_T(tree, torch::kFloat32);
+    for (int64_t nodes = tree_size / 2; nodes >= 1; nodes /= 2)
+        riqn_tree_level_pass(tree.data_ptr<float>(), nodes, cur_stream());
+}
+
+void quantile_huber(torch::Tensor theta, torch::Tensor target_q,
+                    torch::Tensor taus, torch::Tensor returns,
+                    torch::Tensor nonterm,
+                    c10::optional<torch::Tensor> weights, double gamma_n,
+                    double kappa, torch::Tensor out_loss,
+                    torch::Tensor out_per_sample,
+                    torch::Tensor out_grad_theta) {
+    CHECK_T(theta, torch::kFloat32);
+    CHECK_T(target_q, torch::kFloat32);
+    CHECK_T(taus, torch::kFloat32);
+    int B = theta.size(0), N = theta.size(1), Np = target_q.size(1);
+    TORCH_CHECK(N <= 64 && Np <= 64, "N/N' must be <= 64");
+    const float* w = nullptr;
+    if (weights.has_value()) {
+        CHECK_T(weights.value(), torch::kFloat32);
+        w = weights.value().data_ptr<float>();
+    }
+    riqn_quantile_huber(theta.data_ptr<float>(), target_q.data_ptr<float>(),
+                        taus.data_ptr<float>(), returns.data_ptr<float>(),
+                        nonterm.data_ptr<float>(), w, (float)gamma_n,
+                        (float)kappa, B, N, Np, out_loss.data_ptr<float>(),
+                        out_per_sample.data_ptr<float>(),
+                        out_grad_theta.data_ptr<float>(), cur_stream());
+}
+
+void adam_step(torch::Tensor param, torch::Tensor grad, torch::Tensor exp_avg,
+               torch::Tensor exp_avg_sq, double lr, double beta1, double beta2,
+               double eps, torch::Tensor step) {
+    CHECK_T(param, torch::kFloat32);
+    CHECK_T(grad, torch::kFloat32);
+    CHECK_T(step, torch::kInt32);
+    riqn_adam_step(param.data_ptr<float>(), grad.data_ptr<float>(),
+                   exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(),
+                   param.numel(), (float)lr, (float)beta1, (float)beta2,
+                   (float)eps, step.data_ptr<int32_t>(), cur_stream());
+}
+
+void preprocess(torch::Tensor in, torch::Tensor out) {
+    CHECK_T(in, torch::kUInt8);
+    CHECK_T(out, torch::kFloat32);
+    TORCH_CHECK(in.numel() == out.numel());
+    riqn_preprocess(in.data_ptr<uint8_t>(), out.data_ptr<float>(), in.numel(),
+                    cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("replay_append", &replay_append);
+    m.def("replay_sample", &replay_sample);
+    m.def("replay_update_priorities", &replay_update_priorities);
+    m.def("tree_rebuild", &tree_rebuild);
+    m.def("quantile_huber", &quantile_huber);
+    m.def("adam_step", &adam_step);
+    m.def("preprocess", &preprocess);
+}

@@ -1,0 +1,350 @@
+#include "hip/hip_runtime.h"
+// On-GPU prioritized sum-tree replay for MI355X (SURVEY.md kernel K16).
+//
+// The reference keeps its replay as a CPU numpy sum-tree fed through Redis
+// (rainbowiqn/memory.py); here the whole replay -- uint8
+// frame ring, priority sum-tree, stochastic prefix-sum sampling, n-step
+// return assembly, frame-stack reconstruction and importance weights -- lives
+// in HBM3E and runs as three kernel families so the learner step can be
+// hipGraph-captured end to end:
+//
+//   append  : meta scan (1 thread) -> frame blit (C blocks) -> leaf updates
+//   sample  : S1 stratified tree descent + IS weights (1 block)
+//             S2 gather/assemble + fused u8->f32 preprocess (B blocks)
+//   update  : priority delta propagation with atomicExch/atomicAdd
+//
+// Ring layout: capacity = n_rings * ring_cap; ring r (one per Ape-X actor)
+// owns slots [r*ring_cap, (r+1)*ring_cap) so per-actor episode streams stay
+// contiguous for frame-stack/n-step reconstruction.  ring_state[r] =
+// {head, count, episode_t, pad} (int32).  Tree: float32 heap, nodes[1] root,
+// leaves at [tree_size, tree_size + capacity).
+#include "common.h"
+#include "riqn_kernels.h"
+
+#define FRAME_PX 7056  // 84*84
+
+namespace {
+
+DEV void tree_set_leaf(float* tree, int64_t tree_size, int64_t leaf, float p) {
+    float old = atomicExch(&tree[tree_size + leaf], p);
+    float delta = p - old;
+    uint64_t node = (uint64_t)(tree_size + leaf) >> 1;
+    while (node >= 1) {
+        atomicAdd(&tree[node], delta);
+        if (node == 1) break;
+        node >>= 1;
+    }
+}
+
+struct RingView {
+    const int32_t* rs;
+    int64_t ring_cap;
+    DEV int head(int r) const { return rs[4 * r + 0]; }
+    DEV int count(int r) const { return rs[4 * r + 1]; }
+};
+
+DEV bool slot_valid(int64_t leaf, int64_t capacity, int64_t ring_cap,
+                    const int32_t* ring_state, int history, int n_step,
+                    const float* tree, int64_t tree_size) {
+    if (leaf >= capacity) return false;
+    int r = (int)(leaf / ring_cap);
+    int local = (int)(leaf % ring_cap);
+    int head = ring_state[4 * r + 0];
+    int count = ring_state[4 * r + 1];
+    int fwd = (int)((head - local + ring_cap) % ring_cap);
+    if (fwd <= n_step) return false;
+    if (count < ring_cap) {
+        if (local >= count || local + n_step >= count) return false;
+    } else {
+        int bwd = (int)((local - head + ring_cap) % ring_cap);
+        if (bwd < history - 1) return false;
+    }
+    return tree[tree_size + leaf] > 0.0f;
+}
+
+// ---------------------------------------------------------------------------
+// APPEND
+// ---------------------------------------------------------------------------
+__global__ void append_meta_kernel(
+    int32_t* actions, float* rewards, uint8_t* terminals, int32_t* timesteps,
+    int32_t* ring_state, int ring_id, int64_t ring_cap,
+    const float* chunk_meta, int64_t C) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    int head = ring_state[4 * ring_id + 0];
+    int count = ring_state[4 * ring_id + 1];
+    int t = ring_state[4 * ring_id + 2];
+    int64_t base = (int64_t)ring_id * ring_cap;
+    for (int64_t k = 0; k < C; ++k) {
+        int local = (int)((head + k) % ring_cap);
+        int64_t slot = base + local;
+        actions[slot] = (int32_t)chunk_meta[3 * k + 0];
+        rewards[slot] = chunk_meta[3 * k + 1];
+        bool term = chunk_meta[3 * k + 2] > 0.5f;
+        terminals[slot] = term ? 1 : 0;
+        timesteps[slot] = t;
+        t = term ? 0 : t + 1;
+    }
+    ring_state[4 * ring_id + 0] = (int)((head + C) % ring_cap);
+    ring_state[4 * ring_id + 1] = (int)min((int64_t)count + C, ring_cap);
+    ring_state[4 * ring_id + 2] = t;
+}
+
+__global__ void append_frames_kernel(
+    uint8_t* frames, const int32_t* ring_state, int ring_id, int64_t ring_cap,
+    const uint8_t* chunk_frames, int64_t C) {
+    // runs AFTER append_meta: old head = head - C (mod ring_cap)
+    int head_new = ring_state[4 * ring_id + 0];
+    int64_t head_old = (head_new - C % ring_cap + ring_cap) % ring_cap;
+    int64_t k = blockIdx.x;  // one block per chunk entry
+    int local = (int)((head_old + k) % ring_cap);
+    int64_t slot = (int64_t)ring_id * ring_cap + local;
+    // 7056 B per frame = 1764 u32; 256 threads stride
+    const uint32_t* src = (const uint32_t*)(chunk_frames + k * FRAME_PX);
+    uint32_t* dst = (uint32_t*)(frames + slot * FRAME_PX);
+    for (int i = threadIdx.x; i < FRAME_PX / 4; i += blockDim.x) dst[i] = src[i];
+}
+
+__global__ void append_tree_kernel(
+    float* tree, int64_t tree_size, const int32_t* ring_state, int ring_id,
+    int64_t ring_cap, const float* max_priority, int64_t C) {
+    int head_new = ring_state[4 * ring_id + 0];
+    int64_t head_old = (head_new - C % ring_cap + ring_cap) % ring_cap;
+    float p = max_priority[0];
+    for (int64_t k = threadIdx.x + (int64_t)blockIdx.x * blockDim.x; k < C;
+         k += (int64_t)gridDim.x * blockDim.x) {
+        int local = (int)((head_old + k) % ring_cap);
+        int64_t leaf = (int64_t)ring_id * ring_cap + local;
+        tree_set_leaf(tree, tree_size, leaf, p);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// SAMPLE
+// ---------------------------------------------------------------------------
+__global__ void sample_indices_kernel(
+    const float* tree, int64_t tree_size, int64_t capacity,
+    const int32_t* ring_state, int n_rings, int64_t ring_cap, int history,
+    int n_step, const float* beta_ptr, uint64_t* rng_state, int batch,
+    int32_t* out_idx, float* out_weights) {
+    __shared__ float lds[32];
+    __shared__ int total_count_s;
+    int b = threadIdx.x;
+    if (b == 0) {
+        int tc = 0;
+        for (int r = 0; r < n_rings; ++r) tc += ring_state[4 * r + 1];
+        total_count_s = tc;
+    }
+    __syncthreads();
+    float total = tree[1];
+    float beta = beta_ptr[0];
+    int64_t leaf = -1;
+    float prob = 0.0f;
+    if (b < batch) {
+        Philox rng(rng_state[0], rng_state[1], (uint32_t)b);
+        for (int attempt = 0; attempt < 64; ++attempt) {
+            float u;
+            if (attempt < 8)
+                u = ((float)b + rng.uniform()) * (total / batch);
+            else
+                u = rng.uniform() * total;
+            u = fminf(u, total * (1.0f - 1e-6f));
+            // prefix-sum descent
+            int64_t node = 1;
+            float v = u;
+            while (node < tree_size) {
+                float left = tree[2 * node];
+                if (v < left) {
+                    node = 2 * node;
+                } else {
+                    v -= left;
+                    node = 2 * node + 1;
+                }
+            }
+            int64_t cand = node - tree_size;
+            if (slot_valid(cand, capacity, ring_cap, ring_state, history,
+                           n_step, tree, tree_size)) {
+                leaf = cand;
+                break;
+            }
+        }
+        if (leaf < 0) {
+            // last resort: uniform over a definitely-valid region of a ring
+            for (int rr = 0; rr < n_rings && leaf < 0; ++rr) {
+                int r = (b + rr) % n_rings;
+                int head = ring_state[4 * r + 0];
+                int count = ring_state[4 * r + 1];
+                if (count <= history + n_step + 1) continue;
+                int local;
+                if (count < ring_cap) {
+                    int span = max(1, count - n_step - 1);
+                    local = (int)(rng.uniform() * span);
+                } else {
+                    int lo = (int)((head + history - 1) % ring_cap);
+                    int span = max(1, (int)(ring_cap - (n_step + history)));
+                    local = (int)((lo + (int)(rng.uniform() * span)) % ring_cap);
+                }
+                leaf = (int64_t)r * ring_cap + local;
+            }
+            if (leaf < 0) leaf = 0;  // empty replay guard (caller prevents)
+        }
+        prob = fmaxf(tree[tree_size + leaf], 1e-12f * fmaxf(total, 1.0f)) / total;
+        out_idx[b] = (int32_t)leaf;
+    }
+    // importance weights: w = (N * P)^-beta / max_b w
+    float w = (b < batch) ? __powf((float)total_count_s * prob, -beta) : 0.0f;
+    __syncthreads();
+    float wmax = block_reduce_max(w, lds);
+    if (threadIdx.x == 0) lds[0] = wmax;
+    __syncthreads();
+    if (b < batch) out_weights[b] = w / lds[0];
+    if (threadIdx.x == 0) rng_state[1] += 1;  // advance philox offset
+}
+
+__global__ void sample_assemble_kernel(
+    const uint8_t* frames, const int32_t* actions, const float* rewards,
+    const uint8_t* terminals, const int32_t* timesteps, int64_t ring_cap,
+    int history, int n_step, float discount, const int32_t* in_idx,
+    float* out_states, float* out_next_states, int64_t* out_actions,
+    float* out_returns, float* out_nonterminals) {
+    int b = blockIdx.x;
+    int64_t leaf = in_idx[b];
+    int r = (int)(leaf / ring_cap);
+    int local = (int)(leaf % ring_cap);
+    int64_t base = (int64_t)r * ring_cap;
+
+    __shared__ float ret_s;
+    __shared__ int alive_s;
+    if (threadIdx.x == 0) {
+        float ret = 0.0f, g = 1.0f;
+        int alive = 1;
+        for (int s = 0; s < n_step; ++s) {
+            int64_t j = base + (local + s) % ring_cap;
+            ret += g * rewards[j];
+            if (terminals[j]) { alive = 0; break; }
+            g *= discount;
+        }
+        ret_s = ret;
+        alive_s = alive;
+        out_actions[b] = (int64_t)actions[leaf];
+        out_returns[b] = ret;
+        out_nonterminals[b] = (float)alive;
+    }
+    __syncthreads();
+    int alive = alive_s;
+
+    const float inv255 = 1.0f / 255.0f;
+    int64_t out_base = (int64_t)b * history * FRAME_PX;
+    int ts_cur = timesteps[leaf];
+    int local_n = (local + n_step) % (int)ring_cap;
+    int ts_next = timesteps[base + local_n];
+    for (int j = 0; j < history; ++j) {
+        int back = history - 1 - j;
+        // state frame j
+        float* dst = out_states + out_base + (int64_t)j * FRAME_PX;
+        if (ts_cur >= back) {
+            const uint8_t* src = frames + (base + (local - back + ring_cap) % ring_cap) * (int64_t)FRAME_PX;
+            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x)
+                dst[i] = (float)src[i] * inv255;
+        } else {
+            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x) dst[i] = 0.0f;
+        }
+        // next-state frame j (zeros when the n-step window hit a terminal)
+        float* dstn = out_next_states + out_base + (int64_t)j * FRAME_PX;
+        if (alive && ts_next >= back) {
+            const uint8_t* srcn = frames + (base + (local_n - back + ring_cap) % ring_cap) * (int64_t)FRAME_PX;
+            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x)
+                dstn[i] = (float)srcn[i] * inv255;
+        } else {
+            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x) dstn[i] = 0.0f;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// PRIORITY UPDATE
+// ---------------------------------------------------------------------------
+__global__ void update_priorities_kernel(
+    float* tree, int64_t tree_size, float* max_priority, const int32_t* idxs,
+    const float* priorities, int64_t k, float omega) {
+    int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    if (i >= k) return;
+    float p = priorities[i];
+    if (!isfinite(p)) p = 1.0f;
+    p = fmaxf(__powf(fabsf(p), omega), 1e-8f);
+    tree_set_leaf(tree, tree_size, idxs[i], p);
+    atomic_max_pos_float(max_priority, p);
+}
+
+__global__ void tree_level_pass_kernel(float* tree, int64_t level_nodes) {
+    // parents at [level_nodes, 2*level_nodes)
+    for (int64_t i = level_nodes + threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+         i < 2 * level_nodes; i += (int64_t)gridDim.x * blockDim.x)
+        tree[i] = tree[2 * i] + tree[2 * i + 1];
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// C ABI wrappers
+// ---------------------------------------------------------------------------
+extern "C" {
+
+void riqn_replay_append(
+    uint8_t* frames, int32_t* actions, float* rewards, uint8_t* terminals,
+    int32_t* timesteps, float* tree, int64_t tree_size, int64_t capacity,
+    int32_t* ring_state, int n_rings, int64_t ring_cap, float* max_priority,
+    const uint8_t* chunk_frames, const float* chunk_meta, int64_t chunk_len,
+    int ring_id, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    hipLaunchKernelGGL(append_meta_kernel, dim3(1), dim3(64), 0, s,
+                       actions, rewards, terminals, timesteps, ring_state,
+                       ring_id, ring_cap, chunk_meta, chunk_len);
+    hipLaunchKernelGGL(append_frames_kernel, dim3((uint32_t)chunk_len), dim3(256), 0, s,
+                       frames, ring_state, ring_id, ring_cap, chunk_frames,
+                       chunk_len);
+    int blocks = (int)((chunk_len + 255) / 256);
+    hipLaunchKernelGGL(append_tree_kernel, dim3(blocks), dim3(256), 0, s,
+                       tree, tree_size, ring_state, ring_id, ring_cap,
+                       max_priority, chunk_len);
+}
+
+void riqn_replay_sample(
+    const uint8_t* frames, const int32_t* actions, const float* rewards,
+    const uint8_t* terminals, const int32_t* timesteps, const float* tree,
+    int64_t tree_size, int64_t capacity, const int32_t* ring_state, int n_rings,
+    int64_t ring_cap, int history, int n_step, float discount,
+    const float* beta, uint64_t* rng_state, int batch, int32_t* out_idx,
+    float* out_states, float* out_next_states, int64_t* out_actions,
+    float* out_returns, float* out_nonterminals, float* out_weights,
+    void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int bt = ((batch + 63) / 64) * 64;
+    hipLaunchKernelGGL(sample_indices_kernel, dim3(1), dim3(bt), 0, s,
+                       (float*)tree, tree_size, capacity, ring_state, n_rings,
+                       ring_cap, history, n_step, beta, rng_state, batch,
+                       out_idx, out_weights);
+    hipLaunchKernelGGL(sample_assemble_kernel, dim3(batch), dim3(256), 0, s,
+                       frames, actions, rewards, terminals, timesteps,
+                       ring_cap, history, n_step, discount, out_idx,
+                       out_states, out_next_states, out_actions, out_returns,
+                       out_nonterminals);
+}
+
+void riqn_replay_update_priorities(
+    float* tree, int64_t tree_size, float* max_priority, const int32_t* idxs,
+    const float* priorities, int64_t k, float priority_exponent, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int blocks = (int)((k + 255) / 256);
+    hipLaunchKernelGGL(update_priorities_kernel, dim3(blocks), dim3(256), 0, s,
+                       tree, tree_size, max_priority, idxs, priorities, k,
+                       priority_exponent);
+}
+
+void riqn_tree_level_pass(float* tree, int64_t level_nodes, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int blocks = (int)min((level_nodes + 255) / 256, (int64_t)2048);
+    hipLaunchKernelGGL(tree_level_pass_kernel, dim3(blocks), dim3(256), 0, s,
+                       tree, level_nodes);
+}
+
+}  // extern "C"

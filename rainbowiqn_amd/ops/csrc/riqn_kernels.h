@@ -1,0 +1,66 @@
+// C ABI between the torch bindings (bindings.cpp, built against torch headers)
+// and the HIP kernels (*.hip, built by hipcc for gfx950 only).  Plain types
+// keep torch headers out of device compilation.
+#pragma once
+#include <cstdint>
+
+extern "C" {
+
+// ---- replay (GPU prioritized sum-tree) ----------------------
+// meta/ring_state layout documented in replay.hip.
+
+void riqn_replay_append(
+    // replay state
+    uint8_t* frames, int32_t* actions, float* rewards, uint8_t* terminals,
+    int32_t* timesteps, float* tree, int64_t tree_size, int64_t capacity,
+    int32_t* ring_state, int n_rings, int64_t ring_cap, float* max_priority,
+    // chunk
+    const uint8_t* chunk_frames, const float* chunk_meta, int64_t chunk_len,
+    int ring_id, void* stream);
+
+void riqn_replay_sample(
+    const uint8_t* frames, const int32_t* actions, const float* rewards,
+    const uint8_t* terminals, const int32_t* timesteps, const float* tree,
+    int64_t tree_size, int64_t capacity, const int32_t* ring_state, int n_rings,
+    int64_t ring_cap, int history, int n_step, float discount,
+    const float* beta,       // [1] device scalar (annealed outside the graph)
+    uint64_t* rng_state,     // [2] device (seed, offset)
+    int batch,
+    // outputs
+    int32_t* out_idx, float* out_states, float* out_next_states,
+    int64_t* out_actions, float* out_returns, float* out_nonterminals,
+    float* out_weights, void* stream);
+
+void riqn_replay_update_priorities(
+    float* tree, int64_t tree_size, float* max_priority,
+    const int32_t* idxs, const float* priorities, int64_t k,
+    float priority_exponent, void* stream);
+
+void riqn_tree_level_pass(float* tree, int64_t level_nodes, void* stream);
+
+// ---- fused IQN quantile-Huber loss (K10+K11+K12 loss part) -----------------
+// Computes loss, per-sample priorities, and d(loss)/d(theta) in one kernel.
+void riqn_quantile_huber(
+    const float* theta,      // [B, N] online quantiles (taken action)
+    const float* target_q,   // [B, Np] target-net quantiles Z'(s', a*)
+    const float* taus,       // [B, N]
+    const float* returns,    // [B]
+    const float* nonterm,    // [B]
+    const float* weights,    // [B] or nullptr
+    float gamma_n, float kappa, int B, int N, int Np,
+    float* out_loss,         // [1] (mean over batch of weighted per-sample)
+    float* out_per_sample,   // [B]
+    float* out_grad_theta,   // [B, N]  d(out_loss)/d(theta)
+    void* stream);
+
+// ---- fused Adam over the flat parameter buffer (K13) -----------------------
+void riqn_adam_step(
+    float* param, const float* grad, float* exp_avg, float* exp_avg_sq,
+    int64_t numel, float lr, float beta1, float beta2, float eps,
+    int32_t* step,           // [1] device step counter (graph-safe)
+    void* stream);
+
+// ---- frame preprocess: u8 -> f32/255 (K15) ---------------------------------
+void riqn_preprocess(const uint8_t* in, float* out, int64_t numel, void* stream);
+
+}  // extern "C"

@@ -106,6 +106,15 @@ class ReplayMemory:
         return (state * 255).astype(np.uint8)
 
     # -- sampling ----------------------------------------------------------
+    def _uniform_valid_slots(self, k: int) -> np.ndarray:
+        """Uniform draw over slots whose [t-h+1, t+n] windows are in-bounds."""
+        if self.count < self.capacity:
+            span = max(1, self.count - self.n - 1)
+            return np.random.randint(0, span, k)
+        lo = (self.index + self.history - 1) % self.capacity
+        span = max(1, self.capacity - (self.n + self.history))
+        return (lo + np.random.randint(0, span, k)) % self.capacity
+
     def _valid(self, idxs: np.ndarray) -> np.ndarray:
         """A sampled slot t is valid iff the windows [t-h+1, t+n] do not cross
         the ring write head and the slot has nonzero priority."""
@@ -130,8 +139,14 @@ class ReplayMemory:
         rng = np.random
         idxs = np.zeros(batch_size, dtype=np.int64)
         pending = np.arange(batch_size)
-        for _ in range(64):  # bounded rejection loop
-            u = (pending + rng.random(len(pending))) * seg
+        for attempt in range(64):  # bounded rejection loop
+            if attempt < 8:
+                # stratified: redraw within each element's own segment
+                u = (pending + rng.random(len(pending))) * seg
+            else:
+                # a segment can lie entirely inside the invalid window near
+                # the write head -- redraw stragglers over the full range
+                u = rng.random(len(pending)) * total
             cand = self.tree.find(np.minimum(u, total * (1 - 1e-12)))
             idxs[pending] = cand
             good = self._valid(cand)
@@ -139,13 +154,12 @@ class ReplayMemory:
             if len(pending) == 0:
                 break
         else:
-            # fall back: redraw failures uniformly over definitely-valid slots
-            safe_lo = (self.index + self.history) % self.capacity
-            span = max(1, self.count - self.history - self.n - 1)
-            idxs[pending] = (safe_lo + rng.randint(0, span, len(pending))) % self.capacity
+            # last resort: uniform over the definitely-valid index region
+            idxs[pending] = self._uniform_valid_slots(len(pending))
 
         states, next_states, returns, nonterminals = self._assemble(idxs)
-        probs = self.tree.get(idxs) / total
+        # floor: last-resort fallback slots can carry (near-)zero priority
+        probs = np.maximum(self.tree.get(idxs), 1e-12 * max(total, 1.0)) / total
         n_valid = self.count
         weights = (n_valid * probs) ** -self.priority_weight
         weights = weights / weights.max()
@@ -199,7 +213,11 @@ class ReplayMemory:
             priorities.detach().cpu().numpy() if isinstance(priorities, torch.Tensor) else priorities,
             dtype=np.float64,
         )
-        self.tree.update(np.asarray(idxs, dtype=np.int64), p ** self.priority_exponent)
+        # floor keeps every stored transition reachable and the tree finite
+        # (a zero/NaN priority would otherwise poison totals and IS weights)
+        p = np.where(np.isfinite(p), np.abs(p), 1.0)
+        p = np.maximum(p ** self.priority_exponent, 1e-8)
+        self.tree.update(np.asarray(idxs, dtype=np.int64), p)
 
     # -- checkpoint (reference: redis RDB persistence) ----------------------
     def state_dict(self) -> dict:
