@@ -44,11 +44,14 @@ def build_runner(cfg, comm, n_envs, chunk_steps):
     # the benchmark shape so grad steps run from the first timed round (the
     # timed region must contain full learner work; data is synthetic either way).
     if comm.is_learner:
+        from rainbowiqn_amd.env import n_actions_for
+
+        n_act = n_actions_for(cfg.game)
         rng = np.random.default_rng(0)
         per_ring = cfg.learn_start // comm.n_actors + 256
         for a in range(comm.n_actors):
             frames = rng.integers(0, 255, size=(per_ring, 84, 84), dtype=np.uint8)
-            actions = rng.integers(0, runner.agent.action_space, size=per_ring)
+            actions = rng.integers(0, n_act, size=per_ring)
             rewards = rng.random(per_ring).astype(np.float32)
             dones = rng.random(per_ring) < 0.01
             runner.replay.append_actor_chunk(a, frames, actions, rewards, dones)

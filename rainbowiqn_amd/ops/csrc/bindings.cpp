@@ -131,6 +131,47 @@ void preprocess(torch::Tensor in, torch::Tensor out) {
                     cur_stream());
 }
 
+void rand_uniform(torch::Tensor out, torch::Tensor rng_state) {
+    CHECK_T(out, torch::kFloat32);
+    riqn_rand_uniform(out.data_ptr<float>(), out.numel(),
+                      (uint64_t*)rng_state.data_ptr<int64_t>(), cur_stream());
+}
+
+void noisy_reset(torch::Tensor eps_in, torch::Tensor eps_out,
+                 torch::Tensor weight_eps, torch::Tensor bias_eps,
+                 torch::Tensor rng_state) {
+    CHECK_T(weight_eps, torch::kFloat32);
+    riqn_noisy_reset(eps_in.data_ptr<float>(), eps_in.numel(),
+                     eps_out.data_ptr<float>(), eps_out.numel(),
+                     weight_eps.data_ptr<float>(), bias_eps.data_ptr<float>(),
+                     (uint64_t*)rng_state.data_ptr<int64_t>(), cur_stream());
+}
+
+void env_step(torch::Tensor actions, torch::Tensor env_state,
+              torch::Tensor rng_state, torch::Tensor ring, int64_t history,
+              int64_t ring_head_new, double reward_clip, double p_done,
+              int64_t max_episode_frames, torch::Tensor chunk_frames,
+              torch::Tensor chunk_meta, int64_t chunk_pos) {
+    CHECK_T(actions, torch::kInt64);
+    CHECK_T(ring, torch::kUInt8);
+    int n_envs = actions.size(0);
+    riqn_env_step(actions.data_ptr<int64_t>(), env_state.data_ptr<int32_t>(),
+                  (uint64_t*)rng_state.data_ptr<int64_t>(),
+                  ring.data_ptr<uint8_t>(), (int)history, (int)ring_head_new,
+                  n_envs, (float)reward_clip, (float)p_done,
+                  (int)max_episode_frames, chunk_frames.data_ptr<uint8_t>(),
+                  chunk_meta.data_ptr<float>(), (int)chunk_pos, cur_stream());
+}
+
+void stack_gather(torch::Tensor ring, torch::Tensor env_state, int64_t history,
+                  int64_t ring_head, int64_t n_envs, torch::Tensor out) {
+    CHECK_T(ring, torch::kUInt8);
+    CHECK_T(out, torch::kFloat32);
+    riqn_stack_gather(ring.data_ptr<uint8_t>(), env_state.data_ptr<int32_t>(),
+                      (int)history, (int)ring_head, (int)n_envs,
+                      out.data_ptr<float>(), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -141,4 +182,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("quantile_huber", &quantile_huber);
     m.def("adam_step", &adam_step);
     m.def("preprocess", &preprocess);
+    m.def("rand_uniform", &rand_uniform);
+    m.def("noisy_reset", &noisy_reset);
+    m.def("env_step", &env_step);
+    m.def("stack_gather", &stack_gather);
 }

@@ -63,4 +63,20 @@ void riqn_adam_step(
 // ---- frame preprocess: u8 -> f32/255 (K15) ---------------------------------
 void riqn_preprocess(const uint8_t* in, float* out, int64_t numel, void* stream);
 
+// ---- device counter RNG fills (K4/K7) --------------------------------------
+void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state, void* stream);
+void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
+                      int64_t n_out, float* weight_eps, float* bias_eps,
+                      uint64_t* rng_state, void* stream);
+
+// ---- device synthetic env (bench frame source; env.hip) --------------------
+void riqn_env_step(const int64_t* actions, int32_t* env_state,
+                   uint64_t* rng_state, uint8_t* ring, int history,
+                   int ring_head_new, int n_envs, float reward_clip,
+                   float p_done, int max_episode_frames, uint8_t* chunk_frames,
+                   float* chunk_meta, int chunk_pos, void* stream);
+void riqn_stack_gather(const uint8_t* ring, const int32_t* env_state,
+                       int history, int ring_head, int n_envs, float* out,
+                       void* stream);
+
 }  // extern "C"

@@ -1,0 +1,89 @@
+// Device-side counter RNG fills: uniform tau draws and
+// factorised NoisyLinear epsilon regeneration, all hipGraph-replayable (the
+// philox offset lives in device memory and is bumped by the kernels).
+#include "common.h"
+#include "riqn_kernels.h"
+
+namespace {
+
+__global__ void rand_uniform_kernel(float* __restrict__ out, int64_t n,
+                                    uint64_t* rng_state) {
+    uint64_t seed = rng_state[0], off = rng_state[1];
+    int64_t tid = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    Philox rng(seed, off, (uint32_t)tid);
+    float u[4];
+    rng.uniform4(u);
+    int64_t base = tid * 4;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+        if (base + i < n) out[base + i] = u[i];
+    if (tid == 0) rng_state[1] = off + 1;
+}
+
+// f(x) = sign(x) * sqrt(|x|) of a standard normal
+DEV float scale_noise(float z) { return copysignf(sqrtf(fabsf(z)), z); }
+
+__global__ void noisy_eps_vec_kernel(float* __restrict__ eps_in, int64_t n_in,
+                                     float* __restrict__ eps_out,
+                                     int64_t n_out, uint64_t* rng_state) {
+    uint64_t seed = rng_state[0], off = rng_state[1];
+    int64_t tid = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    Philox rng(seed, off, (uint32_t)(tid + 0x10000));
+    float z0, z1;
+    rng.normal2(z0, z1);
+    int64_t i = tid * 2;
+    float* buf;
+    // first n_in/2 thread-pairs fill eps_in, rest fill eps_out
+    if (i < n_in) {
+        if (i < n_in) eps_in[i] = scale_noise(z0);
+        if (i + 1 < n_in) eps_in[i + 1] = scale_noise(z1);
+    } else {
+        int64_t j = i - n_in;
+        if (j < n_out) eps_out[j] = scale_noise(z0);
+        if (j + 1 < n_out) eps_out[j + 1] = scale_noise(z1);
+    }
+    (void)buf;
+    if (tid == 0) rng_state[1] = off + 1;
+}
+
+__global__ void noisy_eps_outer_kernel(const float* __restrict__ eps_in,
+                                       int64_t n_in,
+                                       const float* __restrict__ eps_out,
+                                       int64_t n_out,
+                                       float* __restrict__ weight_eps,
+                                       float* __restrict__ bias_eps) {
+    for (int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+         i < n_in * n_out; i += (int64_t)gridDim.x * blockDim.x)
+        weight_eps[i] = eps_out[i / n_in] * eps_in[i % n_in];
+    int64_t t = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    if (t < n_out) bias_eps[t] = eps_out[t];
+}
+
+}  // namespace
+
+extern "C" {
+
+void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state,
+                       void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int64_t threads = (n + 3) / 4;
+    int blocks = (int)min((threads + 255) / 256, (int64_t)2048);
+    hipLaunchKernelGGL(rand_uniform_kernel, dim3(max(blocks, 1)), dim3(256), 0,
+                       s, out, n, rng_state);
+}
+
+void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
+                      int64_t n_out, float* weight_eps, float* bias_eps,
+                      uint64_t* rng_state, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int64_t pairs = (n_in + n_out + 1) / 2;
+    int blocks = (int)((pairs + 255) / 256);
+    hipLaunchKernelGGL(noisy_eps_vec_kernel, dim3(max(blocks, 1)), dim3(256),
+                       0, s, eps_in, n_in, eps_out, n_out, rng_state);
+    int64_t total = n_in * n_out;
+    int blocks2 = (int)min((total + 255) / 256, (int64_t)2048);
+    hipLaunchKernelGGL(noisy_eps_outer_kernel, dim3(blocks2), dim3(256), 0, s,
+                       eps_in, n_in, eps_out, n_out, weight_eps, bias_eps);
+}
+
+}  // extern "C"

@@ -140,7 +140,14 @@ class ActorState:
 
 
 class ApexRunner:
-    """Fixed-cadence Ape-X round loop shared by the launch scripts and bench."""
+    """Fixed-cadence Ape-X round loop shared by the launch scripts and bench.
+
+    Two execution paths, same round semantics:
+      * fast path (GPU + HIP extension): hipGraph-captured GpuLearner /
+        GpuActorLoop with on-GPU replay and device-resident synthetic env;
+      * reference path (CPU or --no-hip-kernels): eager Agent + CPU sharded
+        replay + numpy env -- also what the gloo CI tests exercise.
+    """
 
     def __init__(self, cfg, comm: Communicator, n_envs: int = 16,
                  chunk_steps: Optional[int] = None, replay_capacity: Optional[int] = None,
@@ -148,21 +155,106 @@ class ApexRunner:
         self.cfg = cfg
         self.comm = comm
         dev = device or str(comm.device)
-        self.agent = Agent(cfg, VectorSyntheticEnv(cfg, 1).action_space(), dev)
-        self.chunk_steps = chunk_steps or max(1, cfg.transition_chunk_size // n_envs)
         self.n_envs = n_envs
-        self.actor: Optional[ActorState] = None
-        self.replay: Optional[ShardedReplay] = None
-        self.flat_weights = self.agent.flat_weights()
-        if comm.is_actor:
-            self.actor = ActorState(cfg, comm, self.agent, n_envs, self.chunk_steps)
-        if comm.is_learner:
-            self.replay = ShardedReplay(cfg, comm.n_actors, replay_capacity)
+        self.chunk_steps = chunk_steps or max(1, cfg.transition_chunk_size // n_envs)
         self.grad_steps = 0
         self.losses: List[float] = []
+        self.actor = None
+        self.replay = None
+        self.glearner = None
+        self.gactor = None
+
+        from ..env import n_actions_for
+
+        n_act = n_actions_for(cfg.game)
+        self.fast = False
+        if torch.device(dev).type == "cuda" and cfg.use_hip_kernels:
+            from ..ops import extension_available
+
+            self.fast = extension_available()
+            if not self.fast:
+                from ..ops import get_extension
+
+                get_extension()  # raises: native path is mandatory on GPU
+
+        if self.fast:
+            self._init_fast(cfg, comm, dev, n_act, replay_capacity)
+        else:
+            self.agent = Agent(cfg, n_act, dev)
+            self.flat_weights = self.agent.flat_weights()
+            if comm.is_actor:
+                self.actor = ActorState(cfg, comm, self.agent, n_envs, self.chunk_steps)
+            if comm.is_learner:
+                self.replay = ShardedReplay(cfg, comm.n_actors, replay_capacity)
+
+    def _init_fast(self, cfg, comm, dev, n_act, replay_capacity):
+        from ..models.model import RainbowIQN
+        from ..ops.flat_params import flatten_module
+        from ..runtime.gpu_actor import GpuActorLoop
+        from ..runtime.gpu_learner import GpuLearner
+
+        # chunk_steps must be a multiple of history for the actor ring phase
+        h = cfg.history_length
+        self.chunk_steps = max(h, (self.chunk_steps // h) * h)
+        if comm.is_learner:
+            if replay_capacity is not None:
+                cfg_cap = cfg
+                cfg_cap.memory_capacity = replay_capacity
+            self.glearner = GpuLearner(cfg, n_act, n_rings=comm.n_actors,
+                                       device=dev, use_graph=cfg.hipgraph)
+            self.replay = self.glearner.replay
+            self.flat_weights = self.glearner.flat_online
+        if comm.is_actor:
+            if comm.world_size == 1:
+                net = self.glearner.online  # colocated: share the online net
+            else:
+                net = RainbowIQN(cfg, n_act).to(dev)
+                self.flat_weights = flatten_module(net)
+            self.gactor = GpuActorLoop(cfg, net, self.n_envs, self.chunk_steps,
+                                       device=dev, use_graph=cfg.hipgraph,
+                                       seed_offset=comm.rank * 7717)
+            self.actor = self.gactor  # exposes .total_steps
 
     # -- one Ape-X round ----------------------------------------------------
     def run_round(self, learn: bool = True) -> None:
+        if self.fast:
+            self._run_round_fast(learn)
+        else:
+            self._run_round_ref(learn)
+
+    def _run_round_fast(self, learn: bool) -> None:
+        cfg = self.cfg
+        comm = self.comm
+        if self.gactor is not None:
+            self.gactor.reset_noise()
+            self.gactor.run_round()
+            chunk_f, chunk_m = self.gactor.chunk_frames, self.gactor.chunk_meta
+        else:
+            c = self.n_envs * self.chunk_steps
+            chunk_f = torch.zeros(c, 84, 84, dtype=torch.uint8, device=comm.device)
+            chunk_m = torch.zeros(c, 3, device=comm.device)
+
+        if comm.world_size > 1:
+            frames_list = comm.gather_chunks(chunk_f)
+            meta_list = comm.gather_chunks(chunk_m)
+        else:
+            frames_list, meta_list = [chunk_f], [chunk_m]
+        if comm.is_learner:
+            for a, (f, m) in enumerate(zip(frames_list, meta_list)):
+                self.glearner.append_actor_chunk(a, f, meta=m)
+            if learn and len(self.glearner.replay) >= cfg.learn_start:
+                self.glearner.learn_steps(cfg.learn_per_round)
+                self.grad_steps = self.glearner.grad_steps
+                self.losses.append(self.glearner.last_loss())
+                tgt = max(1, cfg.target_update // max(1, cfg.replay_frequency))
+                if self.grad_steps % tgt < cfg.learn_per_round:
+                    self.glearner.update_target()
+                    self.glearner.maintenance()
+
+        if comm.world_size > 1:
+            comm.broadcast_weights(self.flat_weights, src=0)
+
+    def _run_round_ref(self, learn: bool) -> None:
         cfg = self.cfg
         # 1. actors generate a chunk of experience
         if self.actor is not None:

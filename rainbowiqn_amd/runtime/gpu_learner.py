@@ -1,0 +1,257 @@
+"""hipGraph-captured learner and actor steps (MI355X fast path).
+
+The learner's whole step -- GPU-replay sample, online/target forward (bf16),
+fused quantile-Huber loss + gradient, backward, fused Adam, priority
+write-back, device-RNG tau draws and NoisyNet noise regeneration -- is
+captured once into a hip graph and replayed per grad step: one ~10 us replay
+instead of ~10^2 eager launches at 3-4 us each (MI355X_MICROARCH.md
+boundary/graph-replay-floor rows).  Everything dynamic lives in device
+buffers (philox offsets, Adam step counter, ring heads), so replays are
+self-advancing; host-side knobs (beta anneal, target sync, tree rebuild)
+touch device buffers between replays.
+
+The actor act step (kernels K1-K9+K15) is captured the same
+way: static state buffer in, greedy action buffer out.
+"""
+
+from __future__ import annotations
+
+import contextlib
+from typing import Optional
+
+import torch
+
+from ..models.model import RainbowIQN
+from ..ops import get_extension
+from ..ops.flat_params import FusedAdam, attach_flat_grads, flatten_module
+from ..ops.functional import quantile_huber_loss_fused
+from ..replay.gpu_replay import GpuReplay
+
+
+def _noisy_layers(net: RainbowIQN):
+    return [net.fc_h_v, net.fc_h_a, net.fc_z_v, net.fc_z_a]
+
+
+class GpuLearner:
+    def __init__(self, cfg, action_space: int, n_rings: int = 1,
+                 device: str = "cuda", use_graph: bool = True):
+        self.cfg = cfg
+        self.ext = get_extension()
+        self.device = torch.device(device)
+        self.action_space = action_space
+        self.B = cfg.batch_size
+        self.N = cfg.num_tau_samples
+        self.Np = cfg.num_tau_prime_samples
+        self.K = cfg.num_quantile_samples
+
+        self.online = RainbowIQN(cfg, action_space).to(self.device)
+        self.target = RainbowIQN(cfg, action_space).to(self.device)
+        self.online.train()
+        self.target.train()  # target uses its own noise draws (reference learn())
+        self.flat_online = flatten_module(self.online)
+        self.flat_target = flatten_module(self.target)
+        self.flat_target.copy_(self.flat_online)
+        for p in self.target.parameters():
+            p.requires_grad_(False)
+        self.flat_grad = attach_flat_grads(self.online)
+        self.opt = FusedAdam(self.flat_online, self.flat_grad, lr=cfg.lr, eps=cfg.adam_eps)
+
+        self.replay = GpuReplay(cfg, n_rings=n_rings, device=device)
+        self.rng_state = torch.tensor(
+            [cfg.seed * 7919 + 13, 0], dtype=torch.int64, device=self.device
+        )
+        b = self.B
+        self.taus_online = torch.empty(b * self.N, 1, device=self.device)
+        self.taus_target = torch.empty(b * self.Np, 1, device=self.device)
+        self.taus_select = torch.empty(b * self.K, 1, device=self.device)
+        # scratch for noisy eps vectors (max fan 3136)
+        self._eps_in = torch.empty(4096, device=self.device)
+        self._eps_out = torch.empty(4096, device=self.device)
+        self.loss_static = torch.zeros(1, device=self.device)
+        self.use_graph = use_graph
+        self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self.grad_steps = 0
+        self._autocast_on = cfg.dtype == "bf16"
+
+    # -- pieces -------------------------------------------------------------
+    def _reset_noise_device(self, net: RainbowIQN) -> None:
+        for layer in _noisy_layers(net):
+            self.ext.noisy_reset(
+                self._eps_in[: layer.in_features],
+                self._eps_out[: layer.out_features],
+                layer.weight_epsilon, layer.bias_epsilon, self.rng_state,
+            )
+
+    def _autocast(self):
+        if self._autocast_on:
+            return torch.autocast("cuda", dtype=torch.bfloat16, cache_enabled=False)
+        return contextlib.nullcontext()
+
+    def _step_body(self):
+        cfg = self.cfg
+        b = self.B
+        ext = self.ext
+        ext.rand_uniform(self.taus_online, self.rng_state)
+        ext.rand_uniform(self.taus_target, self.rng_state)
+        ext.rand_uniform(self.taus_select, self.rng_state)
+        self._reset_noise_device(self.online)
+        self._reset_noise_device(self.target)
+        idx, s, a, r, s2, nt, w = self.replay.sample()
+        with self._autocast():
+            z, _ = self.online(s, self.N, taus=self.taus_online)
+            theta = z.gather(2, a.view(b, 1, 1).expand(b, self.N, 1)).squeeze(2)
+            with torch.no_grad():
+                zsel, _ = self.online(s2, self.K, taus=self.taus_select)
+                a_star = zsel.mean(dim=1).argmax(dim=1)
+                zp, _ = self.target(s2, self.Np, taus=self.taus_target)
+                theta_p = zp.gather(
+                    2, a_star.view(b, 1, 1).expand(b, self.Np, 1)
+                ).squeeze(2)
+        loss, per_sample = quantile_huber_loss_fused(
+            theta.float(), theta_p.float(),
+            self.taus_online.view(b, self.N).float(), r, nt, w,
+            cfg.discount, cfg.multi_step, cfg.kappa,
+        )
+        self.opt.zero_grad()
+        loss.backward()
+        self.opt.step()
+        self.replay.update_priorities(idx, per_sample)
+        self.loss_static.copy_(loss.detach().view(1))
+
+    # -- graph lifecycle -----------------------------------------------------
+    def capture(self) -> bool:
+        """Capture the learn step into a hip graph (after replay has data)."""
+        if not self.use_graph:
+            return False
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):  # warmup: MIOpen find, workspace allocs
+                self._step_body()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._step_body()
+        self.graph = g
+        return True
+
+    # -- public API -----------------------------------------------------------
+    def append_actor_chunk(self, ring_id: int, frames, meta=None, **kw) -> None:
+        self.replay.append_actor_chunk(ring_id, frames, meta=meta, **kw)
+
+    def learn_steps(self, k: int) -> None:
+        if self.graph is None and self.use_graph and len(self.replay) >= self.cfg.learn_start:
+            self.capture()
+            # the 3 warmup iterations executed real steps; the capture pass
+            # only records (kernels do not run during stream capture)
+            self.grad_steps += 3
+            k = max(0, k - 3)
+        for _ in range(k):
+            if self.graph is not None:
+                self.graph.replay()
+            else:
+                self._step_body()
+        self.grad_steps += k
+
+    def last_loss(self) -> float:
+        return float(self.loss_static.item())
+
+    def update_target(self) -> None:
+        self.flat_target.copy_(self.flat_online)
+
+    def anneal_beta(self, beta: float) -> None:
+        self.replay.beta.fill_(min(1.0, beta))
+
+    def maintenance(self) -> None:
+        """Periodic fp32-drift cleanup of the tree (outside the graph)."""
+        self.replay.rebuild_tree()
+
+    # -- checkpoint (reference-parity state_dict) ------------------------------
+    def state_dict(self):
+        return self.online.state_dict()
+
+    def save(self, path: str, name: str = "model.pth") -> None:
+        import os
+
+        os.makedirs(path, exist_ok=True)
+        torch.save(self.online.state_dict(), os.path.join(path, name))
+
+
+class GraphedActor:
+    """hipGraph-captured greedy act step over a batch of E env streams.
+
+    In: static u8 frame-stack buffer [E, h, 84, 84]; out: static action
+    buffer [E].  NoisyNet exploration noise and tau draws are regenerated
+    in-graph from device philox state.
+    """
+
+    def __init__(self, cfg, net: RainbowIQN, n_envs: int, device="cuda",
+                 use_graph: bool = True, seed_offset: int = 0):
+        self.cfg = cfg
+        self.ext = get_extension()
+        self.net = net
+        self.E = n_envs
+        self.K = cfg.num_quantile_samples
+        self.device = torch.device(device)
+        self.stack_u8 = torch.zeros(n_envs, cfg.history_length, 84, 84,
+                                    dtype=torch.uint8, device=self.device)
+        self.stack_f32 = torch.zeros(n_envs, cfg.history_length, 84, 84,
+                                     device=self.device)
+        self.taus = torch.empty(n_envs * self.K, 1, device=self.device)
+        self.actions = torch.zeros(n_envs, dtype=torch.int64, device=self.device)
+        self.rng_state = torch.tensor(
+            [cfg.seed * 104729 + 7 + seed_offset, 0], dtype=torch.int64,
+            device=self.device,
+        )
+        self._autocast_on = cfg.dtype == "bf16"
+        self.use_graph = use_graph
+        self.graph: Optional[torch.cuda.CUDAGraph] = None
+
+    def _autocast(self):
+        if self._autocast_on:
+            return torch.autocast("cuda", dtype=torch.bfloat16, cache_enabled=False)
+        return contextlib.nullcontext()
+
+    def _body(self):
+        self.ext.preprocess(self.stack_u8, self.stack_f32)
+        self.ext.rand_uniform(self.taus, self.rng_state)
+        with torch.no_grad(), self._autocast():
+            z, _ = self.net(self.stack_f32, self.K, taus=self.taus)
+            self.actions.copy_(z.float().mean(dim=1).argmax(dim=1))
+
+    def reset_noise(self) -> None:
+        for layer in _noisy_layers(self.net):
+            nin = torch.empty(layer.in_features, device=self.device)
+            nout = torch.empty(layer.out_features, device=self.device)
+            self.ext.noisy_reset(nin, nout, layer.weight_epsilon,
+                                 layer.bias_epsilon, self.rng_state)
+
+    def capture(self) -> None:
+        if not self.use_graph:
+            return
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                self._body()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._body()
+        self.graph = g
+
+    def act(self) -> torch.Tensor:
+        """Run the captured act step on the current stack_u8; returns actions."""
+        if self.graph is None:
+            if self.use_graph:
+                self.capture()
+                self.graph.replay()
+            else:
+                self._body()
+        else:
+            self.graph.replay()
+        return self.actions

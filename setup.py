@@ -23,6 +23,8 @@ ext = CUDAExtension(
         os.path.join(CSRC, "replay.hip"),
         os.path.join(CSRC, "loss.hip"),
         os.path.join(CSRC, "adam.hip"),
+        os.path.join(CSRC, "rng.hip"),
+        os.path.join(CSRC, "env.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
