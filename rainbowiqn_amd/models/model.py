@@ -110,12 +110,15 @@ class RainbowIQN(nn.Module):
     # -- pieces (exposed so kernel-parity tests can target each stage) -----
     def encode(self, x: torch.Tensor) -> torch.Tensor:
         """x: [B, history, 84, 84] in [0,1] -> [B, 3136]."""
+        w_dtype = self.convs[0].weight.dtype
+        if x.dtype != w_dtype:
+            x = x.to(w_dtype)
         return self.convs(x).flatten(1)
 
     def embed_taus(self, taus: torch.Tensor) -> torch.Tensor:
         """taus: [R, 1] -> [R, 3136] cosine embedding through iqn_fc."""
-        cos = torch.cos(taus * self.cos_basis)  # [R, 64]
-        return F.relu(self.iqn_fc(cos))
+        cos = torch.cos(taus.float() * self.cos_basis.float())  # [R, 64]
+        return F.relu(self.iqn_fc(cos.to(self.iqn_fc.weight.dtype)))
 
     def head(self, h: torch.Tensor) -> torch.Tensor:
         """h: [R, 3136] -> quantile Q values [R, A] (dueling combine)."""

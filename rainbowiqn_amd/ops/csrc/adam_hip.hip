@@ -36,6 +36,28 @@ __global__ void adam_kernel(float* __restrict__ param,
     }
 }
 
+__global__ void adam_mixed_kernel(
+    float* __restrict__ master, const unsigned short* __restrict__ grad_bf16,
+    unsigned short* __restrict__ param_bf16, float* __restrict__ m,
+    float* __restrict__ v, int64_t numel, float lr, float beta1, float beta2,
+    float eps, const int32_t* __restrict__ step) {
+    int t = step[0];
+    float bc1 = 1.0f - __powf(beta1, (float)t);
+    float bc2 = 1.0f - __powf(beta2, (float)t);
+    float step_size = lr / bc1;
+    for (int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x; i < numel;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float g = bf16_to_f32(grad_bf16[i]);
+        float mi = beta1 * m[i] + (1.0f - beta1) * g;
+        float vi = beta2 * v[i] + (1.0f - beta2) * g * g;
+        m[i] = mi;
+        v[i] = vi;
+        float p = master[i] - step_size * mi / (sqrtf(vi / bc2) + eps);
+        master[i] = p;
+        param_bf16[i] = f32_to_bf16(p);
+    }
+}
+
 __global__ void preprocess_kernel(const uint8_t* __restrict__ in,
                                   float* __restrict__ out, int64_t numel) {
     const float inv255 = 1.0f / 255.0f;
@@ -65,6 +87,20 @@ void riqn_adam_step(float* param, const float* grad, float* exp_avg,
     int blocks = (int)min((numel + 255) / 256, (int64_t)2048);
     hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, s, param, grad,
                        exp_avg, exp_avg_sq, numel, lr, beta1, beta2, eps, step);
+}
+
+void riqn_adam_step_mixed(float* master, const uint16_t* grad_bf16,
+                          uint16_t* param_bf16, float* exp_avg,
+                          float* exp_avg_sq, int64_t numel, float lr,
+                          float beta1, float beta2, float eps, int32_t* step,
+                          void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    hipLaunchKernelGGL(bump_step_kernel, dim3(1), dim3(64), 0, s, step);
+    int blocks = (int)min((numel + 255) / 256, (int64_t)2048);
+    hipLaunchKernelGGL(adam_mixed_kernel, dim3(blocks), dim3(256), 0, s,
+                       master, (const unsigned short*)grad_bf16,
+                       (unsigned short*)param_bf16, exp_avg, exp_avg_sq, numel,
+                       lr, beta1, beta2, eps, step);
 }
 
 void riqn_preprocess(const uint8_t* in, float* out, int64_t numel,

@@ -50,7 +50,10 @@ void replay_sample(torch::Tensor frames, torch::Tensor actions,
                    torch::Tensor out_next_states, torch::Tensor out_actions,
                    torch::Tensor out_returns, torch::Tensor out_nonterminals,
                    torch::Tensor out_weights) {
-    CHECK_T(out_states, torch::kFloat32);
+    CHECK_DEV(out_states); CHECK_CONT(out_states);
+    bool out_bf16 = out_states.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(out_bf16 || out_states.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(out_next_states.scalar_type() == out_states.scalar_type());
     CHECK_T(out_actions, torch::kInt64);
     CHECK_T(out_idx, torch::kInt32);
     TORCH_CHECK(batch <= 512, "sample batch must be <= 512 (single block)");
@@ -61,10 +64,10 @@ void replay_sample(torch::Tensor frames, torch::Tensor actions,
         capacity, ring_state.data_ptr<int32_t>(), (int)n_rings, ring_cap,
         (int)history, (int)n_step, (float)discount, beta.data_ptr<float>(),
         (uint64_t*)rng_state.data_ptr<int64_t>(), (int)batch,
-        out_idx.data_ptr<int32_t>(), out_states.data_ptr<float>(),
-        out_next_states.data_ptr<float>(), out_actions.data_ptr<int64_t>(),
+        out_idx.data_ptr<int32_t>(), (float*)out_states.data_ptr(),
+        (float*)out_next_states.data_ptr(), out_actions.data_ptr<int64_t>(),
         out_returns.data_ptr<float>(), out_nonterminals.data_ptr<float>(),
-        out_weights.data_ptr<float>(), cur_stream());
+        out_weights.data_ptr<float>(), out_bf16 ? 1 : 0, cur_stream());
 }
 
 void replay_update_priorities(torch::Tensor tree, int64_t tree_size,
@@ -140,11 +143,31 @@ void rand_uniform(torch::Tensor out, torch::Tensor rng_state) {
 void noisy_reset(torch::Tensor eps_in, torch::Tensor eps_out,
                  torch::Tensor weight_eps, torch::Tensor bias_eps,
                  torch::Tensor rng_state) {
-    CHECK_T(weight_eps, torch::kFloat32);
+    bool eps_bf16 = weight_eps.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(eps_bf16 || weight_eps.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(bias_eps.scalar_type() == weight_eps.scalar_type());
     riqn_noisy_reset(eps_in.data_ptr<float>(), eps_in.numel(),
                      eps_out.data_ptr<float>(), eps_out.numel(),
-                     weight_eps.data_ptr<float>(), bias_eps.data_ptr<float>(),
+                     (float*)weight_eps.data_ptr(), (float*)bias_eps.data_ptr(),
+                     eps_bf16 ? 1 : 0,
                      (uint64_t*)rng_state.data_ptr<int64_t>(), cur_stream());
+}
+
+void adam_step_mixed(torch::Tensor master, torch::Tensor grad_bf16,
+                     torch::Tensor param_bf16, torch::Tensor exp_avg,
+                     torch::Tensor exp_avg_sq, double lr, double beta1,
+                     double beta2, double eps, torch::Tensor step) {
+    CHECK_T(master, torch::kFloat32);
+    CHECK_T(grad_bf16, torch::kBFloat16);
+    CHECK_T(param_bf16, torch::kBFloat16);
+    CHECK_T(step, torch::kInt32);
+    riqn_adam_step_mixed(master.data_ptr<float>(),
+                         (const uint16_t*)grad_bf16.data_ptr(),
+                         (uint16_t*)param_bf16.data_ptr(),
+                         exp_avg.data_ptr<float>(),
+                         exp_avg_sq.data_ptr<float>(), master.numel(),
+                         (float)lr, (float)beta1, (float)beta2, (float)eps,
+                         step.data_ptr<int32_t>(), cur_stream());
 }
 
 void env_step(torch::Tensor actions, torch::Tensor env_state,
@@ -166,10 +189,11 @@ void env_step(torch::Tensor actions, torch::Tensor env_state,
 void stack_gather(torch::Tensor ring, torch::Tensor env_state, int64_t history,
                   int64_t ring_head, int64_t n_envs, torch::Tensor out) {
     CHECK_T(ring, torch::kUInt8);
-    CHECK_T(out, torch::kFloat32);
+    bool out_bf16 = out.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(out_bf16 || out.scalar_type() == torch::kFloat32);
     riqn_stack_gather(ring.data_ptr<uint8_t>(), env_state.data_ptr<int32_t>(),
                       (int)history, (int)ring_head, (int)n_envs,
-                      out.data_ptr<float>(), cur_stream());
+                      (float*)out.data_ptr(), out_bf16 ? 1 : 0, cur_stream());
 }
 
 }  // namespace
@@ -185,5 +209,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rand_uniform", &rand_uniform);
     m.def("noisy_reset", &noisy_reset);
     m.def("env_step", &env_step);
+    m.def("adam_step_mixed", &adam_step_mixed);
     m.def("stack_gather", &stack_gather);
 }

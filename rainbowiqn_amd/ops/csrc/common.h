@@ -104,6 +104,24 @@ DEV float block_reduce_max(float v, float* lds) {
     return v;
 }
 
+// f32 -> bf16 (round-to-nearest-even)
+DEV unsigned short f32_to_bf16(float f) {
+    unsigned int b = __float_as_uint(f);
+    b += 0x7FFF + ((b >> 16) & 1);
+    return (unsigned short)(b >> 16);
+}
+DEV float bf16_to_f32(unsigned short h) {
+    unsigned int b = ((unsigned int)h) << 16;
+    return __uint_as_float(b);
+}
+struct ushort4_t { unsigned short x, y, z, w; };
+DEV ushort4_t f32x4_to_bf16x4(float a, float b, float c, float d) {
+    ushort4_t r;
+    r.x = f32_to_bf16(a); r.y = f32_to_bf16(b);
+    r.z = f32_to_bf16(c); r.w = f32_to_bf16(d);
+    return r;
+}
+
 // atomicMax for non-negative floats via ordered uint bit pattern
 DEV void atomic_max_pos_float(float* addr, float val) {
     atomicMax((unsigned int*)addr, __float_as_uint(val));

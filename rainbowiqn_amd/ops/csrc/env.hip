@@ -103,6 +103,7 @@ __global__ void env_step_kernel(
     if (threadIdx.x == 0 && e == 0) rng_state[1] += 1;
 }
 
+template <int OUT_BF16>
 __global__ void stack_gather_kernel(
     const uint8_t* __restrict__ ring, const int32_t* __restrict__ env_state,
     int history, int ring_head,  // slot of the NEWEST frame
@@ -112,15 +113,35 @@ __global__ void stack_gather_kernel(
     const float inv255 = 1.0f / 255.0f;
     for (int j = 0; j < history; ++j) {
         int back = history - 1 - j;
-        float* dst = out + ((int64_t)e * history + j) * FRAME_PX;
-        if (steps >= back) {
-            int slot = (ring_head - back + 2 * history) % history;
-            const uint8_t* src = ring + ((int64_t)slot * n_envs + e) * FRAME_PX;
-            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x)
-                dst[i] = (float)src[i] * inv255;
+        int64_t off = ((int64_t)e * history + j) * FRAME_PX;
+        bool fill = steps >= back;
+        int slot = (ring_head - back + 2 * history) % history;
+        const uchar4* src =
+            (const uchar4*)(ring + ((int64_t)slot * n_envs + e) * FRAME_PX);
+        if (OUT_BF16) {
+            ushort4_t* d4 = (ushort4_t*)((unsigned short*)(void*)out + off);
+            ushort4_t z = {0, 0, 0, 0};
+            for (int i = threadIdx.x; i < FRAME_PX / 4; i += blockDim.x) {
+                if (fill) {
+                    uchar4 u = src[i];
+                    d4[i] = f32x4_to_bf16x4(u.x * inv255, u.y * inv255,
+                                            u.z * inv255, u.w * inv255);
+                } else {
+                    d4[i] = z;
+                }
+            }
         } else {
-            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x)
-                dst[i] = 0.0f;
+            float4* d4 = (float4*)(out + off);
+            float4 z = make_float4(0.f, 0.f, 0.f, 0.f);
+            for (int i = threadIdx.x; i < FRAME_PX / 4; i += blockDim.x) {
+                if (fill) {
+                    uchar4 u = src[i];
+                    d4[i] = make_float4(u.x * inv255, u.y * inv255,
+                                        u.z * inv255, u.w * inv255);
+                } else {
+                    d4[i] = z;
+                }
+            }
         }
     }
 }
@@ -143,10 +164,15 @@ void riqn_env_step(const int64_t* actions, int32_t* env_state,
 
 void riqn_stack_gather(const uint8_t* ring, const int32_t* env_state,
                        int history, int ring_head, int n_envs, float* out,
-                       void* stream) {
-    hipLaunchKernelGGL(stack_gather_kernel, dim3(n_envs), dim3(256), 0,
-                       (hipStream_t)stream, ring, env_state, history,
-                       ring_head, n_envs, out);
+                       int out_bf16, void* stream) {
+    if (out_bf16)
+        hipLaunchKernelGGL(stack_gather_kernel<1>, dim3(n_envs), dim3(256), 0,
+                           (hipStream_t)stream, ring, env_state, history,
+                           ring_head, n_envs, out);
+    else
+        hipLaunchKernelGGL(stack_gather_kernel<0>, dim3(n_envs), dim3(256), 0,
+                           (hipStream_t)stream, ring, env_state, history,
+                           ring_head, n_envs, out);
 }
 
 }  // extern "C"

@@ -200,62 +200,88 @@ __global__ void sample_indices_kernel(
     if (threadIdx.x == 0) rng_state[1] += 1;  // advance philox offset
 }
 
-__global__ void sample_assemble_kernel(
-    const uint8_t* frames, const int32_t* actions, const float* rewards,
-    const uint8_t* terminals, const int32_t* timesteps, int64_t ring_cap,
-    int history, int n_step, float discount, const int32_t* in_idx,
-    float* out_states, float* out_next_states, int64_t* out_actions,
-    float* out_returns, float* out_nonterminals) {
-    int b = blockIdx.x;
+__global__ void sample_scalars_kernel(
+    const int32_t* actions, const float* rewards, const uint8_t* terminals,
+    int64_t ring_cap, int n_step, float discount, const int32_t* in_idx,
+    int batch, int64_t* out_actions, float* out_returns,
+    float* out_nonterminals) {
+    int b = threadIdx.x + blockIdx.x * blockDim.x;
+    if (b >= batch) return;
     int64_t leaf = in_idx[b];
     int r = (int)(leaf / ring_cap);
     int local = (int)(leaf % ring_cap);
     int64_t base = (int64_t)r * ring_cap;
-
-    __shared__ float ret_s;
-    __shared__ int alive_s;
-    if (threadIdx.x == 0) {
-        float ret = 0.0f, g = 1.0f;
-        int alive = 1;
-        for (int s = 0; s < n_step; ++s) {
-            int64_t j = base + (local + s) % ring_cap;
-            ret += g * rewards[j];
-            if (terminals[j]) { alive = 0; break; }
-            g *= discount;
-        }
-        ret_s = ret;
-        alive_s = alive;
-        out_actions[b] = (int64_t)actions[leaf];
-        out_returns[b] = ret;
-        out_nonterminals[b] = (float)alive;
+    float ret = 0.0f, g = 1.0f;
+    int alive = 1;
+    for (int s = 0; s < n_step; ++s) {
+        int64_t j = base + (local + s) % ring_cap;
+        ret += g * rewards[j];
+        if (terminals[j]) { alive = 0; break; }
+        g *= discount;
     }
-    __syncthreads();
-    int alive = alive_s;
+    out_actions[b] = (int64_t)actions[leaf];
+    out_returns[b] = ret;
+    out_nonterminals[b] = (float)alive;
+}
 
+// grid (batch, 2*history): block (b, j<h: state frame j; j>=h: next frame)
+// OUT_BF16: write __hip_bfloat16-packed ushort instead of f32 (native-bf16
+// model path -- no separate cast kernels in the learn graph)
+template <int OUT_BF16>
+__global__ void sample_assemble_kernel(
+    const uint8_t* __restrict__ frames, const uint8_t* __restrict__ terminals,
+    const int32_t* __restrict__ timesteps, int64_t ring_cap, int history,
+    int n_step, const int32_t* __restrict__ in_idx,
+    const float* __restrict__ out_nonterminals, float* __restrict__ out_states,
+    float* __restrict__ out_next_states) {
+    int b = blockIdx.x;
+    int jj = blockIdx.y;
+    bool is_next = jj >= history;
+    int j = is_next ? jj - history : jj;
+    int back = history - 1 - j;
+    int64_t leaf = in_idx[b];
+    int r = (int)(leaf / ring_cap);
+    int local = (int)(leaf % ring_cap);
+    int64_t base = (int64_t)r * ring_cap;
+    bool fill;
+    int src_local;
+    int64_t elem_off = ((int64_t)b * history + j) * FRAME_PX;
+    void* out = is_next ? (void*)out_next_states : (void*)out_states;
+    if (!is_next) {
+        fill = timesteps[leaf] >= back;
+        src_local = (local - back + (int)ring_cap) % (int)ring_cap;
+    } else {
+        int local_n = (local + n_step) % (int)ring_cap;
+        bool alive = out_nonterminals[b] > 0.5f;
+        fill = alive && timesteps[base + local_n] >= back;
+        src_local = (local_n - back + (int)ring_cap) % (int)ring_cap;
+    }
     const float inv255 = 1.0f / 255.0f;
-    int64_t out_base = (int64_t)b * history * FRAME_PX;
-    int ts_cur = timesteps[leaf];
-    int local_n = (local + n_step) % (int)ring_cap;
-    int ts_next = timesteps[base + local_n];
-    for (int j = 0; j < history; ++j) {
-        int back = history - 1 - j;
-        // state frame j
-        float* dst = out_states + out_base + (int64_t)j * FRAME_PX;
-        if (ts_cur >= back) {
-            const uint8_t* src = frames + (base + (local - back + ring_cap) % ring_cap) * (int64_t)FRAME_PX;
-            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x)
-                dst[i] = (float)src[i] * inv255;
-        } else {
-            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x) dst[i] = 0.0f;
+    const uchar4* src =
+        (const uchar4*)(frames + (base + src_local) * (int64_t)FRAME_PX);
+    if (OUT_BF16) {
+        ushort4_t* d4 = (ushort4_t*)((unsigned short*)out + elem_off);
+        ushort4_t z = {0, 0, 0, 0};
+        for (int i = threadIdx.x; i < FRAME_PX / 4; i += blockDim.x) {
+            if (fill) {
+                uchar4 u = src[i];
+                d4[i] = f32x4_to_bf16x4(u.x * inv255, u.y * inv255,
+                                        u.z * inv255, u.w * inv255);
+            } else {
+                d4[i] = z;
+            }
         }
-        // next-state frame j (zeros when the n-step window hit a terminal)
-        float* dstn = out_next_states + out_base + (int64_t)j * FRAME_PX;
-        if (alive && ts_next >= back) {
-            const uint8_t* srcn = frames + (base + (local_n - back + ring_cap) % ring_cap) * (int64_t)FRAME_PX;
-            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x)
-                dstn[i] = (float)srcn[i] * inv255;
-        } else {
-            for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x) dstn[i] = 0.0f;
+    } else {
+        float4* d4 = (float4*)((float*)out + elem_off);
+        float4 z = make_float4(0.f, 0.f, 0.f, 0.f);
+        for (int i = threadIdx.x; i < FRAME_PX / 4; i += blockDim.x) {
+            if (fill) {
+                uchar4 u = src[i];
+                d4[i] = make_float4(u.x * inv255, u.y * inv255, u.z * inv255,
+                                    u.w * inv255);
+            } else {
+                d4[i] = z;
+            }
         }
     }
 }
@@ -316,18 +342,29 @@ void riqn_replay_sample(
     const float* beta, uint64_t* rng_state, int batch, int32_t* out_idx,
     float* out_states, float* out_next_states, int64_t* out_actions,
     float* out_returns, float* out_nonterminals, float* out_weights,
-    void* stream) {
+    int out_bf16, void* stream) {
     hipStream_t s = (hipStream_t)stream;
     int bt = ((batch + 63) / 64) * 64;
     hipLaunchKernelGGL(sample_indices_kernel, dim3(1), dim3(bt), 0, s,
                        (float*)tree, tree_size, capacity, ring_state, n_rings,
                        ring_cap, history, n_step, beta, rng_state, batch,
                        out_idx, out_weights);
-    hipLaunchKernelGGL(sample_assemble_kernel, dim3(batch), dim3(256), 0, s,
-                       frames, actions, rewards, terminals, timesteps,
-                       ring_cap, history, n_step, discount, out_idx,
-                       out_states, out_next_states, out_actions, out_returns,
-                       out_nonterminals);
+    hipLaunchKernelGGL(sample_scalars_kernel, dim3((batch + 63) / 64),
+                       dim3(64), 0, s, actions, rewards, terminals, ring_cap,
+                       n_step, discount, out_idx, batch, out_actions,
+                       out_returns, out_nonterminals);
+    if (out_bf16)
+        hipLaunchKernelGGL(sample_assemble_kernel<1>,
+                           dim3(batch, 2 * history), dim3(256), 0, s, frames,
+                           terminals, timesteps, ring_cap, history, n_step,
+                           out_idx, out_nonterminals, out_states,
+                           out_next_states);
+    else
+        hipLaunchKernelGGL(sample_assemble_kernel<0>,
+                           dim3(batch, 2 * history), dim3(256), 0, s, frames,
+                           terminals, timesteps, ring_cap, history, n_step,
+                           out_idx, out_nonterminals, out_states,
+                           out_next_states);
 }
 
 void riqn_replay_update_priorities(

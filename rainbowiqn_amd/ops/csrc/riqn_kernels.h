@@ -29,7 +29,7 @@ void riqn_replay_sample(
     // outputs
     int32_t* out_idx, float* out_states, float* out_next_states,
     int64_t* out_actions, float* out_returns, float* out_nonterminals,
-    float* out_weights, void* stream);
+    float* out_weights, int out_bf16, void* stream);
 
 void riqn_replay_update_priorities(
     float* tree, int64_t tree_size, float* max_priority,
@@ -60,6 +60,13 @@ void riqn_adam_step(
     int32_t* step,           // [1] device step counter (graph-safe)
     void* stream);
 
+// mixed-precision Adam: fp32 master + bf16 grads -> fp32 master AND a bf16
+// parameter copy (the module weights) updated in one pass
+void riqn_adam_step_mixed(
+    float* master, const uint16_t* grad_bf16, uint16_t* param_bf16,
+    float* exp_avg, float* exp_avg_sq, int64_t numel, float lr, float beta1,
+    float beta2, float eps, int32_t* step, void* stream);
+
 // ---- frame preprocess: u8 -> f32/255 (K15) ---------------------------------
 void riqn_preprocess(const uint8_t* in, float* out, int64_t numel, void* stream);
 
@@ -67,7 +74,7 @@ void riqn_preprocess(const uint8_t* in, float* out, int64_t numel, void* stream)
 void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state, void* stream);
 void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
                       int64_t n_out, float* weight_eps, float* bias_eps,
-                      uint64_t* rng_state, void* stream);
+                      int eps_bf16, uint64_t* rng_state, void* stream);
 
 // ---- device synthetic env (bench frame source; env.hip) --------------------
 void riqn_env_step(const int64_t* actions, int32_t* env_state,
@@ -77,6 +84,6 @@ void riqn_env_step(const int64_t* actions, int32_t* env_state,
                    float* chunk_meta, int chunk_pos, void* stream);
 void riqn_stack_gather(const uint8_t* ring, const int32_t* env_state,
                        int history, int ring_head, int n_envs, float* out,
-                       void* stream);
+                       int out_bf16, void* stream);
 
 }  // extern "C"

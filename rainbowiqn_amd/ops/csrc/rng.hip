@@ -46,6 +46,7 @@ __global__ void noisy_eps_vec_kernel(float* __restrict__ eps_in, int64_t n_in,
     if (tid == 0) rng_state[1] = off + 1;
 }
 
+template <int OUT_BF16>
 __global__ void noisy_eps_outer_kernel(const float* __restrict__ eps_in,
                                        int64_t n_in,
                                        const float* __restrict__ eps_out,
@@ -53,10 +54,20 @@ __global__ void noisy_eps_outer_kernel(const float* __restrict__ eps_in,
                                        float* __restrict__ weight_eps,
                                        float* __restrict__ bias_eps) {
     for (int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
-         i < n_in * n_out; i += (int64_t)gridDim.x * blockDim.x)
-        weight_eps[i] = eps_out[i / n_in] * eps_in[i % n_in];
+         i < n_in * n_out; i += (int64_t)gridDim.x * blockDim.x) {
+        float v = eps_out[i / n_in] * eps_in[i % n_in];
+        if (OUT_BF16)
+            ((unsigned short*)(void*)weight_eps)[i] = f32_to_bf16(v);
+        else
+            weight_eps[i] = v;
+    }
     int64_t t = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
-    if (t < n_out) bias_eps[t] = eps_out[t];
+    if (t < n_out) {
+        if (OUT_BF16)
+            ((unsigned short*)(void*)bias_eps)[t] = f32_to_bf16(eps_out[t]);
+        else
+            bias_eps[t] = eps_out[t];
+    }
 }
 
 }  // namespace
@@ -74,7 +85,7 @@ void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state,
 
 void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
                       int64_t n_out, float* weight_eps, float* bias_eps,
-                      uint64_t* rng_state, void* stream) {
+                      int eps_bf16, uint64_t* rng_state, void* stream) {
     hipStream_t s = (hipStream_t)stream;
     int64_t pairs = (n_in + n_out + 1) / 2;
     int blocks = (int)((pairs + 255) / 256);
@@ -82,8 +93,14 @@ void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
                        0, s, eps_in, n_in, eps_out, n_out, rng_state);
     int64_t total = n_in * n_out;
     int blocks2 = (int)min((total + 255) / 256, (int64_t)2048);
-    hipLaunchKernelGGL(noisy_eps_outer_kernel, dim3(blocks2), dim3(256), 0, s,
-                       eps_in, n_in, eps_out, n_out, weight_eps, bias_eps);
+    if (eps_bf16)
+        hipLaunchKernelGGL(noisy_eps_outer_kernel<1>, dim3(blocks2), dim3(256),
+                           0, s, eps_in, n_in, eps_out, n_out, weight_eps,
+                           bias_eps);
+    else
+        hipLaunchKernelGGL(noisy_eps_outer_kernel<0>, dim3(blocks2), dim3(256),
+                           0, s, eps_in, n_in, eps_out, n_out, weight_eps,
+                           bias_eps);
 }
 
 }  // extern "C"

@@ -27,7 +27,7 @@ def flatten_module(module: nn.Module) -> torch.Tensor:
     """
     params: List[nn.Parameter] = [p for p in module.parameters()]
     total = sum(p.numel() for p in params)
-    flat = torch.empty(total, dtype=torch.float32,
+    flat = torch.empty(total, dtype=params[0].dtype,
                        device=params[0].device if params else "cpu")
     off = 0
     for p in params:
@@ -43,7 +43,7 @@ def attach_flat_grads(module: nn.Module) -> torch.Tensor:
     """Give every parameter a .grad view into one flat buffer and return it."""
     params = [p for p in module.parameters()]
     total = sum(p.numel() for p in params)
-    flat_grad = torch.zeros(total, dtype=torch.float32, device=params[0].device)
+    flat_grad = torch.zeros(total, dtype=params[0].dtype, device=params[0].device)
     off = 0
     for p in params:
         k = p.numel()
@@ -83,3 +83,57 @@ class FusedAdam:
         self.m.copy_(sd["m"])
         self.v.copy_(sd["v"])
         self.step_t.copy_(sd["step"])
+
+
+class FusedAdamMixed:
+    """fp32 master + bf16 module weights/grads: one kernel reads the bf16
+    grads, updates the fp32 master and writes the bf16 params (K13 + the
+    master->weight cast fused)."""
+
+    def __init__(self, flat_param_bf16: torch.Tensor, flat_grad_bf16: torch.Tensor,
+                 lr: float, eps: float, betas=(0.9, 0.999)):
+        assert flat_param_bf16.dtype == torch.bfloat16
+        self.ext = get_extension()
+        self.param = flat_param_bf16
+        self.grad = flat_grad_bf16
+        self.master = flat_param_bf16.float()
+        self.m = torch.zeros_like(self.master)
+        self.v = torch.zeros_like(self.master)
+        self.step_t = torch.zeros(1, dtype=torch.int32, device=flat_param_bf16.device)
+        self.lr = lr
+        self.eps = eps
+        self.betas = betas
+
+    def zero_grad(self) -> None:
+        self.grad.zero_()
+
+    def step(self) -> None:
+        self.ext.adam_step_mixed(self.master, self.grad, self.param, self.m,
+                                 self.v, self.lr, self.betas[0], self.betas[1],
+                                 self.eps, self.step_t)
+
+    def state_dict(self) -> dict:
+        return {"m": self.m, "v": self.v, "step": self.step_t,
+                "master": self.master}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.m.copy_(sd["m"])
+        self.v.copy_(sd["v"])
+        self.step_t.copy_(sd["step"])
+        self.master.copy_(sd["master"])
+        self.param.copy_(self.master.to(torch.bfloat16))
+
+
+def master_state_dict(module: nn.Module, master: torch.Tensor) -> dict:
+    """Reference-format fp32 state_dict from the fp32 master flat buffer
+    (params) + the module's buffers (cast to fp32)."""
+    sd = {}
+    off = 0
+    pmap = {}
+    for name, p in module.named_parameters():
+        k = p.numel()
+        pmap[name] = master[off : off + k].view(p.shape).clone()
+        off += k
+    for name, t in module.state_dict().items():
+        sd[name] = pmap[name] if name in pmap else t.float().clone()
+    return sd

@@ -208,7 +208,8 @@ class ApexRunner:
             if comm.world_size == 1:
                 net = self.glearner.online  # colocated: share the online net
             else:
-                net = RainbowIQN(cfg, n_act).to(dev)
+                dt = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
+                net = RainbowIQN(cfg, n_act).to(dev).to(dt)
                 self.flat_weights = flatten_module(net)
             self.gactor = GpuActorLoop(cfg, net, self.n_envs, self.chunk_steps,
                                        device=dev, use_graph=cfg.hipgraph,
@@ -226,7 +227,7 @@ class ApexRunner:
         cfg = self.cfg
         comm = self.comm
         if self.gactor is not None:
-            self.gactor.reset_noise()
+            # NoisyNet noise reset happens inside the captured round graph
             self.gactor.run_round()
             chunk_f, chunk_m = self.gactor.chunk_frames, self.gactor.chunk_meta
         else:

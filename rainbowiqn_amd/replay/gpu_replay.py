@@ -31,7 +31,7 @@ def _next_pow2(n: int) -> int:
 class GpuReplay:
     def __init__(self, cfg, n_rings: int = 1, capacity: Optional[int] = None,
                  device: str = "cuda", batch_size: Optional[int] = None,
-                 seed: Optional[int] = None):
+                 seed: Optional[int] = None, out_dtype: torch.dtype = torch.float32):
         self.ext = get_extension()
         self.cfg = cfg
         self.device = torch.device(device)
@@ -62,8 +62,9 @@ class GpuReplay:
         b = self.batch_size
         # static output buffers (graph-capture targets)
         self.out_idx = torch.zeros(b, dtype=torch.int32, device=d)
-        self.out_states = torch.zeros(b, self.history, 84, 84, device=d)
-        self.out_next_states = torch.zeros(b, self.history, 84, 84, device=d)
+        # bf16 output feeds the native-bf16 model path with no cast kernels
+        self.out_states = torch.zeros(b, self.history, 84, 84, dtype=out_dtype, device=d)
+        self.out_next_states = torch.zeros(b, self.history, 84, 84, dtype=out_dtype, device=d)
         self.out_actions = torch.zeros(b, dtype=torch.int64, device=d)
         self.out_returns = torch.zeros(b, device=d)
         self.out_nonterminals = torch.zeros(b, device=d)

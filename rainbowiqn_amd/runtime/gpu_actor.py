@@ -12,7 +12,6 @@ chunk (u8 frames + f32 meta) feeds the RCCL gather directly.
 
 from __future__ import annotations
 
-import contextlib
 from typing import Optional
 
 import torch
@@ -46,25 +45,23 @@ class GpuActorLoop:
         self.rng_state = torch.tensor(
             [cfg.seed * 48271 + 101 + seed_offset, 0], dtype=torch.int64, device=d
         )
-        self.stack_f32 = torch.zeros(e, self.h, 84, 84, device=d)
+        self.dtype = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
+        self.stack_f32 = torch.zeros(e, self.h, 84, 84, dtype=self.dtype, device=d)
         self.taus = torch.empty(e * self.K, 1, device=d)
         self.actions = torch.zeros(e, dtype=torch.int64, device=d)
         c = chunk_steps * e
         self.chunk_frames = torch.zeros(c, 84, 84, dtype=torch.uint8, device=d)
         self.chunk_meta = torch.zeros(c, 3, device=d)
-        self._autocast_on = cfg.dtype == "bf16"
         self.use_graph = use_graph
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.total_steps = 0
         self._newest = 0  # ring slot of the most recent frame (round-invariant)
-
-    def _autocast(self):
-        if self._autocast_on:
-            return torch.autocast("cuda", dtype=torch.bfloat16, cache_enabled=False)
-        return contextlib.nullcontext()
+        self._eps_in = torch.empty(4096, device=d)
+        self._eps_out = torch.empty(4096, device=d)
 
     def _round_body(self) -> None:
         cfg = self.cfg
+        self.reset_noise()  # in-graph: device philox, fixed launch sequence
         newest = self._newest
         for s in range(self.chunk_steps):
             self.ext.stack_gather(
@@ -72,7 +69,7 @@ class GpuActorLoop:
                 newest, self.E, self.stack_f32,
             )
             self.ext.rand_uniform(self.taus, self.rng_state)
-            with torch.no_grad(), self._autocast():
+            with torch.no_grad():
                 z, _ = self.net(self.stack_f32, self.K, taus=self.taus)
                 self.actions.copy_(z.float().mean(dim=1).argmax(dim=1))
             newest = (newest + 1) % self.h
@@ -87,10 +84,10 @@ class GpuActorLoop:
     def reset_noise(self) -> None:
         for layer in (self.net.fc_h_v, self.net.fc_h_a, self.net.fc_z_v,
                       self.net.fc_z_a):
-            nin = torch.empty(layer.in_features, device=self.device)
-            nout = torch.empty(layer.out_features, device=self.device)
-            self.ext.noisy_reset(nin, nout, layer.weight_epsilon,
-                                 layer.bias_epsilon, self.rng_state)
+            self.ext.noisy_reset(self._eps_in[: layer.in_features],
+                                 self._eps_out[: layer.out_features],
+                                 layer.weight_epsilon, layer.bias_epsilon,
+                                 self.rng_state)
 
     def capture(self) -> None:
         torch.cuda.synchronize()

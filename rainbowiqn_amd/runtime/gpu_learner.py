@@ -16,14 +16,14 @@ way: static state buffer in, greedy action buffer out.
 
 from __future__ import annotations
 
-import contextlib
 from typing import Optional
 
 import torch
 
 from ..models.model import RainbowIQN
 from ..ops import get_extension
-from ..ops.flat_params import FusedAdam, attach_flat_grads, flatten_module
+from ..ops.flat_params import (FusedAdam, FusedAdamMixed, attach_flat_grads,
+                               flatten_module, master_state_dict)
 from ..ops.functional import quantile_huber_loss_fused
 from ..replay.gpu_replay import GpuReplay
 
@@ -44,8 +44,14 @@ class GpuLearner:
         self.Np = cfg.num_tau_prime_samples
         self.K = cfg.num_quantile_samples
 
-        self.online = RainbowIQN(cfg, action_space).to(self.device)
-        self.target = RainbowIQN(cfg, action_space).to(self.device)
+        # Native-bf16 weights (cfg.dtype == "bf16"): the modules' parameters,
+        # buffers and activations are bf16 end to end -- no autocast, no
+        # per-forward weight casts -- with an fp32 Adam master kept by the
+        # fused mixed-precision optimizer kernel.
+        self.bf16 = cfg.dtype == "bf16"
+        dt = torch.bfloat16 if self.bf16 else torch.float32
+        self.online = RainbowIQN(cfg, action_space).to(self.device).to(dt)
+        self.target = RainbowIQN(cfg, action_space).to(self.device).to(dt)
         self.online.train()
         self.target.train()  # target uses its own noise draws (reference learn())
         self.flat_online = flatten_module(self.online)
@@ -54,9 +60,14 @@ class GpuLearner:
         for p in self.target.parameters():
             p.requires_grad_(False)
         self.flat_grad = attach_flat_grads(self.online)
-        self.opt = FusedAdam(self.flat_online, self.flat_grad, lr=cfg.lr, eps=cfg.adam_eps)
+        if self.bf16:
+            self.opt = FusedAdamMixed(self.flat_online, self.flat_grad,
+                                      lr=cfg.lr, eps=cfg.adam_eps)
+        else:
+            self.opt = FusedAdam(self.flat_online, self.flat_grad, lr=cfg.lr,
+                                 eps=cfg.adam_eps)
 
-        self.replay = GpuReplay(cfg, n_rings=n_rings, device=device)
+        self.replay = GpuReplay(cfg, n_rings=n_rings, device=device, out_dtype=dt)
         self.rng_state = torch.tensor(
             [cfg.seed * 7919 + 13, 0], dtype=torch.int64, device=self.device
         )
@@ -70,8 +81,8 @@ class GpuLearner:
         self.loss_static = torch.zeros(1, device=self.device)
         self.use_graph = use_graph
         self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self.steps_per_replay = 1
         self.grad_steps = 0
-        self._autocast_on = cfg.dtype == "bf16"
 
     # -- pieces -------------------------------------------------------------
     def _reset_noise_device(self, net: RainbowIQN) -> None:
@@ -81,11 +92,6 @@ class GpuLearner:
                 self._eps_out[: layer.out_features],
                 layer.weight_epsilon, layer.bias_epsilon, self.rng_state,
             )
-
-    def _autocast(self):
-        if self._autocast_on:
-            return torch.autocast("cuda", dtype=torch.bfloat16, cache_enabled=False)
-        return contextlib.nullcontext()
 
     def _step_body(self):
         cfg = self.cfg
@@ -97,16 +103,15 @@ class GpuLearner:
         self._reset_noise_device(self.online)
         self._reset_noise_device(self.target)
         idx, s, a, r, s2, nt, w = self.replay.sample()
-        with self._autocast():
-            z, _ = self.online(s, self.N, taus=self.taus_online)
-            theta = z.gather(2, a.view(b, 1, 1).expand(b, self.N, 1)).squeeze(2)
-            with torch.no_grad():
-                zsel, _ = self.online(s2, self.K, taus=self.taus_select)
-                a_star = zsel.mean(dim=1).argmax(dim=1)
-                zp, _ = self.target(s2, self.Np, taus=self.taus_target)
-                theta_p = zp.gather(
-                    2, a_star.view(b, 1, 1).expand(b, self.Np, 1)
-                ).squeeze(2)
+        z, _ = self.online(s, self.N, taus=self.taus_online)
+        theta = z.gather(2, a.view(b, 1, 1).expand(b, self.N, 1)).squeeze(2)
+        with torch.no_grad():
+            zsel, _ = self.online(s2, self.K, taus=self.taus_select)
+            a_star = zsel.mean(dim=1).argmax(dim=1)
+            zp, _ = self.target(s2, self.Np, taus=self.taus_target)
+            theta_p = zp.gather(
+                2, a_star.view(b, 1, 1).expand(b, self.Np, 1)
+            ).squeeze(2)
         loss, per_sample = quantile_huber_loss_fused(
             theta.float(), theta_p.float(),
             self.taus_online.view(b, self.N).float(), r, nt, w,
@@ -119,10 +124,13 @@ class GpuLearner:
         self.loss_static.copy_(loss.detach().view(1))
 
     # -- graph lifecycle -----------------------------------------------------
-    def capture(self) -> bool:
-        """Capture the learn step into a hip graph (after replay has data)."""
+    def capture(self, steps_per_replay: int = 1) -> bool:
+        """Capture `steps_per_replay` unrolled learn steps into one hip graph
+        (after replay has data): one graph replay per learn round removes all
+        per-step host launch/boundary cost."""
         if not self.use_graph:
             return False
+        self.steps_per_replay = steps_per_replay
         torch.cuda.synchronize()
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
@@ -133,7 +141,8 @@ class GpuLearner:
         torch.cuda.synchronize()
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            self._step_body()
+            for _ in range(steps_per_replay):
+                self._step_body()
         self.graph = g
         return True
 
@@ -143,16 +152,20 @@ class GpuLearner:
 
     def learn_steps(self, k: int) -> None:
         if self.graph is None and self.use_graph and len(self.replay) >= self.cfg.learn_start:
-            self.capture()
+            # one replay per learn round: unroll the round's steps in-graph
+            self.capture(steps_per_replay=max(1, self.cfg.learn_per_round))
             # the 3 warmup iterations executed real steps; the capture pass
             # only records (kernels do not run during stream capture)
             self.grad_steps += 3
             k = max(0, k - 3)
-        for _ in range(k):
-            if self.graph is not None:
+        if self.graph is not None:
+            spr = self.steps_per_replay
+            while k >= spr:
                 self.graph.replay()
-            else:
-                self._step_body()
+                k -= spr
+                self.grad_steps += spr
+        for _ in range(k):
+            self._step_body()
         self.grad_steps += k
 
     def last_loss(self) -> float:
@@ -161,6 +174,10 @@ class GpuLearner:
     def update_target(self) -> None:
         self.flat_target.copy_(self.flat_online)
 
+    @property
+    def master_flat(self) -> torch.Tensor:
+        return self.opt.master if self.bf16 else self.flat_online
+
     def anneal_beta(self, beta: float) -> None:
         self.replay.beta.fill_(min(1.0, beta))
 
@@ -168,15 +185,17 @@ class GpuLearner:
         """Periodic fp32-drift cleanup of the tree (outside the graph)."""
         self.replay.rebuild_tree()
 
-    # -- checkpoint (reference-parity state_dict) ------------------------------
+    # -- checkpoint (reference-parity state_dict: fp32 from the Adam master) ---
     def state_dict(self):
+        if self.bf16:
+            return master_state_dict(self.online, self.opt.master)
         return self.online.state_dict()
 
     def save(self, path: str, name: str = "model.pth") -> None:
         import os
 
         os.makedirs(path, exist_ok=True)
-        torch.save(self.online.state_dict(), os.path.join(path, name))
+        torch.save(self.state_dict(), os.path.join(path, name))
 
 
 class GraphedActor:
@@ -195,36 +214,36 @@ class GraphedActor:
         self.E = n_envs
         self.K = cfg.num_quantile_samples
         self.device = torch.device(device)
+        self.dtype = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
         self.stack_u8 = torch.zeros(n_envs, cfg.history_length, 84, 84,
                                     dtype=torch.uint8, device=self.device)
         self.stack_f32 = torch.zeros(n_envs, cfg.history_length, 84, 84,
-                                     device=self.device)
+                                     dtype=self.dtype, device=self.device)
         self.taus = torch.empty(n_envs * self.K, 1, device=self.device)
         self.actions = torch.zeros(n_envs, dtype=torch.int64, device=self.device)
         self.rng_state = torch.tensor(
             [cfg.seed * 104729 + 7 + seed_offset, 0], dtype=torch.int64,
             device=self.device,
         )
-        self._autocast_on = cfg.dtype == "bf16"
         self.use_graph = use_graph
         self.graph: Optional[torch.cuda.CUDAGraph] = None
 
-    def _autocast(self):
-        if self._autocast_on:
-            return torch.autocast("cuda", dtype=torch.bfloat16, cache_enabled=False)
-        return contextlib.nullcontext()
-
     def _body(self):
-        self.ext.preprocess(self.stack_u8, self.stack_f32)
+        # stack_gather-style preprocess: u8 -> model dtype
+        if self.dtype == torch.float32:
+            self.ext.preprocess(self.stack_u8, self.stack_f32)
+        else:
+            self.stack_f32.copy_(self.stack_u8)
+            self.stack_f32.div_(255.0)
         self.ext.rand_uniform(self.taus, self.rng_state)
-        with torch.no_grad(), self._autocast():
+        with torch.no_grad():
             z, _ = self.net(self.stack_f32, self.K, taus=self.taus)
             self.actions.copy_(z.float().mean(dim=1).argmax(dim=1))
 
     def reset_noise(self) -> None:
         for layer in _noisy_layers(self.net):
-            nin = torch.empty(layer.in_features, device=self.device)
-            nout = torch.empty(layer.out_features, device=self.device)
+            nin = torch.zeros(layer.in_features, device=self.device)
+            nout = torch.zeros(layer.out_features, device=self.device)
             self.ext.noisy_reset(nin, nout, layer.weight_epsilon,
                                  layer.bias_epsilon, self.rng_state)
 
