@@ -71,13 +71,51 @@ class NoisyLinear(nn.Module):
         self.weight_epsilon.copy_(eps_out.outer(eps_in))
         self.bias_epsilon.copy_(eps_out)
 
+    def enable_fused_noise(self) -> None:
+        """GPU fast path: `weight_noisy`/`bias_noisy` buffers are refreshed by
+        the riqn noisy_material kernel once per noise reset, so every forward
+        is a single GEMM (no per-forward w = mu + sigma*eps materialization);
+        backward routes grads to mu/sigma through _NoisyLinearFn."""
+        self.register_buffer("weight_noisy", self.weight_mu.detach().clone(),
+                             persistent=False)
+        self.register_buffer("bias_noisy", self.bias_mu.detach().clone(),
+                             persistent=False)
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.training:
+            if getattr(self, "weight_noisy", None) is not None:
+                return _NoisyLinearFn.apply(
+                    x, self.weight_mu, self.weight_sigma, self.bias_mu,
+                    self.bias_sigma, self.weight_noisy, self.bias_noisy,
+                    self.weight_epsilon, self.bias_epsilon,
+                )
             w = self.weight_mu + self.weight_sigma * self.weight_epsilon
             b = self.bias_mu + self.bias_sigma * self.bias_epsilon
         else:
             w, b = self.weight_mu, self.bias_mu
         return F.linear(x, w, b)
+
+
+class _NoisyLinearFn(torch.autograd.Function):
+    """F.linear on the pre-materialized noisy weight; backward applies the
+    noise chain rule: dmu = dW, dsigma = dW * eps (same for bias)."""
+
+    @staticmethod
+    def forward(ctx, x, w_mu, w_sigma, b_mu, b_sigma, w_noisy, b_noisy,
+                w_eps, b_eps):
+        ctx.save_for_backward(x, w_noisy, w_eps, b_eps)
+        return F.linear(x, w_noisy, b_noisy)
+
+    @staticmethod
+    def backward(ctx, g):
+        x, w_noisy, w_eps, b_eps = ctx.saved_tensors
+        g2 = g.reshape(-1, g.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        grad_x = (g2 @ w_noisy).view_as(x)
+        gw = g2.t() @ x2
+        gb = g2.sum(0)
+        return (grad_x, gw, gw * w_eps, gb, gb * b_eps,
+                None, None, None, None)
 
 
 class RainbowIQN(nn.Module):

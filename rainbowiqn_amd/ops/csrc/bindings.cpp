@@ -153,6 +153,25 @@ void noisy_reset(torch::Tensor eps_in, torch::Tensor eps_out,
                      (uint64_t*)rng_state.data_ptr<int64_t>(), cur_stream());
 }
 
+void noisy_material(torch::Tensor eps_in, torch::Tensor eps_out,
+                    torch::Tensor w_mu, torch::Tensor w_sigma,
+                    torch::Tensor b_mu, torch::Tensor b_sigma,
+                    torch::Tensor weight_eps, torch::Tensor bias_eps,
+                    torch::Tensor w_noisy, torch::Tensor b_noisy,
+                    torch::Tensor rng_state) {
+    bool bf16 = w_mu.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(bf16 || w_mu.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(w_noisy.sizes() == w_mu.sizes());
+    riqn_noisy_material(eps_in.data_ptr<float>(), eps_in.numel(),
+                        eps_out.data_ptr<float>(), eps_out.numel(),
+                        w_mu.data_ptr(), w_sigma.data_ptr(), b_mu.data_ptr(),
+                        b_sigma.data_ptr(), weight_eps.data_ptr(),
+                        bias_eps.data_ptr(), w_noisy.data_ptr(),
+                        b_noisy.data_ptr(), bf16 ? 1 : 0,
+                        (uint64_t*)rng_state.data_ptr<int64_t>(),
+                        cur_stream());
+}
+
 void adam_step_mixed(torch::Tensor master, torch::Tensor grad_bf16,
                      torch::Tensor param_bf16, torch::Tensor exp_avg,
                      torch::Tensor exp_avg_sq, double lr, double beta1,
@@ -210,5 +229,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("noisy_reset", &noisy_reset);
     m.def("env_step", &env_step);
     m.def("adam_step_mixed", &adam_step_mixed);
+    m.def("noisy_material", &noisy_material);
     m.def("stack_gather", &stack_gather);
 }

@@ -75,6 +75,12 @@ void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state, void* stream)
 void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
                       int64_t n_out, float* weight_eps, float* bias_eps,
                       int eps_bf16, uint64_t* rng_state, void* stream);
+void riqn_noisy_material(float* eps_in, int64_t n_in, float* eps_out,
+                         int64_t n_out, const void* w_mu, const void* w_sigma,
+                         const void* b_mu, const void* b_sigma,
+                         void* weight_eps, void* bias_eps, void* w_noisy,
+                         void* b_noisy, int bf16, uint64_t* rng_state,
+                         void* stream);
 
 // ---- device synthetic env (bench frame source; env.hip) --------------------
 void riqn_env_step(const int64_t* actions, int32_t* env_state,

@@ -70,6 +70,50 @@ __global__ void noisy_eps_outer_kernel(const float* __restrict__ eps_in,
     }
 }
 
+// eps outer product AND the materialized noisy weight w = mu + sigma*eps in
+// one pass (K6/K7 fused): the per-forward w/b materialization kernels vanish
+// from the learn/act graphs -- noise (and thus w_noisy) changes once per step.
+template <int BF16>
+__global__ void noisy_material_kernel(
+    const float* __restrict__ eps_in, int64_t n_in,
+    const float* __restrict__ eps_out, int64_t n_out,
+    const void* __restrict__ w_mu, const void* __restrict__ w_sigma,
+    const void* __restrict__ b_mu, const void* __restrict__ b_sigma,
+    void* __restrict__ weight_eps, void* __restrict__ bias_eps,
+    void* __restrict__ w_noisy, void* __restrict__ b_noisy) {
+    for (int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+         i < n_in * n_out; i += (int64_t)gridDim.x * blockDim.x) {
+        float e = eps_out[i / n_in] * eps_in[i % n_in];
+        float mu, sg;
+        if (BF16) {
+            mu = bf16_to_f32(((const unsigned short*)w_mu)[i]);
+            sg = bf16_to_f32(((const unsigned short*)w_sigma)[i]);
+            ((unsigned short*)weight_eps)[i] = f32_to_bf16(e);
+            ((unsigned short*)w_noisy)[i] = f32_to_bf16(mu + sg * e);
+        } else {
+            mu = ((const float*)w_mu)[i];
+            sg = ((const float*)w_sigma)[i];
+            ((float*)weight_eps)[i] = e;
+            ((float*)w_noisy)[i] = mu + sg * e;
+        }
+    }
+    int64_t t = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    if (t < n_out) {
+        float e = eps_out[t];
+        if (BF16) {
+            float mu = bf16_to_f32(((const unsigned short*)b_mu)[t]);
+            float sg = bf16_to_f32(((const unsigned short*)b_sigma)[t]);
+            ((unsigned short*)bias_eps)[t] = f32_to_bf16(e);
+            ((unsigned short*)b_noisy)[t] = f32_to_bf16(mu + sg * e);
+        } else {
+            float mu = ((const float*)b_mu)[t];
+            float sg = ((const float*)b_sigma)[t];
+            ((float*)bias_eps)[t] = e;
+            ((float*)b_noisy)[t] = mu + sg * e;
+        }
+    }
+}
+
 }  // namespace
 
 extern "C" {
@@ -101,6 +145,31 @@ void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
         hipLaunchKernelGGL(noisy_eps_outer_kernel<0>, dim3(blocks2), dim3(256),
                            0, s, eps_in, n_in, eps_out, n_out, weight_eps,
                            bias_eps);
+}
+
+void riqn_noisy_material(float* eps_in, int64_t n_in, float* eps_out,
+                         int64_t n_out, const void* w_mu, const void* w_sigma,
+                         const void* b_mu, const void* b_sigma,
+                         void* weight_eps, void* bias_eps, void* w_noisy,
+                         void* b_noisy, int bf16, uint64_t* rng_state,
+                         void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int64_t pairs = (n_in + n_out + 1) / 2;
+    int blocks = (int)((pairs + 255) / 256);
+    hipLaunchKernelGGL(noisy_eps_vec_kernel, dim3(max(blocks, 1)), dim3(256),
+                       0, s, eps_in, n_in, eps_out, n_out, rng_state);
+    int64_t total = n_in * n_out;
+    int blocks2 = (int)min((total + 255) / 256, (int64_t)2048);
+    if (bf16)
+        hipLaunchKernelGGL(noisy_material_kernel<1>, dim3(blocks2), dim3(256),
+                           0, s, eps_in, n_in, eps_out, n_out, w_mu, w_sigma,
+                           b_mu, b_sigma, weight_eps, bias_eps, w_noisy,
+                           b_noisy);
+    else
+        hipLaunchKernelGGL(noisy_material_kernel<0>, dim3(blocks2), dim3(256),
+                           0, s, eps_in, n_in, eps_out, n_out, w_mu, w_sigma,
+                           b_mu, b_sigma, weight_eps, bias_eps, w_noisy,
+                           b_noisy);
 }
 
 }  // extern "C"

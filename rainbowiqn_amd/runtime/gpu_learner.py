@@ -52,6 +52,9 @@ class GpuLearner:
         dt = torch.bfloat16 if self.bf16 else torch.float32
         self.online = RainbowIQN(cfg, action_space).to(self.device).to(dt)
         self.target = RainbowIQN(cfg, action_space).to(self.device).to(dt)
+        for net in (self.online, self.target):
+            for layer in _noisy_layers(net):
+                layer.enable_fused_noise()
         self.online.train()
         self.target.train()  # target uses its own noise draws (reference learn())
         self.flat_online = flatten_module(self.online)
@@ -87,10 +90,12 @@ class GpuLearner:
     # -- pieces -------------------------------------------------------------
     def _reset_noise_device(self, net: RainbowIQN) -> None:
         for layer in _noisy_layers(net):
-            self.ext.noisy_reset(
+            self.ext.noisy_material(
                 self._eps_in[: layer.in_features],
                 self._eps_out[: layer.out_features],
-                layer.weight_epsilon, layer.bias_epsilon, self.rng_state,
+                layer.weight_mu, layer.weight_sigma, layer.bias_mu,
+                layer.bias_sigma, layer.weight_epsilon, layer.bias_epsilon,
+                layer.weight_noisy, layer.bias_noisy, self.rng_state,
             )
 
     def _step_body(self):
