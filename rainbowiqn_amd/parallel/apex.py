@@ -88,6 +88,16 @@ class ShardedReplay:
             m = ring_ids == a
             self.rings[int(a)].update_priorities(idxs[m], p[m])
 
+    # -- persistence (reference: Redis RDB snapshot) ----------
+    def state_dict(self) -> dict:
+        return {"kind": "sharded_replay",
+                "rings": [r.state_dict() for r in self.rings]}
+
+    def load_state_dict(self, sd: dict) -> None:
+        assert len(sd["rings"]) == len(self.rings), "actor-ring count mismatch"
+        for r, rsd in zip(self.rings, sd["rings"]):
+            r.load_state_dict(rsd)
+
 
 class ActorState:
     """Device-side actor loop state: E env streams with frame-stack buffers."""

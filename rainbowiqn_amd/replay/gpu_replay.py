@@ -144,6 +144,24 @@ class GpuReplay:
         Called outside graph capture (e.g. at target-update cadence)."""
         self.ext.tree_rebuild(self.tree, self.tree_size)
 
+    # -- persistence (reference: Redis RDB snapshot, SURVEY.md 6.4) ----------
+    _PERSIST = ("frames", "actions", "rewards", "terminals", "timesteps",
+                "tree", "ring_state", "max_priority", "beta", "rng_state")
+
+    def state_dict(self) -> dict:
+        sd = {k: getattr(self, k).cpu() for k in self._PERSIST}
+        sd["len"] = self._len_host
+        sd["kind"] = "gpu_replay"
+        return sd
+
+    def load_state_dict(self, sd: dict) -> None:
+        assert sd["frames"].shape[0] == self.capacity, (
+            f"replay checkpoint capacity {sd['frames'].shape[0]} != {self.capacity}"
+        )
+        for k in self._PERSIST:
+            getattr(self, k).copy_(sd[k].to(self.device))
+        self._len_host = int(sd["len"])
+
     # -- introspection for tests --------------------------------------------
     def tree_total(self) -> float:
         return float(self.tree[1].item())

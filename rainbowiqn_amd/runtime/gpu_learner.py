@@ -202,6 +202,18 @@ class GpuLearner:
         os.makedirs(path, exist_ok=True)
         torch.save(self.state_dict(), os.path.join(path, name))
 
+    def load_state_dict_ref(self, sd) -> None:
+        """Load a reference-format fp32 state_dict (checkpoint resume): module
+        weights (cast to the compute dtype), Adam fp32 master, and target."""
+        from ..utils.checkpoint import remap_reference_state_dict
+
+        sd = remap_reference_state_dict(sd, self.online)
+        dt = torch.bfloat16 if self.bf16 else torch.float32
+        self.online.load_state_dict({k: v.to(dt) for k, v in sd.items()})
+        if self.bf16:
+            self.opt.master.copy_(self.flat_online.float())
+        self.update_target()
+
 
 class GraphedActor:
     """hipGraph-captured greedy act step over a batch of E env streams.
