@@ -215,6 +215,73 @@ void stack_gather(torch::Tensor ring, torch::Tensor env_state, int64_t history,
                       (float*)out.data_ptr(), out_bf16 ? 1 : 0, cur_stream());
 }
 
+// ---- fused RainbowIQN forward (model_fwd.hip) ------------------------------
+
+#define CHECK_BF16(t) CHECK_T(t, torch::kBFloat16)
+
+void conv_pack3(torch::Tensor w3, torch::Tensor w3p) {
+    CHECK_BF16(w3);
+    CHECK_BF16(w3p);
+    TORCH_CHECK(w3.numel() == 64 * 64 * 9 && w3p.numel() == 64 * 64 * 16);
+    riqn_conv_pack3(w3.data_ptr(), w3p.data_ptr(), cur_stream());
+}
+
+void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
+                 torch::Tensor w2, torch::Tensor b2, torch::Tensor w3p,
+                 torch::Tensor b3, torch::Tensor feat) {
+    CHECK_BF16(x);
+    CHECK_BF16(w1); CHECK_BF16(w2); CHECK_BF16(w3p);
+    CHECK_BF16(b1); CHECK_BF16(b2); CHECK_BF16(b3);
+    CHECK_BF16(feat);
+    int B = (int)x.size(0);
+    TORCH_CHECK(x.numel() == (int64_t)B * 4 * 84 * 84, "history must be 4");
+    TORCH_CHECK(w1.numel() == 32 * 256 && w2.numel() == 64 * 512);
+    TORCH_CHECK(feat.numel() == (int64_t)B * 3136);
+    riqn_conv_encode(x.data_ptr(), w1.data_ptr(), b1.data_ptr(), w2.data_ptr(),
+                     b2.data_ptr(), w3p.data_ptr(), b3.data_ptr(),
+                     feat.data_ptr(), B, cur_stream());
+}
+
+void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
+               torch::Tensor feat, torch::Tensor h, int64_t nq) {
+    CHECK_T(taus, torch::kFloat32);
+    CHECK_BF16(w); CHECK_BF16(bias); CHECK_BF16(feat); CHECK_BF16(h);
+    int R = (int)taus.numel();
+    TORCH_CHECK(h.numel() == (int64_t)R * 3136);
+    TORCH_CHECK(w.numel() == 3136 * 64);
+    TORCH_CHECK(feat.numel() * nq == h.numel(), "feat rows * nq != R");
+    riqn_tau_embed(taus.data_ptr<float>(), w.data_ptr(), bias.data_ptr(),
+                   feat.data_ptr(), h.data_ptr(), R, (int)nq, cur_stream());
+}
+
+void heads_gemm(torch::Tensor h, torch::Tensor wcat, torch::Tensor bcat,
+                torch::Tensor out) {
+    CHECK_BF16(h); CHECK_BF16(wcat); CHECK_BF16(bcat); CHECK_BF16(out);
+    int R = (int)h.size(0);
+    TORCH_CHECK(h.size(1) == 3136 && wcat.numel() == 1024 * 3136);
+    TORCH_CHECK(out.numel() == (int64_t)R * 1024);
+    riqn_heads_gemm(h.data_ptr(), wcat.data_ptr(), bcat.data_ptr(),
+                    out.data_ptr(), R, cur_stream());
+}
+
+// mode 0: q_out [R,A] f32; mode 1: act_io out [R/nq]; mode 2: q_out [R] f32,
+// act_io in [R/nq]
+void heads_finish(torch::Tensor hh, torch::Tensor wz, torch::Tensor bz,
+                  int64_t A, int64_t nq, int64_t mode, torch::Tensor q_out,
+                  torch::Tensor act_io) {
+    CHECK_BF16(hh); CHECK_BF16(wz); CHECK_BF16(bz);
+    int R = (int)hh.size(0);
+    TORCH_CHECK(hh.size(1) == 1024);
+    TORCH_CHECK(wz.numel() == (1 + A) * 512 && A <= 18);
+    TORCH_CHECK(q_out.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(act_io.scalar_type() == torch::kInt64);
+    if (mode == 1 || mode == 2) TORCH_CHECK(act_io.numel() == R / nq);
+    if (mode == 1) TORCH_CHECK(R % nq == 0);
+    riqn_heads_finish(hh.data_ptr(), wz.data_ptr(), bz.data_ptr(), R, (int)A,
+                      (int)nq, (int)mode, q_out.data_ptr<float>(),
+                      act_io.data_ptr<int64_t>(), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -231,4 +298,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adam_step_mixed", &adam_step_mixed);
     m.def("noisy_material", &noisy_material);
     m.def("stack_gather", &stack_gather);
+    m.def("conv_pack3", &conv_pack3);
+    m.def("conv_encode", &conv_encode);
+    m.def("tau_embed", &tau_embed);
+    m.def("heads_gemm", &heads_gemm);
+    m.def("heads_finish", &heads_finish);
 }

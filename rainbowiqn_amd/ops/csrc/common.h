@@ -74,6 +74,12 @@ DEV float wave_reduce_sum(float v) {
     for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
     return v;
 }
+// butterfly: every lane ends with the full sum
+DEV float wave_allreduce_sum(float v) {
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+    return v;
+}
 DEV float wave_reduce_max(float v) {
     #pragma unroll
     for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off, 64));

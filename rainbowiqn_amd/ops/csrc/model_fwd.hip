@@ -1,0 +1,501 @@
+// Fused RainbowIQN forward for CDNA4 (gfx950) -- SURVEY.md K1-K9 as four
+// hand-written MFMA kernels instead of ~40 library launches:
+//
+//   1. conv_encode : Nature-CNN (4->32 8x8s4, 32->64 4x4s2, 64->64 3x3s1,
+//      ReLU after each) fused into ONE kernel.  One workgroup per state;
+//      the input frame-stack and both intermediate activation planes stay
+//      in LDS (59+30+14 KB of the 160 KB/CU); each conv layer is an
+//      implicit GEMM on v_mfma_f32_16x16x32_bf16 tiles with im2col gathers
+//      straight from LDS.  Weights ride in from L2 (re-read by all B
+//      workgroups).  Replaces MIOpen igemm + NHWC transposes + eager ReLU.
+//   2. tau_embed   : cos(pi*i*tau) basis computed in-register (no [R,64]
+//      cosine tensor in HBM), GEMM against iqn_fc.weight via MFMA, bias +
+//      ReLU + broadcast-multiply with the conv features fused in the
+//      epilogue -> h [R,3136] bf16.
+//   3. heads_gemm  : h [R,3136] x Wcat^T [3136,1024] -> [R,1024] where Wcat
+//      is fc_h_v/fc_h_a's noisy weights CONCATENATED row-wise (the
+//      noisy_material kernel writes them into one buffer, so both dueling
+//      branches are one LDS-tiled MFMA GEMM).  Bias + ReLU fused.
+//   4. heads_finish: z-layer GEMV (Wz = [fc_z_v; fc_z_a] concat, row 0 from
+//      the hv half, rows 1..A from the ha half) + dueling combine
+//      Q = V + A - mean_a(A), with three epilogue modes: quantile values
+//      out (learn), mean-over-taus argmax -> actions (act, K9 fused), or
+//      gather at a given action -> theta (target-net path, K10 feed).
+//
+// Numerics: bf16 operands, fp32 MFMA accumulation (same contract as the
+// hipBLASLt/MIOpen path it replaces); parity tests in
+// tests/test_model_fused_gpu.py compare against the eager module.
+#include "common.h"
+#include "riqn_kernels.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+typedef __attribute__((ext_vector_type(4))) unsigned short u16x4;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+DEV f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+DEV bf16x8 as_bf16x8(u16x8 v) { return __builtin_bit_cast(bf16x8, v); }
+DEV float bf(const unsigned short u) { return bf16_to_f32(u); }
+
+// ---------------------------------------------------------------------------
+// 1. Fused conv encoder.  One WG (256 threads, 4 waves) per state.
+// LDS layout (ushorts):
+//   in   : [4][84][88]   -> 29568 ush (59136 B)
+//   c1out: [32][20][24]  -> 15360 ush (30720 B)
+//   c2out: [64][9][12]   ->  6912 ush (13824 B)
+// ---------------------------------------------------------------------------
+#define IN_STRIDE 88
+#define C1_STRIDE 24
+#define C2_STRIDE 12
+#define LDS_IN 0
+#define LDS_C1 29568
+#define LDS_C2 (29568 + 15360)
+#define LDS_TOTAL (29568 + 15360 + 6912)
+
+__global__ __launch_bounds__(256, 1) void conv_encode_kernel(
+    const unsigned short* __restrict__ x,    // [B,4,84,84] bf16
+    const unsigned short* __restrict__ w1,   // [32,256]  bf16
+    const unsigned short* __restrict__ b1,   // [32]      bf16
+    const unsigned short* __restrict__ w2,   // [64,512]  bf16
+    const unsigned short* __restrict__ b2,   // [64]
+    const unsigned short* __restrict__ w3p,  // [64,1024] bf16 (9->16 padded)
+    const unsigned short* __restrict__ b3,   // [64]
+    unsigned short* __restrict__ feat) {     // [B,3136]  bf16
+    __shared__ unsigned short lds[LDS_TOTAL];
+    const int b = blockIdx.x;
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wave = t >> 6;
+    const int l15 = lane & 15;
+    const int l4 = lane >> 4;  // 0..3
+
+    // ---- stage input state into LDS (row stride 88) ----------------------
+    {
+        const unsigned short* src = x + (int64_t)b * (4 * 84 * 84);
+        for (int e = t * 4; e < 4 * 84 * 84; e += 256 * 4) {
+            u16x4 v = *(const u16x4*)(src + e);
+            int c = e / 7056, r = (e % 7056) / 84, cc = e % 84;
+            *(u16x4*)&lds[LDS_IN + (c * 84 + r) * IN_STRIDE + cc] = v;
+        }
+    }
+    __syncthreads();
+
+    // ---- conv1: M=400 (25 tiles), N=32 (2 tiles), K=256 (8 steps) --------
+    for (int mt = wave; mt < 25; mt += 4) {
+        f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+        const int m = mt * 16 + l15;           // output pixel (A-row)
+        const int oy = m / 20, ox = m % 20;
+        #pragma unroll
+        for (int ks = 0; ks < 8; ++ks) {
+            const int k0 = ks * 32 + l4 * 8;
+            const int c = k0 >> 6, ky = (k0 >> 3) & 7;
+            // A: 8 taps = input row segment [c][oy*4+ky][ox*4 .. +7]
+            u16x8 av;
+            {
+                const int idx = LDS_IN + (c * 84 + oy * 4 + ky) * IN_STRIDE + ox * 4;
+                u16x4 lo = *(const u16x4*)&lds[idx];
+                u16x4 hi = *(const u16x4*)&lds[idx + 4];
+                av = u16x8{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+            }
+            bf16x8 a = as_bf16x8(av);
+            #pragma unroll
+            for (int nt = 0; nt < 2; ++nt) {
+                const int n = nt * 16 + l15;
+                bf16x8 bfr = as_bf16x8(*(const u16x8*)(w1 + n * 256 + k0));
+                acc[nt] = mfma16(a, bfr, acc[nt]);
+            }
+        }
+        // epilogue: D row = mt*16 + l4*4 + j (pixel), col = l15 (channel)
+        #pragma unroll
+        for (int nt = 0; nt < 2; ++nt) {
+            const int n = nt * 16 + l15;
+            const float bias = bf(b1[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int mm = mt * 16 + l4 * 4 + j;
+                if (mm < 400) {
+                    float v = fmaxf(acc[nt][j] + bias, 0.0f);
+                    lds[LDS_C1 + (n * 20 + mm / 20) * C1_STRIDE + mm % 20] =
+                        f32_to_bf16(v);
+                }
+            }
+        }
+    }
+    __syncthreads();
+
+    // ---- conv2: M=81 (6 tiles), N=64 (4 tiles), K=512 (16 steps) ---------
+    for (int mt = wave; mt < 6; mt += 4) {
+        f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
+                        f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+        const int m = mt * 16 + l15;
+        const int oy = m / 9, ox = m % 9;
+        const bool mv = m < 81;
+        #pragma unroll 4
+        for (int ks = 0; ks < 16; ++ks) {
+            const int k0 = ks * 32 + l4 * 8;
+            const int c = k0 >> 4, t16 = k0 & 15;  // t16 in {0,8}
+            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (mv) {
+                // taps (ky0, kx 0..3) and (ky0+1, kx 0..3)
+                const int ky0 = t16 >> 2;
+                const int base =
+                    LDS_C1 + (c * 20 + oy * 2 + ky0) * C1_STRIDE + ox * 2;
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) av[j] = lds[base + j];
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) av[4 + j] = lds[base + C1_STRIDE + j];
+            }
+            bf16x8 a = as_bf16x8(av);
+            #pragma unroll
+            for (int nt = 0; nt < 4; ++nt) {
+                const int n = nt * 16 + l15;
+                bf16x8 bfr = as_bf16x8(*(const u16x8*)(w2 + n * 512 + k0));
+                acc[nt] = mfma16(a, bfr, acc[nt]);
+            }
+        }
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+            const int n = nt * 16 + l15;
+            const float bias = bf(b2[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int mm = mt * 16 + l4 * 4 + j;
+                if (mm < 81) {
+                    float v = fmaxf(acc[nt][j] + bias, 0.0f);
+                    lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] =
+                        f32_to_bf16(v);
+                }
+            }
+        }
+    }
+    __syncthreads();
+
+    // ---- conv3: M=49 (4 tiles), N=64 (4 tiles), K=1024 padded (32 steps) -
+    {
+        const int mt = wave;  // exactly 4 tiles
+        f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
+                        f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+        const int m = mt * 16 + l15;
+        const int oy = m / 7, ox = m % 7;
+        const bool mv = m < 49;
+        #pragma unroll 4
+        for (int ks = 0; ks < 32; ++ks) {
+            const int k0 = ks * 32 + l4 * 8;
+            const int c = k0 >> 4, t16 = k0 & 15;
+            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (mv) {
+                #pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+                    if (ky < 3 && kx < 3)
+                        av[j] = lds[LDS_C2 + (c * 9 + oy + ky) * C2_STRIDE +
+                                    ox + kx];
+                }
+            }
+            bf16x8 a = as_bf16x8(av);
+            #pragma unroll
+            for (int nt = 0; nt < 4; ++nt) {
+                const int n = nt * 16 + l15;
+                bf16x8 bfr = as_bf16x8(*(const u16x8*)(w3p + n * 1024 + k0));
+                acc[nt] = mfma16(a, bfr, acc[nt]);
+            }
+        }
+        unsigned short* out = feat + (int64_t)b * 3136;
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt) {
+            const int n = nt * 16 + l15;
+            const float bias = bf(b3[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int mm = mt * 16 + l4 * 4 + j;
+                if (mm < 49)
+                    out[n * 49 + mm] = f32_to_bf16(fmaxf(acc[nt][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
+// pack conv3 weight [64][64*3*3] -> [64][64*16] (tap grid padded 3x3 -> 4x4)
+__global__ void pack_conv3_kernel(const unsigned short* __restrict__ w3,
+                                  unsigned short* __restrict__ w3p) {
+    int i = threadIdx.x + blockIdx.x * blockDim.x;
+    if (i >= 64 * 1024) return;
+    int n = i >> 10, rem = i & 1023, c = rem >> 4, t16 = rem & 15;
+    int ky = t16 >> 2, kx = t16 & 3;
+    w3p[i] = (ky < 3 && kx < 3) ? w3[n * 576 + c * 9 + ky * 3 + kx]
+                                : (unsigned short)0;
+}
+
+// ---------------------------------------------------------------------------
+// 2. tau embed: h[r][n] = relu(sum_k cos(pi*k*tau_r) * W[n][k] + b[n]) * feat[r/nq][n]
+// Grid (ceil(R/16), NSPLIT=4); 196 n-tiles split across gridDim.y * 4 waves.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void tau_embed_kernel(
+    const float* __restrict__ taus,         // [R]
+    const unsigned short* __restrict__ w,   // [3136,64] bf16 (iqn_fc.weight)
+    const unsigned short* __restrict__ bias,// [3136] bf16
+    const unsigned short* __restrict__ feat,// [B,3136] bf16
+    unsigned short* __restrict__ h,         // [R,3136] bf16
+    int R, int nq) {
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int l15 = lane & 15;
+    const int l4 = lane >> 4;
+    const int mt = blockIdx.x;
+    const int r_a = mt * 16 + l15;               // A-fragment row
+    const float tau = (r_a < R) ? taus[r_a] : 0.0f;
+
+    // A-fragments for both k-steps, computed once (k = l4*8 + j + 32*ks)
+    bf16x8 afrag[2];
+    #pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+        u16x8 av;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float k = (float)(ks * 32 + l4 * 8 + j);
+            av[j] = f32_to_bf16(__cosf(3.14159265358979f * k * tau));
+        }
+        afrag[ks] = as_bf16x8(av);
+    }
+
+    // n-tiles: 196 total; this block owns [blockIdx.y*49, +49), wave-strided
+    const int nt0 = blockIdx.y * 49;
+    for (int nt = nt0 + wave; nt < nt0 + 49; nt += 4) {
+        const int n = nt * 16 + l15;
+        f32x4 acc = {0, 0, 0, 0};
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+            bf16x8 bfr = as_bf16x8(*(const u16x8*)(w + n * 64 + ks * 32 + l4 * 8));
+            acc = mfma16(afrag[ks], bfr, acc);
+        }
+        const float bb = bf(bias[n]);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const int r = mt * 16 + l4 * 4 + j;  // D row
+            if (r < R) {
+                float phi = bf16_to_f32(f32_to_bf16(fmaxf(acc[j] + bb, 0.0f)));
+                float fv = bf(feat[(int64_t)(r / nq) * 3136 + n]);
+                h[(int64_t)r * 3136 + n] = f32_to_bf16(phi * fv);
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// 3. heads GEMM: out[R,1024] = relu(h[R,3136] @ Wcat^T + bcat), bf16.
+// Block tile 64x64, BK=32 (one MFMA K), 4 waves each on a 32x32 quadrant.
+// ---------------------------------------------------------------------------
+#define HG_BM 64
+#define HG_BN 64
+#define HG_BK 32
+__global__ __launch_bounds__(256) void heads_gemm_kernel(
+    const unsigned short* __restrict__ hin,  // [R,3136]
+    const unsigned short* __restrict__ wcat, // [1024,3136]
+    const unsigned short* __restrict__ bcat, // [1024]
+    unsigned short* __restrict__ out,        // [R,1024]
+    int R) {
+    __shared__ unsigned short smem[2 * HG_BM * HG_BK];  // As | Bs
+    unsigned short* As = smem;                           // [64][32]
+    unsigned short* Bs = smem + HG_BM * HG_BK;           // [64][32]
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;
+    const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid of 32x32
+    const int l15 = lane & 15, l4 = lane >> 4;
+    const int bm = blockIdx.x * HG_BM, bn = blockIdx.y * HG_BN;
+
+    f32x4 acc[2][2] = {{{0, 0, 0, 0}, {0, 0, 0, 0}},
+                       {{0, 0, 0, 0}, {0, 0, 0, 0}}};
+    const int ldrow = t >> 2, ldk = (t & 3) * 8;  // 256 thr: 64 rows x 4 pieces
+
+    for (int k0 = 0; k0 < 3136; k0 += HG_BK) {
+        // stage A and B tiles (each thread one 16B piece per tile)
+        {
+            const int gr = bm + ldrow;
+            u16x8 va = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (gr < R) va = *(const u16x8*)(hin + (int64_t)gr * 3136 + k0 + ldk);
+            *(u16x8*)&As[ldrow * HG_BK + ldk] = va;
+            u16x8 vb = *(const u16x8*)(wcat + (int64_t)(bn + ldrow) * 3136 + k0 + ldk);
+            *(u16x8*)&Bs[ldrow * HG_BK + ldk] = vb;
+        }
+        __syncthreads();
+        #pragma unroll
+        for (int fm = 0; fm < 2; ++fm) {
+            bf16x8 a = as_bf16x8(
+                *(const u16x8*)&As[(wr * 32 + fm * 16 + l15) * HG_BK + l4 * 8]);
+            #pragma unroll
+            for (int fn = 0; fn < 2; ++fn) {
+                bf16x8 b = as_bf16x8(
+                    *(const u16x8*)&Bs[(wc * 32 + fn * 16 + l15) * HG_BK + l4 * 8]);
+                acc[fm][fn] = mfma16(a, b, acc[fm][fn]);
+            }
+        }
+        __syncthreads();
+    }
+    #pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+            const int n = bn + wc * 32 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * 32 + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// 4. finish: z-layer + dueling.  hh[r] = [hv(512) | ha(512)];
+//   v = Wz[0] . hv + bz[0];  a_i = Wz[1+i] . ha + bz[1+i]
+//   q[r][i] = v + a_i - mean(a)
+// mode 0: write q [R,A] f32
+// mode 1: actions[e] = argmax_i mean_{k<nq} q[e*nq+k][i]   (one WG per env)
+// mode 2: theta[r] = q[r][act[r/nq]]                        (f32 [R])
+// ---------------------------------------------------------------------------
+template <int MODE>
+__global__ __launch_bounds__(256) void heads_finish_kernel(
+    const unsigned short* __restrict__ hh,  // [R,1024]
+    const unsigned short* __restrict__ wz,  // [1+A,512]
+    const unsigned short* __restrict__ bz,  // [1+A]
+    int R, int A, int nq,
+    float* __restrict__ q_out,              // mode 0: [R,A]; mode 2: [R]
+    int64_t* __restrict__ act_io) {         // mode 1: out [R/nq]; mode 2: in [R/nq]
+    __shared__ float qsum[32];              // mode 1 scratch: sum over rows
+    __shared__ unsigned short wzs[19 * 512];
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;
+    // stage Wz (<= 19*512 bf16 = 19 KB)
+    for (int i = t * 4; i < (1 + A) * 512; i += 256 * 4)
+        *(u16x4*)&wzs[i] = *(const u16x4*)(wz + i);
+    if (MODE == 1)
+        for (int i = t; i < 32; i += 256) qsum[i] = 0.0f;
+    __syncthreads();
+
+    const int rows_per_wg = (MODE == 1) ? nq : 64;
+    const int r0 = blockIdx.x * rows_per_wg;
+    for (int rr = wave; rr < rows_per_wg; rr += 4) {
+        const int r = r0 + rr;
+        if (r >= R) break;
+        const unsigned short* hrow = hh + (int64_t)r * 1024;
+        // lane-parallel dot products: lane covers elems lane*8..+8
+        float hv[8], ha[8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            hv[j] = bf(hrow[lane * 8 + j]);
+            ha[j] = bf(hrow[512 + lane * 8 + j]);
+        }
+        float v = 0.0f;
+        {
+            float p = 0.0f;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+                p += hv[j] * bf(wzs[lane * 8 + j]);
+            v = wave_allreduce_sum(p) + bf(bz[0]);
+        }
+        float amean = 0.0f;
+        float aval[18];
+        for (int i = 0; i < A; ++i) {
+            float p = 0.0f;
+            const unsigned short* wrow = &wzs[(1 + i) * 512];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+                p += ha[j] * bf(wrow[lane * 8 + j]);
+            float s = wave_allreduce_sum(p) + bf(bz[1 + i]);
+            aval[i] = s;
+            amean += s;
+        }
+        amean /= (float)A;
+        if (MODE == 0) {
+            if (lane < A) q_out[(int64_t)r * A + lane] = v + aval[lane] - amean;
+        } else if (MODE == 1) {
+            if (lane < A)
+                atomicAdd(&qsum[lane], v + aval[lane] - amean);
+        } else {  // MODE 2
+            if (lane == 0) {
+                int a = (int)act_io[r / nq];
+                q_out[r] = v + aval[a] - amean;
+            }
+        }
+    }
+    if (MODE == 1) {
+        __syncthreads();
+        if (t == 0) {
+            int best = 0;
+            float bv = qsum[0];
+            for (int i = 1; i < A; ++i)
+                if (qsum[i] > bv) { bv = qsum[i]; best = i; }
+            act_io[blockIdx.x] = (int64_t)best;
+        }
+    }
+}
+
+}  // namespace
+
+extern "C" {
+
+void riqn_conv_pack3(const void* w3, void* w3p, void* stream) {
+    hipLaunchKernelGGL(pack_conv3_kernel, dim3(64), dim3(1024), 0,
+                       (hipStream_t)stream, (const unsigned short*)w3,
+                       (unsigned short*)w3p);
+}
+
+void riqn_conv_encode(const void* x, const void* w1, const void* b1,
+                      const void* w2, const void* b2, const void* w3p,
+                      const void* b3, void* feat, int B, void* stream) {
+    hipLaunchKernelGGL(conv_encode_kernel, dim3(B), dim3(256), 0,
+                       (hipStream_t)stream, (const unsigned short*)x,
+                       (const unsigned short*)w1, (const unsigned short*)b1,
+                       (const unsigned short*)w2, (const unsigned short*)b2,
+                       (const unsigned short*)w3p, (const unsigned short*)b3,
+                       (unsigned short*)feat);
+}
+
+void riqn_tau_embed(const float* taus, const void* w, const void* bias,
+                    const void* feat, void* h, int R, int nq, void* stream) {
+    dim3 grid((R + 15) / 16, 4);
+    hipLaunchKernelGGL(tau_embed_kernel, grid, dim3(256), 0,
+                       (hipStream_t)stream, taus, (const unsigned short*)w,
+                       (const unsigned short*)bias,
+                       (const unsigned short*)feat, (unsigned short*)h, R, nq);
+}
+
+void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
+                     void* out, int R, void* stream) {
+    dim3 grid((R + HG_BM - 1) / HG_BM, 1024 / HG_BN);
+    hipLaunchKernelGGL(heads_gemm_kernel, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const unsigned short*)h,
+                       (const unsigned short*)wcat,
+                       (const unsigned short*)bcat, (unsigned short*)out, R);
+}
+
+void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
+                       int A, int nq, int mode, float* q_out, int64_t* act_io,
+                       void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    if (mode == 1) {
+        hipLaunchKernelGGL(heads_finish_kernel<1>, dim3(R / nq), dim3(256), 0,
+                           s, (const unsigned short*)hh,
+                           (const unsigned short*)wz,
+                           (const unsigned short*)bz, R, A, nq, q_out, act_io);
+    } else if (mode == 2) {
+        hipLaunchKernelGGL(heads_finish_kernel<2>, dim3((R + 63) / 64),
+                           dim3(256), 0, s, (const unsigned short*)hh,
+                           (const unsigned short*)wz,
+                           (const unsigned short*)bz, R, A, nq, q_out, act_io);
+    } else {
+        hipLaunchKernelGGL(heads_finish_kernel<0>, dim3((R + 63) / 64),
+                           dim3(256), 0, s, (const unsigned short*)hh,
+                           (const unsigned short*)wz,
+                           (const unsigned short*)bz, R, A, nq, q_out, act_io);
+    }
+}
+
+}  // extern "C"

@@ -82,6 +82,20 @@ void riqn_noisy_material(float* eps_in, int64_t n_in, float* eps_out,
                          void* b_noisy, int bf16, uint64_t* rng_state,
                          void* stream);
 
+// ---- fused RainbowIQN forward (model_fwd.hip) -------------
+// All tensors bf16 unless noted.  Layouts documented in model_fwd.hip.
+void riqn_conv_pack3(const void* w3, void* w3p, void* stream);
+void riqn_conv_encode(const void* x, const void* w1, const void* b1,
+                      const void* w2, const void* b2, const void* w3p,
+                      const void* b3, void* feat, int B, void* stream);
+void riqn_tau_embed(const float* taus, const void* w, const void* bias,
+                    const void* feat, void* h, int R, int nq, void* stream);
+void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
+                     void* out, int R, void* stream);
+void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
+                       int A, int nq, int mode, float* q_out, int64_t* act_io,
+                       void* stream);
+
 // ---- device synthetic env (bench frame source; env.hip) --------------------
 void riqn_env_step(const int64_t* actions, int32_t* env_state,
                    uint64_t* rng_state, uint8_t* ring, int history,

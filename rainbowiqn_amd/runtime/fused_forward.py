@@ -1,0 +1,133 @@
+"""Fused inference forward for RainbowIQN (MI355X fast path).
+
+Binds a bf16 RainbowIQN module to the four hand-written gfx950 kernels in
+ops/csrc/model_fwd.hip (fused 3-conv encoder, cos-tau-embed GEMM, concat
+noisy-heads GEMM, dueling finish).  One forward = 4 kernel launches instead
+of the ~40 eager library launches (MIOpen igemm + NHWC transposes +
+hipBLASLt GEMMs + elementwise ReLU/add/mean chains) it replaces.
+
+Weight plumbing (zero repack per step):
+  * fc_h_v/fc_h_a ``weight_noisy``/``bias_noisy`` buffers are re-pointed to
+    row-slices of ONE concatenated [1024, 3136] buffer, and fc_z_v/fc_z_a
+    to slices of a [1+A, 512] buffer, so the per-round noisy_material
+    kernels write straight into the GEMM operands;
+  * conv1/conv2/iqn weights are read in their natural (flat-param) layout;
+  * conv3's 3x3 taps are padded to a 4x4 grid ([64,1024] bf16) by a tiny
+    pack kernel -- the only repack, re-run in-graph when weights change.
+
+All methods are hipGraph-capturable: fixed shapes, static scratch buffers
+(cached per R), device RNG taus supplied by the caller.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Tuple
+
+import torch
+
+from ..models.model import RainbowIQN
+from ..ops import get_extension
+
+
+class FusedForward:
+    def __init__(self, net: RainbowIQN, device):
+        self.ext = get_extension()
+        self.net = net
+        self.device = torch.device(device)
+        p = net.fc_h_v.weight_mu
+        assert p.dtype == torch.bfloat16, "fused forward is the bf16 path"
+        self.A = net.action_space
+        d = self.device
+        bf = torch.bfloat16
+
+        # concat noisy-weight storage; layer buffers become views into it
+        self.wcat = torch.empty(1024, 3136, dtype=bf, device=d)
+        self.bcat = torch.empty(1024, dtype=bf, device=d)
+        self.wz = torch.empty(1 + self.A, 512, dtype=bf, device=d)
+        self.bz = torch.empty(1 + self.A, dtype=bf, device=d)
+        slices = [
+            (net.fc_h_v, self.wcat[0:512], self.bcat[0:512]),
+            (net.fc_h_a, self.wcat[512:1024], self.bcat[512:1024]),
+            (net.fc_z_v, self.wz[0:1], self.bz[0:1]),
+            (net.fc_z_a, self.wz[1:], self.bz[1:]),
+        ]
+        for layer, wslice, bslice in slices:
+            if getattr(layer, "weight_noisy", None) is None:
+                layer.enable_fused_noise()
+            with torch.no_grad():
+                wslice.copy_(layer.weight_noisy.view(wslice.shape))
+                bslice.copy_(layer.bias_noisy.view(bslice.shape))
+            layer.weight_noisy = wslice.view(layer.weight_noisy.shape)
+            layer.bias_noisy = bslice.view(layer.bias_noisy.shape)
+
+        # conv weights in natural layout (views into the flat param buffer)
+        self.w1 = net.convs[0].weight.detach().view(32, 256)
+        self.b1 = net.convs[0].bias.detach()
+        self.w2 = net.convs[2].weight.detach().view(64, 512)
+        self.b2 = net.convs[2].bias.detach()
+        self.w3 = net.convs[4].weight.detach().view(64, 576)
+        self.b3 = net.convs[4].bias.detach()
+        self.w3p = torch.empty(64, 1024, dtype=bf, device=d)
+        self.wiqn = net.iqn_fc.weight.detach()  # [3136, 64]
+        self.biqn = net.iqn_fc.bias.detach()
+        self.pack_weights()
+
+        self._scratch: Dict[Tuple[int, int], tuple] = {}
+
+    def pack_weights(self) -> None:
+        """Refresh the conv3 padded layout (call after conv weights change;
+        graph-capturable)."""
+        self.ext.conv_pack3(self.w3.contiguous(), self.w3p)
+
+    def _bufs(self, B: int, nq: int):
+        key = (B, nq)
+        if key not in self._scratch:
+            d, bf = self.device, torch.bfloat16
+            R = B * nq
+            self._scratch[key] = (
+                torch.empty(B, 3136, dtype=bf, device=d),     # feat
+                torch.empty(R, 3136, dtype=bf, device=d),     # h
+                torch.empty(R, 1024, dtype=bf, device=d),     # hh
+            )
+        return self._scratch[key]
+
+    def encode(self, x: torch.Tensor) -> torch.Tensor:
+        """Conv trunk only: x [B,4,84,84] bf16 -> features [B,3136] bf16."""
+        feat, _, _ = self._bufs(x.shape[0], 1)
+        self.ext.conv_encode(x, self.w1, self.b1, self.w2, self.b2, self.w3p,
+                             self.b3, feat)
+        return feat
+
+    def _trunk(self, x: torch.Tensor, taus: torch.Tensor, nq: int):
+        """x [B,4,84,84] bf16, taus [B*nq] f32 -> hh [B*nq, 1024]."""
+        B = x.shape[0]
+        feat, h, hh = self._bufs(B, nq)
+        self.ext.conv_encode(x, self.w1, self.b1, self.w2, self.b2, self.w3p,
+                             self.b3, feat)
+        self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h, nq)
+        self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
+        return hh
+
+    def quantiles(self, x, taus, nq: int, out: torch.Tensor) -> torch.Tensor:
+        """Full forward -> quantile values out [B, nq, A] (f32)."""
+        hh = self._trunk(x, taus, nq)
+        dummy = torch.zeros(1, dtype=torch.int64, device=self.device)
+        self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 0,
+                              out.view(-1, self.A), dummy)
+        return out
+
+    def act(self, x, taus, nq: int, actions: torch.Tensor) -> torch.Tensor:
+        """argmax_a mean_k Z_tau_k(x, a) -> actions [B] (K9 fused)."""
+        hh = self._trunk(x, taus, nq)
+        dummy = torch.zeros(1, dtype=torch.float32, device=self.device)
+        self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 1, dummy,
+                              actions)
+        return actions
+
+    def theta(self, x, taus, nq: int, actions: torch.Tensor,
+              out: torch.Tensor) -> torch.Tensor:
+        """Z_tau(x, actions[b]) -> out [B, nq] (target-net path, K10 feed)."""
+        hh = self._trunk(x, taus, nq)
+        self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 2,
+                              out.view(-1), actions)
+        return out

@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "adam.hip"),
         os.path.join(CSRC, "rng.hip"),
         os.path.join(CSRC, "env.hip"),
+        os.path.join(CSRC, "model_fwd.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -1,0 +1,120 @@
+"""Parity tests: fused gfx950 forward kernels (ops/csrc/model_fwd.hip via
+runtime/fused_forward.FusedForward) against the eager RainbowIQN module.
+
+Reference = the fp32 eager module with IDENTICAL weights/noise; the bf16
+eager module brackets the tolerance (fused and eager-bf16 should both sit
+within bf16 rounding of the fp32 truth).
+"""
+
+import pytest
+import torch
+
+from rainbowiqn_amd.config import Config
+
+pytestmark = pytest.mark.gpu
+
+
+def make_nets(A=6, seed=0):
+    from rainbowiqn_amd.models.model import RainbowIQN
+
+    torch.manual_seed(seed)
+    cfg = Config(dtype="bf16", device="cuda")
+    net32 = RainbowIQN(cfg, A).cuda().train()
+    net16 = RainbowIQN(cfg, A).cuda().train()
+    net16.load_state_dict(net32.state_dict())
+    net16 = net16.to(torch.bfloat16)
+    for layer in (net16.fc_h_v, net16.fc_h_a, net16.fc_z_v, net16.fc_z_a):
+        layer.enable_fused_noise()
+    # identical noise in all three: copy eps from net32, refresh noisy bufs
+    for l32, l16 in [(net32.fc_h_v, net16.fc_h_v), (net32.fc_h_a, net16.fc_h_a),
+                     (net32.fc_z_v, net16.fc_z_v), (net32.fc_z_a, net16.fc_z_a)]:
+        l16.weight_epsilon.copy_(l32.weight_epsilon.to(torch.bfloat16))
+        l16.bias_epsilon.copy_(l32.bias_epsilon.to(torch.bfloat16))
+        with torch.no_grad():
+            l16.weight_noisy.copy_(l16.weight_mu + l16.weight_sigma * l16.weight_epsilon)
+            l16.bias_noisy.copy_(l16.bias_mu + l16.bias_sigma * l16.bias_epsilon)
+    return cfg, net32, net16
+
+
+def fused_of(net16):
+    from rainbowiqn_amd.runtime.fused_forward import FusedForward
+
+    fw = FusedForward(net16, "cuda")
+    # FusedForward re-points weight_noisy to concat storage, preserving values
+    return fw
+
+
+def test_conv_encode_parity():
+    _, net32, net16 = make_nets()
+    fw = fused_of(net16)
+    x = torch.rand(32, 4, 84, 84, device="cuda")
+    ref = net32.encode(x)  # fp32 truth
+    got = fw.encode(x.to(torch.bfloat16)).float()
+    torch.cuda.synchronize()
+    err = (got - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.03 * max(scale, 1.0), f"conv encode err {err} scale {scale}"
+    # bracket: eager bf16 should be comparably far from fp32
+    eag16 = net16.encode(x.to(torch.bfloat16)).float()
+    err16 = (eag16 - ref).abs().max().item()
+    assert err < max(3 * err16, 0.02 * max(scale, 1.0)), (err, err16)
+
+
+def test_full_forward_parity():
+    A = 6
+    _, net32, net16 = make_nets(A)
+    fw = fused_of(net16)
+    B, nq = 32, 8
+    x = torch.rand(B, 4, 84, 84, device="cuda")
+    taus = torch.rand(B * nq, 1, device="cuda")
+    z_ref, _ = net32(x, nq, taus=taus)  # [B, nq, A] fp32 truth
+    z16, _ = net16(x.to(torch.bfloat16), nq, taus=taus)
+    out = torch.empty(B, nq, A, device="cuda")
+    fw.quantiles(x.to(torch.bfloat16), taus.view(-1), nq, out)
+    torch.cuda.synchronize()
+    err = (out - z_ref).abs().max().item()
+    err16 = (z16.float() - z_ref).abs().max().item()
+    scale = z_ref.abs().max().item()
+    assert torch.isfinite(out).all()
+    assert err < 0.05 * max(scale, 1.0) + 0.02, (err, scale)
+    assert err < max(3 * err16 + 0.01, 0.03), (err, err16)
+
+
+def test_act_and_theta_modes_consistent():
+    A = 6
+    _, _, net16 = make_nets(A, seed=3)
+    fw = fused_of(net16)
+    E, K = 64, 32
+    x = (torch.rand(E, 4, 84, 84, device="cuda")).to(torch.bfloat16)
+    taus = torch.rand(E * K, device="cuda")
+    q = torch.empty(E, K, A, device="cuda")
+    fw.quantiles(x, taus, K, q)
+    actions = torch.zeros(E, dtype=torch.int64, device="cuda")
+    fw.act(x, taus, K, actions)
+    torch.cuda.synchronize()
+    expect = q.mean(dim=1).argmax(dim=1)
+    assert (actions == expect).all(), (actions[:8], expect[:8])
+    # mode 2 gather equals mode 0 at the same actions
+    th = torch.empty(E, K, device="cuda")
+    fw.theta(x, taus, K, actions, th)
+    torch.cuda.synchronize()
+    gathered = q.gather(2, actions.view(E, 1, 1).expand(E, K, 1)).squeeze(2)
+    assert torch.allclose(th, gathered, atol=1e-4), (th[0, :4], gathered[0, :4])
+
+
+def test_fused_matches_eager_bf16_closely():
+    """Same weights, same noise, same taus: fused vs eager-bf16 should agree
+    to bf16 accumulation noise (both fp32-accumulate)."""
+    A = 9
+    _, _, net16 = make_nets(A, seed=11)
+    fw = fused_of(net16)
+    B, nq = 16, 8
+    x = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
+    taus = torch.rand(B * nq, 1, device="cuda")
+    z16, _ = net16(x, nq, taus=taus)
+    out = torch.empty(B, nq, A, device="cuda")
+    fw.quantiles(x, taus.view(-1), nq, out)
+    torch.cuda.synchronize()
+    err = (out - z16.float()).abs().max().item()
+    scale = z16.float().abs().max().item()
+    assert err < 0.04 * max(scale, 1.0) + 0.02, (err, scale)
