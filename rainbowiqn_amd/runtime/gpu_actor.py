@@ -58,10 +58,18 @@ class GpuActorLoop:
         self._newest = 0  # ring slot of the most recent frame (round-invariant)
         self._eps_in = torch.empty(4096, device=d)
         self._eps_out = torch.empty(4096, device=d)
+        # hand-written fused forward (4 kernels/step) on the bf16 path
+        self.fused = None
+        if self.dtype == torch.bfloat16 and getattr(cfg, "fused_forward", True):
+            from .fused_forward import FusedForward
+
+            self.fused = FusedForward(net, d)
 
     def _round_body(self) -> None:
         cfg = self.cfg
         self.reset_noise()  # in-graph: device philox, fixed launch sequence
+        if self.fused is not None:
+            self.fused.pack_weights()
         newest = self._newest
         for s in range(self.chunk_steps):
             self.ext.stack_gather(
@@ -69,9 +77,13 @@ class GpuActorLoop:
                 newest, self.E, self.stack_f32,
             )
             self.ext.rand_uniform(self.taus, self.rng_state)
-            with torch.no_grad():
-                z, _ = self.net(self.stack_f32, self.K, taus=self.taus)
-                self.actions.copy_(z.float().mean(dim=1).argmax(dim=1))
+            if self.fused is not None:
+                self.fused.act(self.stack_f32, self.taus.view(-1), self.K,
+                               self.actions)
+            else:
+                with torch.no_grad():
+                    z, _ = self.net(self.stack_f32, self.K, taus=self.taus)
+                    self.actions.copy_(z.float().mean(dim=1).argmax(dim=1))
             newest = (newest + 1) % self.h
             self.ext.env_step(
                 self.actions, self.env_state.view(-1), self.rng_state,

@@ -86,6 +86,16 @@ class GpuLearner:
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.steps_per_replay = 1
         self.grad_steps = 0
+        # fused no-grad forwards (double-Q select + target quantiles): 4
+        # hand-written kernels each instead of the eager module stack
+        self.fused_online = self.fused_target = None
+        if self.bf16 and getattr(cfg, "fused_forward", True):
+            from .fused_forward import FusedForward
+
+            self.fused_online = FusedForward(self.online, self.device)
+            self.fused_target = FusedForward(self.target, self.device)
+            self.a_star = torch.zeros(b, dtype=torch.int64, device=self.device)
+            self.theta_p_buf = torch.zeros(b, self.Np, device=self.device)
 
     # -- pieces -------------------------------------------------------------
     def _reset_noise_device(self, net: RainbowIQN) -> None:
@@ -110,13 +120,22 @@ class GpuLearner:
         idx, s, a, r, s2, nt, w = self.replay.sample()
         z, _ = self.online(s, self.N, taus=self.taus_online)
         theta = z.gather(2, a.view(b, 1, 1).expand(b, self.N, 1)).squeeze(2)
-        with torch.no_grad():
-            zsel, _ = self.online(s2, self.K, taus=self.taus_select)
-            a_star = zsel.mean(dim=1).argmax(dim=1)
-            zp, _ = self.target(s2, self.Np, taus=self.taus_target)
-            theta_p = zp.gather(
-                2, a_star.view(b, 1, 1).expand(b, self.Np, 1)
-            ).squeeze(2)
+        if self.fused_online is not None:
+            self.fused_online.pack_weights()
+            self.fused_target.pack_weights()
+            self.fused_online.act(s2, self.taus_select.view(-1), self.K,
+                                  self.a_star)
+            self.fused_target.theta(s2, self.taus_target.view(-1), self.Np,
+                                    self.a_star, self.theta_p_buf)
+            theta_p = self.theta_p_buf
+        else:
+            with torch.no_grad():
+                zsel, _ = self.online(s2, self.K, taus=self.taus_select)
+                a_star = zsel.mean(dim=1).argmax(dim=1)
+                zp, _ = self.target(s2, self.Np, taus=self.taus_target)
+                theta_p = zp.gather(
+                    2, a_star.view(b, 1, 1).expand(b, self.Np, 1)
+                ).squeeze(2)
         loss, per_sample = quantile_huber_loss_fused(
             theta.float(), theta_p.float(),
             self.taus_online.view(b, self.N).float(), r, nt, w,
