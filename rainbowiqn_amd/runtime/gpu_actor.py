@@ -60,7 +60,9 @@ class GpuActorLoop:
         self._eps_out = torch.empty(4096, device=d)
         # hand-written fused forward (4 kernels/step) on the bf16 path
         self.fused = None
-        if self.dtype == torch.bfloat16 and getattr(cfg, "fused_forward", True):
+        if (self.dtype == torch.bfloat16
+                and net.fc_h_v.weight_mu.dtype == torch.bfloat16
+                and getattr(cfg, "fused_forward", True)):
             from .fused_forward import FusedForward
 
             self.fused = FusedForward(net, d)
