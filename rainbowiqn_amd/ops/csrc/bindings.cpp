@@ -215,6 +215,41 @@ void stack_gather(torch::Tensor ring, torch::Tensor env_state, int64_t history,
                       (float*)out.data_ptr(), out_bf16 ? 1 : 0, cur_stream());
 }
 
+// Batched noise regeneration for the 4 NoisyLinear layers of one net.
+// `ts` = 4 groups of 10 tensors: w_mu, w_sigma, b_mu, b_sigma, w_eps, b_eps,
+// w_noisy, b_noisy, eps_in, eps_out.
+void noisy_material4(std::vector<torch::Tensor> ts, torch::Tensor rng_state) {
+    TORCH_CHECK(ts.size() == 40, "need 4 layers x 10 tensors");
+    NoisyPack4 pack;
+    bool bf16 = ts[0].scalar_type() == torch::kBFloat16;
+    for (int L = 0; L < 4; ++L) {
+        auto* g = &ts[L * 10];
+        for (int j = 0; j < 10; ++j) {
+            CHECK_DEV(g[j]);
+            CHECK_CONT(g[j]);
+        }
+        TORCH_CHECK(g[8].scalar_type() == torch::kFloat32);  // eps_in
+        TORCH_CHECK(g[9].scalar_type() == torch::kFloat32);  // eps_out
+        NoisyDesc& d = pack.l[L];
+        d.w_mu = g[0].data_ptr();
+        d.w_sigma = g[1].data_ptr();
+        d.b_mu = g[2].data_ptr();
+        d.b_sigma = g[3].data_ptr();
+        d.w_eps = g[4].data_ptr();
+        d.b_eps = g[5].data_ptr();
+        d.w_noisy = g[6].data_ptr();
+        d.b_noisy = g[7].data_ptr();
+        d.eps_in = g[8].data_ptr<float>();
+        d.eps_out = g[9].data_ptr<float>();
+        d.n_in = g[0].size(-1);
+        d.n_out = g[0].numel() / d.n_in;
+        TORCH_CHECK(g[8].numel() >= d.n_in && g[9].numel() >= d.n_out);
+    }
+    riqn_noisy_material4(pack, bf16 ? 1 : 0,
+                         (uint64_t*)rng_state.data_ptr<int64_t>(),
+                         cur_stream());
+}
+
 // ---- fused RainbowIQN forward (model_fwd.hip) ------------------------------
 
 #define CHECK_BF16(t) CHECK_T(t, torch::kBFloat16)
@@ -297,6 +332,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("env_step", &env_step);
     m.def("adam_step_mixed", &adam_step_mixed);
     m.def("noisy_material", &noisy_material);
+    m.def("noisy_material4", &noisy_material4);
     m.def("stack_gather", &stack_gather);
     m.def("conv_pack3", &conv_pack3);
     m.def("conv_encode", &conv_encode);

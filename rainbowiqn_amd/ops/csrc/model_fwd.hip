@@ -360,6 +360,9 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
 // mode 1: actions[e] = argmax_i mean_{k<nq} q[e*nq+k][i]   (one WG per env)
 // mode 2: theta[r] = q[r][act[r/nq]]                        (f32 [R])
 // ---------------------------------------------------------------------------
+// MODE 0: q [R,A] f32 out.  MODE 2: theta[r] = q[r][act[r/nq]] out.
+// Two-pass per row: advantage sums stashed in LDS (sv) -- no dynamically
+// indexed register array (which spills to scratch).
 template <int MODE>
 __global__ __launch_bounds__(256) void heads_finish_kernel(
     const unsigned short* __restrict__ hh,  // [R,1024]
@@ -367,73 +370,102 @@ __global__ __launch_bounds__(256) void heads_finish_kernel(
     const unsigned short* __restrict__ bz,  // [1+A]
     int R, int A, int nq,
     float* __restrict__ q_out,              // mode 0: [R,A]; mode 2: [R]
-    int64_t* __restrict__ act_io) {         // mode 1: out [R/nq]; mode 2: in [R/nq]
-    __shared__ float qsum[32];              // mode 1 scratch: sum over rows
+    const int64_t* __restrict__ act_in) {   // mode 2: [R/nq]
     __shared__ unsigned short wzs[19 * 512];
+    __shared__ float sv[4][20];
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
-    // stage Wz (<= 19*512 bf16 = 19 KB)
     for (int i = t * 4; i < (1 + A) * 512; i += 256 * 4)
         *(u16x4*)&wzs[i] = *(const u16x4*)(wz + i);
-    if (MODE == 1)
-        for (int i = t; i < 32; i += 256) qsum[i] = 0.0f;
     __syncthreads();
 
-    const int rows_per_wg = (MODE == 1) ? nq : 64;
-    const int r0 = blockIdx.x * rows_per_wg;
-    for (int rr = wave; rr < rows_per_wg; rr += 4) {
+    const int r0 = blockIdx.x * 16;
+    for (int rr = wave; rr < 16; rr += 4) {
         const int r = r0 + rr;
         if (r >= R) break;
         const unsigned short* hrow = hh + (int64_t)r * 1024;
-        // lane-parallel dot products: lane covers elems lane*8..+8
         float hv[8], ha[8];
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
             hv[j] = bf(hrow[lane * 8 + j]);
             ha[j] = bf(hrow[512 + lane * 8 + j]);
         }
-        float v = 0.0f;
-        {
-            float p = 0.0f;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j)
-                p += hv[j] * bf(wzs[lane * 8 + j]);
-            v = wave_allreduce_sum(p) + bf(bz[0]);
-        }
+        float p = 0.0f;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) p += hv[j] * bf(wzs[lane * 8 + j]);
+        const float v = wave_allreduce_sum(p) + bf(bz[0]);
         float amean = 0.0f;
-        float aval[18];
         for (int i = 0; i < A; ++i) {
-            float p = 0.0f;
+            float q = 0.0f;
             const unsigned short* wrow = &wzs[(1 + i) * 512];
             #pragma unroll
-            for (int j = 0; j < 8; ++j)
-                p += ha[j] * bf(wrow[lane * 8 + j]);
-            float s = wave_allreduce_sum(p) + bf(bz[1 + i]);
-            aval[i] = s;
+            for (int j = 0; j < 8; ++j) q += ha[j] * bf(wrow[lane * 8 + j]);
+            float s = wave_allreduce_sum(q) + bf(bz[1 + i]);
+            if (lane == 0) sv[wave][i] = s;
             amean += s;
         }
         amean /= (float)A;
         if (MODE == 0) {
-            if (lane < A) q_out[(int64_t)r * A + lane] = v + aval[lane] - amean;
-        } else if (MODE == 1) {
-            if (lane < A)
-                atomicAdd(&qsum[lane], v + aval[lane] - amean);
+            if (lane < A) q_out[(int64_t)r * A + lane] = v + sv[wave][lane] - amean;
         } else {  // MODE 2
             if (lane == 0) {
-                int a = (int)act_io[r / nq];
-                q_out[r] = v + aval[a] - amean;
+                int a = (int)act_in[r / nq];
+                q_out[r] = v + sv[wave][a] - amean;
             }
         }
     }
-    if (MODE == 1) {
-        __syncthreads();
-        if (t == 0) {
-            int best = 0;
-            float bv = qsum[0];
-            for (int i = 1; i < A; ++i)
-                if (qsum[i] > bv) { bv = qsum[i]; best = i; }
-            act_io[blockIdx.x] = (int64_t)best;
-        }
+}
+
+// MODE 1 (act, K9): one WG per env.  argmax_a mean_k Q(s, a; tau_k) ==
+// argmax_a [ (sum_k ha_k) . Wz_a + nq * bz_a ] -- the V and mean_a terms are
+// constant across a, so the mean over taus folds into ONE summed-ha vector
+// per env (nq x fewer dot products than the per-row form).
+__global__ __launch_bounds__(256) void heads_act_kernel(
+    const unsigned short* __restrict__ hh,  // [R,1024], R = E*nq
+    const unsigned short* __restrict__ wz,  // [1+A,512]
+    const unsigned short* __restrict__ bz,  // [1+A]
+    int A, int nq,
+    int64_t* __restrict__ actions) {        // [E]
+    __shared__ float habar[4][512];         // per-wave partial -> reduced
+    __shared__ float qbar[20];
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;
+    const int64_t base = (int64_t)blockIdx.x * nq * 1024;
+
+    // per-wave partial sums over this env's nq rows (lane owns 8 columns)
+    float part[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int k = wave; k < nq; k += 4) {
+        const unsigned short* harow = hh + base + (int64_t)k * 1024 + 512;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) part[j] += bf(harow[lane * 8 + j]);
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) habar[wave][lane * 8 + j] = part[j];
+    __syncthreads();
+    // reduce 4 wave partials (thread t owns 2 columns)
+    for (int c = t * 2; c < 512; c += 512) {
+        #pragma unroll
+        for (int u = 0; u < 2; ++u)
+            habar[0][c + u] += habar[1][c + u] + habar[2][c + u] + habar[3][c + u];
+    }
+    __syncthreads();
+    // one dot per action: wave w covers actions w, w+4, ...
+    for (int i = wave; i < A; i += 4) {
+        float p = 0.0f;
+        const unsigned short* wrow = wz + (int64_t)(1 + i) * 512;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            p += habar[0][lane * 8 + j] * bf(wrow[lane * 8 + j]);
+        float s = wave_allreduce_sum(p) + (float)nq * bf(bz[1 + i]);
+        if (lane == 0) qbar[i] = s;
+    }
+    __syncthreads();
+    if (t == 0) {
+        int best = 0;
+        float bv = qbar[0];
+        for (int i = 1; i < A; ++i)
+            if (qbar[i] > bv) { bv = qbar[i]; best = i; }
+        actions[blockIdx.x] = (int64_t)best;
     }
 }
 
@@ -481,17 +513,17 @@ void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
                        void* stream) {
     hipStream_t s = (hipStream_t)stream;
     if (mode == 1) {
-        hipLaunchKernelGGL(heads_finish_kernel<1>, dim3(R / nq), dim3(256), 0,
-                           s, (const unsigned short*)hh,
+        hipLaunchKernelGGL(heads_act_kernel, dim3(R / nq), dim3(256), 0, s,
+                           (const unsigned short*)hh,
                            (const unsigned short*)wz,
-                           (const unsigned short*)bz, R, A, nq, q_out, act_io);
+                           (const unsigned short*)bz, A, nq, act_io);
     } else if (mode == 2) {
-        hipLaunchKernelGGL(heads_finish_kernel<2>, dim3((R + 63) / 64),
+        hipLaunchKernelGGL(heads_finish_kernel<2>, dim3((R + 15) / 16),
                            dim3(256), 0, s, (const unsigned short*)hh,
                            (const unsigned short*)wz,
                            (const unsigned short*)bz, R, A, nq, q_out, act_io);
     } else {
-        hipLaunchKernelGGL(heads_finish_kernel<0>, dim3((R + 63) / 64),
+        hipLaunchKernelGGL(heads_finish_kernel<0>, dim3((R + 15) / 16),
                            dim3(256), 0, s, (const unsigned short*)hh,
                            (const unsigned short*)wz,
                            (const unsigned short*)bz, R, A, nq, q_out, act_io);

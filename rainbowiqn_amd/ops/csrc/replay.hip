@@ -64,28 +64,60 @@ DEV bool slot_valid(int64_t leaf, int64_t capacity, int64_t ring_cap,
 // ---------------------------------------------------------------------------
 // APPEND
 // ---------------------------------------------------------------------------
+// Bulk fields in parallel; the sequential episode-timestep chain
+// (t <- term ? 0 : t+1) is a 64-wide ballot scan on wave 0: position k's t
+// is (k - j - 1) for the last terminal j < k in the chunk, or t0 + k before
+// the first terminal.  One block; ~C/64 scan iterations instead of C.
 __global__ void append_meta_kernel(
     int32_t* actions, float* rewards, uint8_t* terminals, int32_t* timesteps,
     int32_t* ring_state, int ring_id, int64_t ring_cap,
     const float* chunk_meta, int64_t C) {
-    if (threadIdx.x != 0 || blockIdx.x != 0) return;
-    int head = ring_state[4 * ring_id + 0];
-    int count = ring_state[4 * ring_id + 1];
-    int t = ring_state[4 * ring_id + 2];
-    int64_t base = (int64_t)ring_id * ring_cap;
-    for (int64_t k = 0; k < C; ++k) {
-        int local = (int)((head + k) % ring_cap);
-        int64_t slot = base + local;
+    __shared__ int t_final_s;
+    const int head = ring_state[4 * ring_id + 0];
+    const int count = ring_state[4 * ring_id + 1];
+    const int64_t base = (int64_t)ring_id * ring_cap;
+    // bulk: actions/rewards/terminals (all 4 waves)
+    for (int64_t k = threadIdx.x; k < C; k += blockDim.x) {
+        int64_t slot = base + (head + k) % ring_cap;
         actions[slot] = (int32_t)chunk_meta[3 * k + 0];
         rewards[slot] = chunk_meta[3 * k + 1];
-        bool term = chunk_meta[3 * k + 2] > 0.5f;
-        terminals[slot] = term ? 1 : 0;
-        timesteps[slot] = t;
-        t = term ? 0 : t + 1;
+        terminals[slot] = chunk_meta[3 * k + 2] > 0.5f ? 1 : 0;
     }
-    ring_state[4 * ring_id + 0] = (int)((head + C) % ring_cap);
-    ring_state[4 * ring_id + 1] = (int)min((int64_t)count + C, ring_cap);
-    ring_state[4 * ring_id + 2] = t;
+    // timestep scan: wave 0 only, 64 entries per iteration
+    if (threadIdx.x < 64) {
+        const int l = threadIdx.x;
+        int t0 = ring_state[4 * ring_id + 2];  // episode t at chunk start
+        for (int64_t g = 0; g < C; g += 64) {
+            int64_t k = g + l;
+            bool term = (k < C) && (chunk_meta[3 * k + 2] > 0.5f);
+            uint64_t mask = __ballot(term);
+            uint64_t below = mask & ((l == 0) ? 0ull : (~0ull >> (64 - l)));
+            int t;
+            if (below) {
+                int j = 63 - __clzll(below);
+                t = l - j - 1;
+            } else {
+                t = t0 + l;
+            }
+            if (k < C) timesteps[base + (head + k) % ring_cap] = t;
+            // carry episode-t across groups
+            int valid = (int)min((int64_t)64, C - g);
+            uint64_t vmask = mask & ((valid == 64) ? ~0ull : ((1ull << valid) - 1));
+            if (vmask) {
+                int h = 63 - __clzll(vmask);
+                t0 = valid - h - 1;
+            } else {
+                t0 += valid;
+            }
+        }
+        if (l == 0) t_final_s = t0;
+    }
+    __syncthreads();  // everyone has read head/count; scan done
+    if (threadIdx.x == 0) {
+        ring_state[4 * ring_id + 0] = (int)((head + C) % ring_cap);
+        ring_state[4 * ring_id + 1] = (int)min((int64_t)count + C, ring_cap);
+        ring_state[4 * ring_id + 2] = t_final_s;
+    }
 }
 
 __global__ void append_frames_kernel(
@@ -321,7 +353,7 @@ void riqn_replay_append(
     const uint8_t* chunk_frames, const float* chunk_meta, int64_t chunk_len,
     int ring_id, void* stream) {
     hipStream_t s = (hipStream_t)stream;
-    hipLaunchKernelGGL(append_meta_kernel, dim3(1), dim3(64), 0, s,
+    hipLaunchKernelGGL(append_meta_kernel, dim3(1), dim3(256), 0, s,
                        actions, rewards, terminals, timesteps, ring_state,
                        ring_id, ring_cap, chunk_meta, chunk_len);
     hipLaunchKernelGGL(append_frames_kernel, dim3((uint32_t)chunk_len), dim3(256), 0, s,

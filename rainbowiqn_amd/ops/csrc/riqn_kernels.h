@@ -96,6 +96,18 @@ void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
                        int A, int nq, int mode, float* q_out, int64_t* act_io,
                        void* stream);
 
+
+// batched 4-layer NoisyLinear regeneration (rng.hip)
+struct NoisyDesc {
+    const void *w_mu, *w_sigma, *b_mu, *b_sigma;
+    void *w_eps, *b_eps, *w_noisy, *b_noisy;
+    float *eps_in, *eps_out;
+    int64_t n_in, n_out;
+};
+struct NoisyPack4 { NoisyDesc l[4]; };
+void riqn_noisy_material4(NoisyPack4 pack, int bf16, uint64_t* rng_state,
+                          void* stream);
+
 // ---- device synthetic env (bench frame source; env.hip) --------------------
 void riqn_env_step(const int64_t* actions, int32_t* env_state,
                    uint64_t* rng_state, uint8_t* ring, int history,

@@ -114,9 +114,100 @@ __global__ void noisy_material_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Batched noise regeneration: ALL FOUR NoisyLinear layers of one net in two
+// kernels (eps vectors, then outer-product + materialized weights) instead
+// of 16 launches per step.
+// ---------------------------------------------------------------------------
+
+__global__ void noisy_eps_vec4_kernel(NoisyPack4 p, uint64_t* rng_state) {
+    uint64_t seed = rng_state[0], off = rng_state[1];
+    int64_t tid = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    Philox rng(seed, off, (uint32_t)(tid + 0x20000));
+    float z0, z1;
+    rng.normal2(z0, z1);
+    int64_t i = tid * 2;
+    // linear index space: [l0.in | l0.out | l1.in | l1.out | ...]
+    #pragma unroll
+    for (int L = 0; L < 4; ++L) {
+        int64_t nin = p.l[L].n_in, nout = p.l[L].n_out;
+        if (i < nin) {
+            p.l[L].eps_in[i] = scale_noise(z0);
+            if (i + 1 < nin) p.l[L].eps_in[i + 1] = scale_noise(z1);
+            break;
+        }
+        i -= nin;
+        if (i < nout) {
+            p.l[L].eps_out[i] = scale_noise(z0);
+            if (i + 1 < nout) p.l[L].eps_out[i + 1] = scale_noise(z1);
+            break;
+        }
+        i -= nout;
+    }
+    if (tid == 0) rng_state[1] = off + 1;
+}
+
+template <int BF16>
+__global__ void noisy_material4_kernel(NoisyPack4 p) {
+    int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    // weight segments
+    for (int L = 0; L < 4; ++L) {
+        const NoisyDesc& d = p.l[L];
+        const int64_t tot = d.n_in * d.n_out;
+        for (int64_t k = i; k < tot; k += stride) {
+            float e = d.eps_out[k / d.n_in] * d.eps_in[k % d.n_in];
+            if (BF16) {
+                float mu = bf16_to_f32(((const unsigned short*)d.w_mu)[k]);
+                float sg = bf16_to_f32(((const unsigned short*)d.w_sigma)[k]);
+                ((unsigned short*)d.w_eps)[k] = f32_to_bf16(e);
+                ((unsigned short*)d.w_noisy)[k] = f32_to_bf16(mu + sg * e);
+            } else {
+                float mu = ((const float*)d.w_mu)[k];
+                float sg = ((const float*)d.w_sigma)[k];
+                ((float*)d.w_eps)[k] = e;
+                ((float*)d.w_noisy)[k] = mu + sg * e;
+            }
+        }
+        // bias segment
+        for (int64_t k = i; k < d.n_out; k += stride) {
+            float e = d.eps_out[k];
+            if (BF16) {
+                float mu = bf16_to_f32(((const unsigned short*)d.b_mu)[k]);
+                float sg = bf16_to_f32(((const unsigned short*)d.b_sigma)[k]);
+                ((unsigned short*)d.b_eps)[k] = f32_to_bf16(e);
+                ((unsigned short*)d.b_noisy)[k] = f32_to_bf16(mu + sg * e);
+            } else {
+                float mu = ((const float*)d.b_mu)[k];
+                float sg = ((const float*)d.b_sigma)[k];
+                ((float*)d.b_eps)[k] = e;
+                ((float*)d.b_noisy)[k] = mu + sg * e;
+            }
+        }
+    }
+}
+
 }  // namespace
 
 extern "C" {
+
+void riqn_noisy_material4(NoisyPack4 pack, int bf16, uint64_t* rng_state,
+                          void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int64_t vec_total = 0;
+    for (int L = 0; L < 4; ++L)
+        vec_total += pack.l[L].n_in + pack.l[L].n_out;
+    int64_t pairs = (vec_total + 1) / 2;
+    hipLaunchKernelGGL(noisy_eps_vec4_kernel,
+                       dim3((uint32_t)((pairs + 255) / 256)), dim3(256), 0, s,
+                       pack, rng_state);
+    if (bf16)
+        hipLaunchKernelGGL(noisy_material4_kernel<1>, dim3(2048), dim3(256),
+                           0, s, pack);
+    else
+        hipLaunchKernelGGL(noisy_material4_kernel<0>, dim3(2048), dim3(256),
+                           0, s, pack);
+}
 
 void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state,
                        void* stream) {

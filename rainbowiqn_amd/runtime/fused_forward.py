@@ -131,3 +131,20 @@ class FusedForward:
         self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 2,
                               out.view(-1), actions)
         return out
+
+
+def noisy_pack(net: RainbowIQN, device) -> list:
+    """Tensor list for ext.noisy_material4: regenerates noise + materialized
+    noisy weights for all four NoisyLinear layers in 2 kernels (vs 8).
+    Build AFTER FusedForward so weight_noisy points at the concat storage."""
+    ts = []
+    for layer in (net.fc_h_v, net.fc_h_a, net.fc_z_v, net.fc_z_a):
+        if getattr(layer, "weight_noisy", None) is None:
+            layer.enable_fused_noise()
+        eps_in = torch.empty(layer.in_features, device=device)
+        eps_out = torch.empty(layer.out_features, device=device)
+        ts += [layer.weight_mu.detach(), layer.weight_sigma.detach(),
+               layer.bias_mu.detach(), layer.bias_sigma.detach(),
+               layer.weight_epsilon, layer.bias_epsilon,
+               layer.weight_noisy, layer.bias_noisy, eps_in, eps_out]
+    return ts

@@ -96,17 +96,12 @@ class GpuActorLoop:
             )
 
     def reset_noise(self) -> None:
-        for layer in (self.net.fc_h_v, self.net.fc_h_a, self.net.fc_z_v,
-                      self.net.fc_z_a):
-            if getattr(layer, "weight_noisy", None) is None:
-                layer.enable_fused_noise()
-            self.ext.noisy_material(self._eps_in[: layer.in_features],
-                                    self._eps_out[: layer.out_features],
-                                    layer.weight_mu, layer.weight_sigma,
-                                    layer.bias_mu, layer.bias_sigma,
-                                    layer.weight_epsilon, layer.bias_epsilon,
-                                    layer.weight_noisy, layer.bias_noisy,
-                                    self.rng_state)
+        # batched: all 4 NoisyLinear layers in 2 kernels
+        if getattr(self, "_noisy_pack", None) is None:
+            from .fused_forward import noisy_pack
+
+            self._noisy_pack = noisy_pack(self.net, self.device)
+        self.ext.noisy_material4(self._noisy_pack, self.rng_state)
 
     def capture(self) -> None:
         torch.cuda.synchronize()

@@ -99,14 +99,17 @@ class GpuLearner:
 
     # -- pieces -------------------------------------------------------------
     def _reset_noise_device(self, net: RainbowIQN) -> None:
-        for layer in _noisy_layers(net):
-            self.ext.noisy_material(
-                self._eps_in[: layer.in_features],
-                self._eps_out[: layer.out_features],
-                layer.weight_mu, layer.weight_sigma, layer.bias_mu,
-                layer.bias_sigma, layer.weight_epsilon, layer.bias_epsilon,
-                layer.weight_noisy, layer.bias_noisy, self.rng_state,
-            )
+        # batched: all 4 NoisyLinear layers in 2 kernels (K6/K7 fused).
+        # Packs are built lazily AFTER FusedForward re-points weight_noisy.
+        packs = getattr(self, "_noisy_packs", None)
+        if packs is None:
+            packs = self._noisy_packs = {}
+        key = id(net)
+        if key not in packs:
+            from .fused_forward import noisy_pack
+
+            packs[key] = noisy_pack(net, self.device)
+        self.ext.noisy_material4(packs[key], self.rng_state)
 
     def _step_body(self):
         cfg = self.cfg

@@ -1,0 +1,91 @@
+#!/usr/bin/env python3
+"""Microbenchmark of the fused forward kernels at the shapes the Ape-X loop
+uses (act: B=E envs x K taus; learn: B=32 x {K, Np}).
+
+Usage (GPU box): python scripts/kbench.py [iters]
+"""
+import sys
+import time
+
+import torch
+
+from rainbowiqn_amd.config import Config
+from rainbowiqn_amd.models.model import RainbowIQN
+from rainbowiqn_amd.ops import get_extension
+from rainbowiqn_amd.runtime.fused_forward import FusedForward, noisy_pack
+
+
+def timeit(fn, iters):
+    for _ in range(10):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6  # us
+
+
+def main():
+    iters = int(sys.argv[1]) if len(sys.argv) > 1 else 200
+    ext = get_extension()
+    cfg = Config(dtype="bf16", device="cuda")
+    A = 6
+    net = RainbowIQN(cfg, A).cuda().to(torch.bfloat16).train()
+    fw = FusedForward(net, "cuda")
+    pack = noisy_pack(net, "cuda")
+    rng = torch.tensor([123, 0], dtype=torch.int64, device="cuda")
+
+    print(f"{'kernel':<28} {'shape':<18} {'us':>9}")
+    for B, nq, tag in [(256, 32, "act E=256"), (64, 32, "act E=64"),
+                       (32, 32, "learn sel"), (32, 8, "learn tgt")]:
+        R = B * nq
+        x = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
+        taus = torch.rand(R, device="cuda")
+        feat, h, hh = fw._bufs(B, nq)
+        q = torch.empty(R, A, device="cuda")
+        th = torch.empty(B, nq, device="cuda")
+        acts = torch.zeros(B, dtype=torch.int64, device="cuda")
+
+        us = timeit(lambda: ext.conv_encode(x, fw.w1, fw.b1, fw.w2, fw.b2,
+                                            fw.w3p, fw.b3, feat), iters)
+        print(f"{'conv_encode':<28} {f'B={B}':<18} {us:9.2f}")
+        us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq),
+                    iters)
+        print(f"{'tau_embed':<28} {f'R={R}':<18} {us:9.2f}")
+        us = timeit(lambda: ext.heads_gemm(h, fw.wcat, fw.bcat, hh), iters)
+        gf = R * 1024 * 3136 * 2 / 1e9
+        print(f"{'heads_gemm':<28} {f'R={R}':<18} {us:9.2f}  "
+              f"({gf / us * 1e6 / 1e3:.2f} TF)")
+        us = timeit(lambda: ext.heads_finish(hh, fw.wz, fw.bz, A, nq, 1,
+                                             torch.zeros(1, device='cuda'),
+                                             acts), iters)
+        print(f"{'heads_finish act':<28} {f'{tag}':<18} {us:9.2f}")
+        us = timeit(lambda: ext.heads_finish(hh, fw.wz, fw.bz, A, nq, 0, q,
+                                             acts), iters)
+        print(f"{'heads_finish q':<28} {f'R={R}':<18} {us:9.2f}")
+        us = timeit(lambda: ext.heads_finish(hh, fw.wz, fw.bz, A, nq, 2,
+                                             th.view(-1), acts), iters)
+        print(f"{'heads_finish theta':<28} {f'R={R}':<18} {us:9.2f}")
+        print()
+
+    us = timeit(lambda: ext.noisy_material4(pack, rng), iters)
+    print(f"{'noisy_material4':<28} {'4 layers':<18} {us:9.2f}")
+    us = timeit(lambda: ext.conv_pack3(fw.w3, fw.w3p), iters)
+    print(f"{'conv_pack3':<28} {'':<18} {us:9.2f}")
+
+    # replay append at bench chunk size
+    from rainbowiqn_amd.replay.gpu_replay import GpuReplay
+
+    rep = GpuReplay(cfg, n_rings=1, capacity=200_000, device="cuda",
+                    out_dtype=torch.bfloat16)
+    cf = torch.randint(0, 255, (2048, 84, 84), dtype=torch.uint8, device="cuda")
+    cm = torch.rand(2048, 3, device="cuda")
+    us = timeit(lambda: rep.append_actor_chunk(0, cf, meta=cm), 50)
+    print(f"{'replay_append':<28} {'C=2048':<18} {us:9.2f}")
+    us = timeit(lambda: rep.sample(), iters)
+    print(f"{'replay_sample':<28} {'B=32':<18} {us:9.2f}")
+
+
+if __name__ == "__main__":
+    main()
