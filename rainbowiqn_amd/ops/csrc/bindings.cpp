@@ -278,15 +278,22 @@ void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
 }
 
 void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
-               torch::Tensor feat, torch::Tensor h, int64_t nq) {
+               torch::Tensor feat, torch::Tensor h, int64_t nq,
+               c10::optional<torch::Tensor> phi_out = c10::nullopt) {
     CHECK_T(taus, torch::kFloat32);
     CHECK_BF16(w); CHECK_BF16(bias); CHECK_BF16(feat); CHECK_BF16(h);
     int R = (int)taus.numel();
     TORCH_CHECK(h.numel() == (int64_t)R * 3136);
     TORCH_CHECK(w.numel() == 3136 * 64);
     TORCH_CHECK(feat.numel() * nq == h.numel(), "feat rows * nq != R");
+    void* pp = nullptr;
+    if (phi_out.has_value()) {
+        TORCH_CHECK(phi_out->numel() == h.numel());
+        pp = phi_out->data_ptr();
+    }
     riqn_tau_embed(taus.data_ptr<float>(), w.data_ptr(), bias.data_ptr(),
-                   feat.data_ptr(), h.data_ptr(), R, (int)nq, cur_stream());
+                   feat.data_ptr(), h.data_ptr(), pp, R, (int)nq,
+                   cur_stream());
 }
 
 void heads_gemm(torch::Tensor h, torch::Tensor wcat, torch::Tensor bcat,
@@ -336,7 +343,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("stack_gather", &stack_gather);
     m.def("conv_pack3", &conv_pack3);
     m.def("conv_encode", &conv_encode);
-    m.def("tau_embed", &tau_embed);
+    m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt);
     m.def("heads_gemm", &heads_gemm);
     m.def("heads_finish", &heads_finish);
 }

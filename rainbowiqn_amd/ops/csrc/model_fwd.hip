@@ -85,6 +85,17 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
     __syncthreads();
 
     // ---- conv1: M=400 (25 tiles), N=32 (2 tiles), K=256 (8 steps) --------
+    // All 16 B-fragments live in registers, loaded before the input barrier
+    // so their latency hides under the staging; the m-loop then runs on LDS
+    // reads + MFMA only.
+    bf16x8 b1f[2][8];
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+        #pragma unroll
+        for (int ks = 0; ks < 8; ++ks)
+            b1f[nt][ks] = as_bf16x8(
+                *(const u16x8*)(w1 + (nt * 16 + l15) * 256 + ks * 32 + l4 * 8));
+
     for (int mt = wave; mt < 25; mt += 4) {
         f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
         const int m = mt * 16 + l15;           // output pixel (A-row)
@@ -102,12 +113,8 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
                 av = u16x8{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
             }
             bf16x8 a = as_bf16x8(av);
-            #pragma unroll
-            for (int nt = 0; nt < 2; ++nt) {
-                const int n = nt * 16 + l15;
-                bf16x8 bfr = as_bf16x8(*(const u16x8*)(w1 + n * 256 + k0));
-                acc[nt] = mfma16(a, bfr, acc[nt]);
-            }
+            acc[0] = mfma16(a, b1f[0][ks], acc[0]);
+            acc[1] = mfma16(a, b1f[1][ks], acc[1]);
         }
         // epilogue: D row = mt*16 + l4*4 + j (pixel), col = l15 (channel)
         #pragma unroll
@@ -128,46 +135,73 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
     __syncthreads();
 
     // ---- conv2: M=81 (6 tiles), N=64 (4 tiles), K=512 (16 steps) ---------
-    for (int mt = wave; mt < 6; mt += 4) {
-        f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
-                        f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
-        const int m = mt * 16 + l15;
-        const int oy = m / 9, ox = m % 9;
-        const bool mv = m < 81;
-        #pragma unroll 4
+    // k-outer loop with a register double-buffer on the 4 B-fragments: the
+    // next k-step's weight loads are in flight while this step's MFMAs run.
+    {
+        const int mt0 = wave;                // owned m-tiles: wave, wave+4
+        const int nmt = (wave < 2) ? 2 : 1;
+        f32x4 acc[2][4];
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+            #pragma unroll
+            for (int nt = 0; nt < 4; ++nt) acc[mi][nt] = f32x4{0, 0, 0, 0};
+        int oys[2], oxs[2];
+        bool mvs[2];
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+            int m = (mt0 + 4 * mi) * 16 + l15;
+            oys[mi] = m / 9; oxs[mi] = m % 9; mvs[mi] = m < 81;
+        }
+        bf16x8 bbuf[2][4];
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+            bbuf[0][nt] = as_bf16x8(
+                *(const u16x8*)(w2 + (nt * 16 + l15) * 512 + l4 * 8));
+        #pragma unroll 2
         for (int ks = 0; ks < 16; ++ks) {
+            const int cur = ks & 1;
+            if (ks + 1 < 16) {
+                const int k1 = (ks + 1) * 32 + l4 * 8;
+                #pragma unroll
+                for (int nt = 0; nt < 4; ++nt)
+                    bbuf[cur ^ 1][nt] = as_bf16x8(
+                        *(const u16x8*)(w2 + (nt * 16 + l15) * 512 + k1));
+            }
             const int k0 = ks * 32 + l4 * 8;
             const int c = k0 >> 4, t16 = k0 & 15;  // t16 in {0,8}
-            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (mv) {
-                // taps (ky0, kx 0..3) and (ky0+1, kx 0..3)
-                const int ky0 = t16 >> 2;
-                const int base =
-                    LDS_C1 + (c * 20 + oy * 2 + ky0) * C1_STRIDE + ox * 2;
+            const int ky0 = t16 >> 2;
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi) {
+                if (mi >= nmt) break;
+                u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
+                if (mvs[mi]) {
+                    const int base = LDS_C1 +
+                        (c * 20 + oys[mi] * 2 + ky0) * C1_STRIDE + oxs[mi] * 2;
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j) av[j] = lds[base + j];
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j) av[4 + j] = lds[base + C1_STRIDE + j];
+                }
+                bf16x8 a = as_bf16x8(av);
                 #pragma unroll
-                for (int j = 0; j < 4; ++j) av[j] = lds[base + j];
-                #pragma unroll
-                for (int j = 0; j < 4; ++j) av[4 + j] = lds[base + C1_STRIDE + j];
+                for (int nt = 0; nt < 4; ++nt)
+                    acc[mi][nt] = mfma16(a, bbuf[cur][nt], acc[mi][nt]);
             }
-            bf16x8 a = as_bf16x8(av);
+        }
+        for (int mi = 0; mi < nmt; ++mi) {
+            const int mt = mt0 + 4 * mi;
             #pragma unroll
             for (int nt = 0; nt < 4; ++nt) {
                 const int n = nt * 16 + l15;
-                bf16x8 bfr = as_bf16x8(*(const u16x8*)(w2 + n * 512 + k0));
-                acc[nt] = mfma16(a, bfr, acc[nt]);
-            }
-        }
-        #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
-            const int n = nt * 16 + l15;
-            const float bias = bf(b2[n]);
-            #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                const int mm = mt * 16 + l4 * 4 + j;
-                if (mm < 81) {
-                    float v = fmaxf(acc[nt][j] + bias, 0.0f);
-                    lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] =
-                        f32_to_bf16(v);
+                const float bias = bf(b2[n]);
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int mm = mt * 16 + l4 * 4 + j;
+                    if (mm < 81) {
+                        float v = fmaxf(acc[mi][nt][j] + bias, 0.0f);
+                        lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] =
+                            f32_to_bf16(v);
+                    }
                 }
             }
         }
@@ -182,8 +216,21 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
         const int m = mt * 16 + l15;
         const int oy = m / 7, ox = m % 7;
         const bool mv = m < 49;
-        #pragma unroll 4
+        bf16x8 bbuf[2][4];
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+            bbuf[0][nt] = as_bf16x8(
+                *(const u16x8*)(w3p + (nt * 16 + l15) * 1024 + l4 * 8));
+        #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
+            const int cur = ks & 1;
+            if (ks + 1 < 32) {
+                const int k1 = (ks + 1) * 32 + l4 * 8;
+                #pragma unroll
+                for (int nt = 0; nt < 4; ++nt)
+                    bbuf[cur ^ 1][nt] = as_bf16x8(
+                        *(const u16x8*)(w3p + (nt * 16 + l15) * 1024 + k1));
+            }
             const int k0 = ks * 32 + l4 * 8;
             const int c = k0 >> 4, t16 = k0 & 15;
             u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -198,11 +245,8 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
             }
             bf16x8 a = as_bf16x8(av);
             #pragma unroll
-            for (int nt = 0; nt < 4; ++nt) {
-                const int n = nt * 16 + l15;
-                bf16x8 bfr = as_bf16x8(*(const u16x8*)(w3p + n * 1024 + k0));
-                acc[nt] = mfma16(a, bfr, acc[nt]);
-            }
+            for (int nt = 0; nt < 4; ++nt)
+                acc[nt] = mfma16(a, bbuf[cur][nt], acc[nt]);
         }
         unsigned short* out = feat + (int64_t)b * 3136;
         #pragma unroll
@@ -234,20 +278,38 @@ __global__ void pack_conv3_kernel(const unsigned short* __restrict__ w3,
 // 2. tau embed: h[r][n] = relu(sum_k cos(pi*k*tau_r) * W[n][k] + b[n]) * feat[r/nq][n]
 // Grid (ceil(R/16), NSPLIT=4); 196 n-tiles split across gridDim.y * 4 waves.
 // ---------------------------------------------------------------------------
+// Output staged through an LDS tile so the [16 rows x 784 cols] block this
+// WG owns leaves as coalesced 8-byte row stores instead of strided 2-byte
+// scatters; the <=4 feat rows the tile touches are staged the same way.
+// Optionally also writes phi (pre-multiply embedding) for the backward pass.
 __global__ __launch_bounds__(256) void tau_embed_kernel(
     const float* __restrict__ taus,         // [R]
     const unsigned short* __restrict__ w,   // [3136,64] bf16 (iqn_fc.weight)
     const unsigned short* __restrict__ bias,// [3136] bf16
     const unsigned short* __restrict__ feat,// [B,3136] bf16
     unsigned short* __restrict__ h,         // [R,3136] bf16
+    unsigned short* __restrict__ phi_out,   // [R,3136] bf16 or nullptr
     int R, int nq) {
-    const int lane = threadIdx.x & 63;
-    const int wave = threadIdx.x >> 6;
+    __shared__ unsigned short tile[16 * 784];
+    __shared__ unsigned short ftile[4 * 784];
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wave = t >> 6;
     const int l15 = lane & 15;
     const int l4 = lane >> 4;
     const int mt = blockIdx.x;
-    const int r_a = mt * 16 + l15;               // A-fragment row
+    const int nt0 = blockIdx.y * 49;       // first n-tile; cols nt0*16..+784
+    const int r_a = mt * 16 + l15;         // A-fragment row
     const float tau = (r_a < R) ? taus[r_a] : 0.0f;
+
+    // stage the feat rows of the states this tile touches (<= 4 for nq >= 8)
+    const int s0 = (mt * 16) / nq;
+    const int span = min((mt * 16 + 15) / nq - s0 + 1, 4);
+    for (int i = t * 4; i < span * 784; i += 256 * 4) {
+        int si = i / 784, cc = i % 784;
+        int64_t src = (int64_t)min(s0 + si, (R - 1) / nq) * 3136 + nt0 * 16 + cc;
+        *(u16x4*)&ftile[si * 784 + cc] = *(const u16x4*)(feat + src);
+    }
 
     // A-fragments for both k-steps, computed once (k = l4*8 + j + 32*ks)
     bf16x8 afrag[2];
@@ -261,9 +323,8 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
         }
         afrag[ks] = as_bf16x8(av);
     }
+    __syncthreads();
 
-    // n-tiles: 196 total; this block owns [blockIdx.y*49, +49), wave-strided
-    const int nt0 = blockIdx.y * 49;
     for (int nt = nt0 + wave; nt < nt0 + 49; nt += 4) {
         const int n = nt * 16 + l15;
         f32x4 acc = {0, 0, 0, 0};
@@ -273,15 +334,27 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
             acc = mfma16(afrag[ks], bfr, acc);
         }
         const float bb = bf(bias[n]);
+        const int col = (nt - nt0) * 16 + l15;
         #pragma unroll
         for (int j = 0; j < 4; ++j) {
-            const int r = mt * 16 + l4 * 4 + j;  // D row
-            if (r < R) {
-                float phi = bf16_to_f32(f32_to_bf16(fmaxf(acc[j] + bb, 0.0f)));
-                float fv = bf(feat[(int64_t)(r / nq) * 3136 + n]);
-                h[(int64_t)r * 3136 + n] = f32_to_bf16(phi * fv);
-            }
+            const int rr = l4 * 4 + j;           // row within tile
+            const int r = mt * 16 + rr;
+            unsigned short ph =
+                f32_to_bf16(fmaxf(acc[j] + bb, 0.0f));
+            float fv = bf(ftile[min(r / nq - s0, span - 1) * 784 + col]);
+            tile[rr * 784 + col] = f32_to_bf16(bf16_to_f32(ph) * fv);
+            if (phi_out != nullptr && r < R)
+                phi_out[(int64_t)r * 3136 + nt * 16 + l15] = ph;
         }
+    }
+    __syncthreads();
+    // coalesced row stores: 16 rows x 784 cols
+    for (int i = t * 4; i < 16 * 784; i += 256 * 4) {
+        int rr = i / 784, cc = i % 784;
+        int r = mt * 16 + rr;
+        if (r < R)
+            *(u16x4*)(h + (int64_t)r * 3136 + nt0 * 16 + cc) =
+                *(const u16x4*)&tile[rr * 784 + cc];
     }
 }
 
@@ -291,49 +364,80 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
 // ---------------------------------------------------------------------------
 #define HG_BM 64
 #define HG_BN 64
-#define HG_BK 32
+#define HG_BK 64
+// BK=64, two LDS buffers, tiles staged by global_load_lds (16 B per lane,
+// lane-linear LDS image), XOR-swizzled on the SOURCE address so the
+// ds_read_b128 fragment reads don't hit row-stride bank conflicts
+// (cdna_hip_programming.md section 5 glds recipe).  The next K-tile's DMA is
+// in flight while the current one computes; __syncthreads drains it.
 __global__ __launch_bounds__(256) void heads_gemm_kernel(
     const unsigned short* __restrict__ hin,  // [R,3136]
     const unsigned short* __restrict__ wcat, // [1024,3136]
     const unsigned short* __restrict__ bcat, // [1024]
     unsigned short* __restrict__ out,        // [R,1024]
     int R) {
-    __shared__ unsigned short smem[2 * HG_BM * HG_BK];  // As | Bs
-    unsigned short* As = smem;                           // [64][32]
-    unsigned short* Bs = smem + HG_BM * HG_BK;           // [64][32]
+    __shared__ unsigned short smem[2 * 2 * HG_BM * HG_BK];  // 2 bufs x (A|B)
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid of 32x32
     const int l15 = lane & 15, l4 = lane >> 4;
     const int bm = blockIdx.x * HG_BM, bn = blockIdx.y * HG_BN;
 
-    f32x4 acc[2][2] = {{{0, 0, 0, 0}, {0, 0, 0, 0}},
-                       {{0, 0, 0, 0}, {0, 0, 0, 0}}};
-    const int ldrow = t >> 2, ldk = (t & 3) * 8;  // 256 thr: 64 rows x 4 pieces
-
-    for (int k0 = 0; k0 < 3136; k0 += HG_BK) {
-        // stage A and B tiles (each thread one 16B piece per tile)
-        {
-            const int gr = bm + ldrow;
-            u16x8 va = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (gr < R) va = *(const u16x8*)(hin + (int64_t)gr * 3136 + k0 + ldk);
-            *(u16x8*)&As[ldrow * HG_BK + ldk] = va;
-            u16x8 vb = *(const u16x8*)(wcat + (int64_t)(bn + ldrow) * 3136 + k0 + ldk);
-            *(u16x8*)&Bs[ldrow * HG_BK + ldk] = vb;
-        }
-        __syncthreads();
+    // glds partition: per buffer, A = 8 KB = 8 instructions; wave w issues
+    // A-pieces {2w, 2w+1} and B-pieces {2w, 2w+1}.  Instruction i writes LDS
+    // rows 8i..8i+7 (lane l -> row 8i + l/8, 16-B piece l%8); the global
+    // column piece is (l%8) ^ (row%8).
+    auto stage = [&](int buf, int k0) {
+        unsigned short* base = smem + buf * (2 * HG_BM * HG_BK);
         #pragma unroll
-        for (int fm = 0; fm < 2; ++fm) {
-            bf16x8 a = as_bf16x8(
-                *(const u16x8*)&As[(wr * 32 + fm * 16 + l15) * HG_BK + l4 * 8]);
-            #pragma unroll
-            for (int fn = 0; fn < 2; ++fn) {
-                bf16x8 b = as_bf16x8(
-                    *(const u16x8*)&Bs[(wc * 32 + fn * 16 + l15) * HG_BK + l4 * 8]);
-                acc[fm][fn] = mfma16(a, b, acc[fm][fn]);
+        for (int i = 0; i < 2; ++i) {
+            const int inst = wave * 2 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            {   // A rows (clamped so out-of-range rows read safe memory)
+                const int gr = min(bm + row, R - 1);
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (hin + (int64_t)gr * 3136 + k0 + piece * 8);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + inst * 512);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+            }
+            {   // B rows
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (wcat + (int64_t)(bn + row) * 3136 + k0 + piece * 8);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + HG_BM * HG_BK + inst * 512);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
             }
         }
-        __syncthreads();
+    };
+
+    f32x4 acc[2][2] = {{{0, 0, 0, 0}, {0, 0, 0, 0}},
+                       {{0, 0, 0, 0}, {0, 0, 0, 0}}};
+    stage(0, 0);
+    __syncthreads();
+    for (int kt = 0; kt < 3136 / HG_BK; ++kt) {
+        const int cur = kt & 1;
+        if (kt + 1 < 3136 / HG_BK) stage(cur ^ 1, (kt + 1) * HG_BK);
+        const unsigned short* As = smem + cur * (2 * HG_BM * HG_BK);
+        const unsigned short* Bs = As + HG_BM * HG_BK;
+        #pragma unroll
+        for (int ksub = 0; ksub < 2; ++ksub) {
+            #pragma unroll
+            for (int fm = 0; fm < 2; ++fm) {
+                const int row = wr * 32 + fm * 16 + l15;
+                const int piece = (ksub * 4 + l4) ^ (row & 7);
+                bf16x8 a = as_bf16x8(*(const u16x8*)&As[row * 64 + piece * 8]);
+                #pragma unroll
+                for (int fn = 0; fn < 2; ++fn) {
+                    const int nrow = wc * 32 + fn * 16 + l15;
+                    const int bp = (ksub * 4 + l4) ^ (nrow & 7);
+                    bf16x8 b = as_bf16x8(*(const u16x8*)&Bs[nrow * 64 + bp * 8]);
+                    acc[fm][fn] = mfma16(a, b, acc[fm][fn]);
+                }
+            }
+        }
+        __syncthreads();  // drains the in-flight glds for the next buffer
     }
     #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
@@ -491,12 +595,14 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
 }
 
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
-                    const void* feat, void* h, int R, int nq, void* stream) {
+                    const void* feat, void* h, void* phi_out, int R, int nq,
+                    void* stream) {
     dim3 grid((R + 15) / 16, 4);
     hipLaunchKernelGGL(tau_embed_kernel, grid, dim3(256), 0,
                        (hipStream_t)stream, taus, (const unsigned short*)w,
                        (const unsigned short*)bias,
-                       (const unsigned short*)feat, (unsigned short*)h, R, nq);
+                       (const unsigned short*)feat, (unsigned short*)h,
+                       (unsigned short*)phi_out, R, nq);
 }
 
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,

@@ -135,17 +135,41 @@ __global__ void append_frames_kernel(
     for (int i = threadIdx.x; i < FRAME_PX / 4; i += blockDim.x) dst[i] = src[i];
 }
 
-__global__ void append_tree_kernel(
+// Appended leaves are a contiguous ring range, so instead of C atomic
+// delta-walks to the root (serialized near the top), set the leaves with
+// plain stores and recompute the affected ancestor RANGE level by level
+// (node = left + right) in one block with barriers.  Handles ring wrap as
+// two ranges.  Also removes fp32 drift (no += deltas on this path).
+__global__ __launch_bounds__(1024) void append_tree_kernel(
     float* tree, int64_t tree_size, const int32_t* ring_state, int ring_id,
     int64_t ring_cap, const float* max_priority, int64_t C) {
-    int head_new = ring_state[4 * ring_id + 0];
-    int64_t head_old = (head_new - C % ring_cap + ring_cap) % ring_cap;
-    float p = max_priority[0];
-    for (int64_t k = threadIdx.x + (int64_t)blockIdx.x * blockDim.x; k < C;
-         k += (int64_t)gridDim.x * blockDim.x) {
-        int local = (int)((head_old + k) % ring_cap);
-        int64_t leaf = (int64_t)ring_id * ring_cap + local;
-        tree_set_leaf(tree, tree_size, leaf, p);
+    const int head_new = ring_state[4 * ring_id + 0];
+    const int64_t head_old = (head_new - C % ring_cap + ring_cap) % ring_cap;
+    const float p = max_priority[0];
+    const int64_t base = (int64_t)ring_id * ring_cap;
+    const int t = threadIdx.x;
+    // two leaf ranges [s0, e0) and [s1, e1) in absolute leaf index
+    const int64_t first = min(C, ring_cap - head_old);
+    const int64_t s0 = base + head_old, e0 = s0 + first;
+    const int64_t s1 = base, e1 = base + (C - first);
+    for (int64_t k = t; k < C; k += blockDim.x) {
+        int64_t leaf = (k < first) ? (s0 + k) : (s1 + (k - first));
+        tree[tree_size + leaf] = p;
+    }
+    __syncthreads();
+    // recompute ancestors: ranges in node space, inclusive
+    int64_t lo0 = (tree_size + s0) >> 1, hi0 = (tree_size + e0 - 1) >> 1;
+    int64_t lo1 = (tree_size + s1) >> 1, hi1 = (tree_size + e1 - 1) >> 1;
+    const bool two = C > first;
+    while (hi0 >= 1) {
+        for (int64_t n = lo0 + t; n <= hi0; n += blockDim.x)
+            tree[n] = tree[2 * n] + tree[2 * n + 1];
+        if (two && hi0 > 1)  // second range merges with the first near the top
+            for (int64_t n = lo1 + t; n <= hi1 && n < lo0; n += blockDim.x)
+                tree[n] = tree[2 * n] + tree[2 * n + 1];
+        __syncthreads();
+        if (hi0 == 1) break;
+        lo0 >>= 1; hi0 >>= 1; lo1 >>= 1; hi1 >>= 1;
     }
 }
 
@@ -359,8 +383,7 @@ void riqn_replay_append(
     hipLaunchKernelGGL(append_frames_kernel, dim3((uint32_t)chunk_len), dim3(256), 0, s,
                        frames, ring_state, ring_id, ring_cap, chunk_frames,
                        chunk_len);
-    int blocks = (int)((chunk_len + 255) / 256);
-    hipLaunchKernelGGL(append_tree_kernel, dim3(blocks), dim3(256), 0, s,
+    hipLaunchKernelGGL(append_tree_kernel, dim3(1), dim3(1024), 0, s,
                        tree, tree_size, ring_state, ring_id, ring_cap,
                        max_priority, chunk_len);
 }

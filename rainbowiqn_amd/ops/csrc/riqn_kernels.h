@@ -89,7 +89,8 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* w2, const void* b2, const void* w3p,
                       const void* b3, void* feat, int B, void* stream);
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
-                    const void* feat, void* h, int R, int nq, void* stream);
+                    const void* feat, void* h, void* phi_out, int R, int nq,
+                    void* stream);
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
                      void* out, int R, void* stream);
 void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
