@@ -56,7 +56,7 @@ DEV float bf(const unsigned short u) { return bf16_to_f32(u); }
 #define LDS_C2 (29568 + 15360)
 #define LDS_TOTAL (29568 + 15360 + 6912)
 
-__global__ __launch_bounds__(256, 1) void conv_encode_kernel(
+__global__ __launch_bounds__(512, 1) void conv_encode_kernel(
     const unsigned short* __restrict__ x,    // [B,4,84,84] bf16
     const unsigned short* __restrict__ w1,   // [32,256]  bf16
     const unsigned short* __restrict__ b1,   // [32]      bf16
@@ -76,7 +76,7 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
     // ---- stage input state into LDS (row stride 88) ----------------------
     {
         const unsigned short* src = x + (int64_t)b * (4 * 84 * 84);
-        for (int e = t * 4; e < 4 * 84 * 84; e += 256 * 4) {
+        for (int e = t * 4; e < 4 * 84 * 84; e += 512 * 4) {
             u16x4 v = *(const u16x4*)(src + e);
             int c = e / 7056, r = (e % 7056) / 84, cc = e % 84;
             *(u16x4*)&lds[LDS_IN + (c * 84 + r) * IN_STRIDE + cc] = v;
@@ -96,7 +96,7 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
             b1f[nt][ks] = as_bf16x8(
                 *(const u16x8*)(w1 + (nt * 16 + l15) * 256 + ks * 32 + l4 * 8));
 
-    for (int mt = wave; mt < 25; mt += 4) {
+    for (int mt = wave; mt < 25; mt += 8) {
         f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
         const int m = mt * 16 + l15;           // output pixel (A-row)
         const int oy = m / 20, ox = m % 20;
@@ -135,73 +135,57 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
     __syncthreads();
 
     // ---- conv2: M=81 (6 tiles), N=64 (4 tiles), K=512 (16 steps) ---------
-    // k-outer loop with a register double-buffer on the 4 B-fragments: the
-    // next k-step's weight loads are in flight while this step's MFMAs run.
-    {
-        const int mt0 = wave;                // owned m-tiles: wave, wave+4
-        const int nmt = (wave < 2) ? 2 : 1;
-        f32x4 acc[2][4];
+    // Work unit = (m-tile, n-half): 12 units over 8 waves.  k-outer loop
+    // with a register double-buffer on the B-fragments: the next k-step's
+    // weight loads are in flight while this step's MFMAs run.
+    for (int u = wave; u < 12; u += 8) {
+        const int mt = u >> 1, nh = u & 1;
+        f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+        const int m = mt * 16 + l15;
+        const int oy = m / 9, ox = m % 9;
+        const bool mv = m < 81;
+        bf16x8 bbuf[2][2];
         #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
-            #pragma unroll
-            for (int nt = 0; nt < 4; ++nt) acc[mi][nt] = f32x4{0, 0, 0, 0};
-        int oys[2], oxs[2];
-        bool mvs[2];
-        #pragma unroll
-        for (int mi = 0; mi < 2; ++mi) {
-            int m = (mt0 + 4 * mi) * 16 + l15;
-            oys[mi] = m / 9; oxs[mi] = m % 9; mvs[mi] = m < 81;
-        }
-        bf16x8 bbuf[2][4];
-        #pragma unroll
-        for (int nt = 0; nt < 4; ++nt)
-            bbuf[0][nt] = as_bf16x8(
-                *(const u16x8*)(w2 + (nt * 16 + l15) * 512 + l4 * 8));
+        for (int nt = 0; nt < 2; ++nt)
+            bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
+                w2 + ((nh * 2 + nt) * 16 + l15) * 512 + l4 * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 16; ++ks) {
             const int cur = ks & 1;
             if (ks + 1 < 16) {
                 const int k1 = (ks + 1) * 32 + l4 * 8;
                 #pragma unroll
-                for (int nt = 0; nt < 4; ++nt)
-                    bbuf[cur ^ 1][nt] = as_bf16x8(
-                        *(const u16x8*)(w2 + (nt * 16 + l15) * 512 + k1));
+                for (int nt = 0; nt < 2; ++nt)
+                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                        w2 + ((nh * 2 + nt) * 16 + l15) * 512 + k1));
             }
             const int k0 = ks * 32 + l4 * 8;
             const int c = k0 >> 4, t16 = k0 & 15;  // t16 in {0,8}
             const int ky0 = t16 >> 2;
-            #pragma unroll
-            for (int mi = 0; mi < 2; ++mi) {
-                if (mi >= nmt) break;
-                u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
-                if (mvs[mi]) {
-                    const int base = LDS_C1 +
-                        (c * 20 + oys[mi] * 2 + ky0) * C1_STRIDE + oxs[mi] * 2;
-                    #pragma unroll
-                    for (int j = 0; j < 4; ++j) av[j] = lds[base + j];
-                    #pragma unroll
-                    for (int j = 0; j < 4; ++j) av[4 + j] = lds[base + C1_STRIDE + j];
-                }
-                bf16x8 a = as_bf16x8(av);
+            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (mv) {
+                const int base =
+                    LDS_C1 + (c * 20 + oy * 2 + ky0) * C1_STRIDE + ox * 2;
                 #pragma unroll
-                for (int nt = 0; nt < 4; ++nt)
-                    acc[mi][nt] = mfma16(a, bbuf[cur][nt], acc[mi][nt]);
+                for (int j = 0; j < 4; ++j) av[j] = lds[base + j];
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) av[4 + j] = lds[base + C1_STRIDE + j];
             }
+            bf16x8 a = as_bf16x8(av);
+            acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
+            acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
         }
-        for (int mi = 0; mi < nmt; ++mi) {
-            const int mt = mt0 + 4 * mi;
+        #pragma unroll
+        for (int nt = 0; nt < 2; ++nt) {
+            const int n = (nh * 2 + nt) * 16 + l15;
+            const float bias = bf(b2[n]);
             #pragma unroll
-            for (int nt = 0; nt < 4; ++nt) {
-                const int n = nt * 16 + l15;
-                const float bias = bf(b2[n]);
-                #pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    const int mm = mt * 16 + l4 * 4 + j;
-                    if (mm < 81) {
-                        float v = fmaxf(acc[mi][nt][j] + bias, 0.0f);
-                        lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] =
-                            f32_to_bf16(v);
-                    }
+            for (int j = 0; j < 4; ++j) {
+                const int mm = mt * 16 + l4 * 4 + j;
+                if (mm < 81) {
+                    float v = fmaxf(acc[nt][j] + bias, 0.0f);
+                    lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] =
+                        f32_to_bf16(v);
                 }
             }
         }
@@ -209,27 +193,27 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
     __syncthreads();
 
     // ---- conv3: M=49 (4 tiles), N=64 (4 tiles), K=1024 padded (32 steps) -
+    // 8 units = (m-tile, n-half), exactly one per wave.
     {
-        const int mt = wave;  // exactly 4 tiles
-        f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
-                        f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+        const int mt = wave >> 1, nh = wave & 1;
+        f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
         const int m = mt * 16 + l15;
         const int oy = m / 7, ox = m % 7;
         const bool mv = m < 49;
-        bf16x8 bbuf[2][4];
+        bf16x8 bbuf[2][2];
         #pragma unroll
-        for (int nt = 0; nt < 4; ++nt)
-            bbuf[0][nt] = as_bf16x8(
-                *(const u16x8*)(w3p + (nt * 16 + l15) * 1024 + l4 * 8));
+        for (int nt = 0; nt < 2; ++nt)
+            bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
+                w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + l4 * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
             const int cur = ks & 1;
             if (ks + 1 < 32) {
                 const int k1 = (ks + 1) * 32 + l4 * 8;
                 #pragma unroll
-                for (int nt = 0; nt < 4; ++nt)
-                    bbuf[cur ^ 1][nt] = as_bf16x8(
-                        *(const u16x8*)(w3p + (nt * 16 + l15) * 1024 + k1));
+                for (int nt = 0; nt < 2; ++nt)
+                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                        w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
             }
             const int k0 = ks * 32 + l4 * 8;
             const int c = k0 >> 4, t16 = k0 & 15;
@@ -244,14 +228,13 @@ __global__ __launch_bounds__(256, 1) void conv_encode_kernel(
                 }
             }
             bf16x8 a = as_bf16x8(av);
-            #pragma unroll
-            for (int nt = 0; nt < 4; ++nt)
-                acc[nt] = mfma16(a, bbuf[cur][nt], acc[nt]);
+            acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
+            acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
         }
         unsigned short* out = feat + (int64_t)b * 3136;
         #pragma unroll
-        for (int nt = 0; nt < 4; ++nt) {
-            const int n = nt * 16 + l15;
+        for (int nt = 0; nt < 2; ++nt) {
+            const int n = (nh * 2 + nt) * 16 + l15;
             const float bias = bf(b3[n]);
             #pragma unroll
             for (int j = 0; j < 4; ++j) {
@@ -290,68 +273,75 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
     unsigned short* __restrict__ h,         // [R,3136] bf16
     unsigned short* __restrict__ phi_out,   // [R,3136] bf16 or nullptr
     int R, int nq) {
-    __shared__ unsigned short tile[16 * 784];
+    __shared__ unsigned short tile[2 * 16 * 784];  // two m-tiles per WG
     __shared__ unsigned short ftile[4 * 784];
     const int t = threadIdx.x;
     const int lane = t & 63;
     const int wave = t >> 6;
     const int l15 = lane & 15;
     const int l4 = lane >> 4;
-    const int mt = blockIdx.x;
+    const int mt0 = blockIdx.x * 2;        // two m-tiles -> W re-read halved
     const int nt0 = blockIdx.y * 49;       // first n-tile; cols nt0*16..+784
-    const int r_a = mt * 16 + l15;         // A-fragment row
-    const float tau = (r_a < R) ? taus[r_a] : 0.0f;
 
     // stage the feat rows of the states this tile touches (<= 4 for nq >= 8)
-    const int s0 = (mt * 16) / nq;
-    const int span = min((mt * 16 + 15) / nq - s0 + 1, 4);
+    const int s0 = (mt0 * 16) / nq;
+    const int span = min((mt0 * 16 + 31) / nq - s0 + 1, 4);
     for (int i = t * 4; i < span * 784; i += 256 * 4) {
         int si = i / 784, cc = i % 784;
         int64_t src = (int64_t)min(s0 + si, (R - 1) / nq) * 3136 + nt0 * 16 + cc;
         *(u16x4*)&ftile[si * 784 + cc] = *(const u16x4*)(feat + src);
     }
 
-    // A-fragments for both k-steps, computed once (k = l4*8 + j + 32*ks)
-    bf16x8 afrag[2];
+    // A-fragments for both m-tiles x both k-steps (k = l4*8 + j + 32*ks)
+    bf16x8 afrag[2][2];
     #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-        u16x8 av;
+    for (int mi = 0; mi < 2; ++mi) {
+        const int r_a = (mt0 + mi) * 16 + l15;
+        const float tau = (r_a < R) ? taus[r_a] : 0.0f;
         #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-            float k = (float)(ks * 32 + l4 * 8 + j);
-            av[j] = f32_to_bf16(__cosf(3.14159265358979f * k * tau));
+        for (int ks = 0; ks < 2; ++ks) {
+            u16x8 av;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                float k = (float)(ks * 32 + l4 * 8 + j);
+                av[j] = f32_to_bf16(__cosf(3.14159265358979f * k * tau));
+            }
+            afrag[mi][ks] = as_bf16x8(av);
         }
-        afrag[ks] = as_bf16x8(av);
     }
     __syncthreads();
 
     for (int nt = nt0 + wave; nt < nt0 + 49; nt += 4) {
         const int n = nt * 16 + l15;
-        f32x4 acc = {0, 0, 0, 0};
+        f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
             bf16x8 bfr = as_bf16x8(*(const u16x8*)(w + n * 64 + ks * 32 + l4 * 8));
-            acc = mfma16(afrag[ks], bfr, acc);
+            acc[0] = mfma16(afrag[0][ks], bfr, acc[0]);
+            acc[1] = mfma16(afrag[1][ks], bfr, acc[1]);
         }
         const float bb = bf(bias[n]);
         const int col = (nt - nt0) * 16 + l15;
         #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            const int rr = l4 * 4 + j;           // row within tile
-            const int r = mt * 16 + rr;
-            unsigned short ph =
-                f32_to_bf16(fmaxf(acc[j] + bb, 0.0f));
-            float fv = bf(ftile[min(r / nq - s0, span - 1) * 784 + col]);
-            tile[rr * 784 + col] = f32_to_bf16(bf16_to_f32(ph) * fv);
-            if (phi_out != nullptr && r < R)
-                phi_out[(int64_t)r * 3136 + nt * 16 + l15] = ph;
+        for (int mi = 0; mi < 2; ++mi) {
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int rr = l4 * 4 + j;       // row within m-tile
+                const int r = (mt0 + mi) * 16 + rr;
+                unsigned short ph = f32_to_bf16(fmaxf(acc[mi][j] + bb, 0.0f));
+                float fv = bf(ftile[min(r / nq - s0, span - 1) * 784 + col]);
+                tile[(mi * 16 + rr) * 784 + col] =
+                    f32_to_bf16(bf16_to_f32(ph) * fv);
+                if (phi_out != nullptr && r < R)
+                    phi_out[(int64_t)r * 3136 + nt * 16 + l15] = ph;
+            }
         }
     }
     __syncthreads();
-    // coalesced row stores: 16 rows x 784 cols
-    for (int i = t * 4; i < 16 * 784; i += 256 * 4) {
+    // coalesced row stores: 32 rows x 784 cols
+    for (int i = t * 4; i < 32 * 784; i += 256 * 4) {
         int rr = i / 784, cc = i % 784;
-        int r = mt * 16 + rr;
+        int r = mt0 * 16 + rr;
         if (r < R)
             *(u16x4*)(h + (int64_t)r * 3136 + nt0 * 16 + cc) =
                 *(const u16x4*)&tile[rr * 784 + cc];
@@ -586,7 +576,7 @@ void riqn_conv_pack3(const void* w3, void* w3p, void* stream) {
 void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* w2, const void* b2, const void* w3p,
                       const void* b3, void* feat, int B, void* stream) {
-    hipLaunchKernelGGL(conv_encode_kernel, dim3(B), dim3(256), 0,
+    hipLaunchKernelGGL(conv_encode_kernel, dim3(B), dim3(512), 0,
                        (hipStream_t)stream, (const unsigned short*)x,
                        (const unsigned short*)w1, (const unsigned short*)b1,
                        (const unsigned short*)w2, (const unsigned short*)b2,
@@ -597,7 +587,7 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     const void* feat, void* h, void* phi_out, int R, int nq,
                     void* stream) {
-    dim3 grid((R + 15) / 16, 4);
+    dim3 grid((R + 31) / 32, 4);
     hipLaunchKernelGGL(tau_embed_kernel, grid, dim3(256), 0,
                        (hipStream_t)stream, taus, (const unsigned short*)w,
                        (const unsigned short*)bias,
