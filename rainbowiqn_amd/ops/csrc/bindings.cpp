@@ -263,7 +263,9 @@ void conv_pack3(torch::Tensor w3, torch::Tensor w3p) {
 
 void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
                  torch::Tensor w2, torch::Tensor b2, torch::Tensor w3p,
-                 torch::Tensor b3, torch::Tensor feat) {
+                 torch::Tensor b3, torch::Tensor feat,
+                 c10::optional<torch::Tensor> c1g = c10::nullopt,
+                 c10::optional<torch::Tensor> c2g = c10::nullopt) {
     CHECK_BF16(x);
     CHECK_BF16(w1); CHECK_BF16(w2); CHECK_BF16(w3p);
     CHECK_BF16(b1); CHECK_BF16(b2); CHECK_BF16(b3);
@@ -272,9 +274,16 @@ void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
     TORCH_CHECK(x.numel() == (int64_t)B * 4 * 84 * 84, "history must be 4");
     TORCH_CHECK(w1.numel() == 32 * 256 && w2.numel() == 64 * 512);
     TORCH_CHECK(feat.numel() == (int64_t)B * 3136);
+    void *p1 = nullptr, *p2 = nullptr;
+    if (c1g.has_value()) {
+        TORCH_CHECK(c1g->numel() == (int64_t)B * 12800 &&
+                    c2g->numel() == (int64_t)B * 5184);
+        p1 = c1g->data_ptr();
+        p2 = c2g->data_ptr();
+    }
     riqn_conv_encode(x.data_ptr(), w1.data_ptr(), b1.data_ptr(), w2.data_ptr(),
                      b2.data_ptr(), w3p.data_ptr(), b3.data_ptr(),
-                     feat.data_ptr(), B, cur_stream());
+                     feat.data_ptr(), p1, p2, B, cur_stream());
 }
 
 void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
@@ -324,6 +333,132 @@ void heads_finish(torch::Tensor hh, torch::Tensor wz, torch::Tensor bz,
                       act_io.data_ptr<int64_t>(), cur_stream());
 }
 
+
+// ---- hand-written learn-step backward (model_bwd.hip) ----------------------
+
+void finish_bwd_dhh(torch::Tensor dtheta, torch::Tensor actions,
+                    torch::Tensor hh, torch::Tensor wz, torch::Tensor dhh,
+                    int64_t A, int64_t nq) {
+    CHECK_T(dtheta, torch::kFloat32);
+    CHECK_T(actions, torch::kInt64);
+    CHECK_BF16(hh); CHECK_BF16(wz); CHECK_BF16(dhh);
+    int R = (int)dtheta.numel();
+    TORCH_CHECK(hh.numel() == (int64_t)R * 1024 && dhh.numel() == hh.numel());
+    riqn_finish_bwd_dhh(dtheta.data_ptr<float>(),
+                        actions.data_ptr<int64_t>(), hh.data_ptr(),
+                        wz.data_ptr(), dhh.data_ptr(), R, (int)A, (int)nq,
+                        cur_stream());
+}
+
+void zgrad(torch::Tensor dtheta, torch::Tensor actions, torch::Tensor hh,
+           int64_t A, int64_t nq, std::vector<torch::Tensor> eps,
+           std::vector<torch::Tensor> grads) {
+    TORCH_CHECK(eps.size() == 4 && grads.size() == 8);
+    int R = (int)dtheta.numel();
+    riqn_zgrad(dtheta.data_ptr<float>(), actions.data_ptr<int64_t>(),
+               hh.data_ptr(), R, (int)A, (int)nq, eps[0].data_ptr(),
+               eps[1].data_ptr(), eps[2].data_ptr(), eps[3].data_ptr(),
+               grads[0].data_ptr(), grads[1].data_ptr(), grads[2].data_ptr(),
+               grads[3].data_ptr(), grads[4].data_ptr(), grads[5].data_ptr(),
+               grads[6].data_ptr(), grads[7].data_ptr(), cur_stream());
+}
+
+void gemm_dx(torch::Tensor dhh, torch::Tensor wcat, torch::Tensor feat,
+             torch::Tensor phi, torch::Tensor dphi, torch::Tensor dfeat,
+             int64_t nq) {
+    CHECK_BF16(dhh); CHECK_BF16(wcat); CHECK_BF16(feat); CHECK_BF16(phi);
+    CHECK_BF16(dphi);
+    CHECK_T(dfeat, torch::kFloat32);
+    int R = (int)dhh.size(0);
+    TORCH_CHECK(phi.numel() == (int64_t)R * 3136 && dphi.numel() == phi.numel());
+    TORCH_CHECK(dfeat.numel() * nq == (int64_t)R * 3136);
+    riqn_gemm_dx(dhh.data_ptr(), wcat.data_ptr(), feat.data_ptr(),
+                 phi.data_ptr(), dphi.data_ptr(), dfeat.data_ptr<float>(), R,
+                 (int)nq, cur_stream());
+}
+
+void gemm_dwcat(torch::Tensor dhh, torch::Tensor h, torch::Tensor weps_v,
+                torch::Tensor weps_a, std::vector<torch::Tensor> grads) {
+    TORCH_CHECK(grads.size() == 4);
+    CHECK_BF16(dhh); CHECK_BF16(h);
+    int R = (int)dhh.size(0);
+    riqn_gemm_dwcat(dhh.data_ptr(), h.data_ptr(), weps_v.data_ptr(),
+                    weps_a.data_ptr(), grads[0].data_ptr(),
+                    grads[1].data_ptr(), grads[2].data_ptr(),
+                    grads[3].data_ptr(), R, cur_stream());
+}
+
+void bias_grad_heads(torch::Tensor dhh, torch::Tensor beps_v,
+                     torch::Tensor beps_a, std::vector<torch::Tensor> grads) {
+    TORCH_CHECK(grads.size() == 4);
+    int R = (int)dhh.size(0);
+    riqn_bias_grad_heads(dhh.data_ptr(), R, beps_v.data_ptr(),
+                         beps_a.data_ptr(), grads[0].data_ptr(),
+                         grads[1].data_ptr(), grads[2].data_ptr(),
+                         grads[3].data_ptr(), cur_stream());
+}
+
+void bias_grad_cols(torch::Tensor X, torch::Tensor db) {
+    CHECK_BF16(X); CHECK_BF16(db);
+    riqn_bias_grad_cols(X.data_ptr(), (int)X.size(0), (int)X.size(1),
+                        db.data_ptr(), cur_stream());
+}
+
+void tau_dw(torch::Tensor dphi, torch::Tensor taus, torch::Tensor dwiqn) {
+    CHECK_BF16(dphi); CHECK_BF16(dwiqn);
+    CHECK_T(taus, torch::kFloat32);
+    TORCH_CHECK(dwiqn.numel() == 3136 * 64);
+    riqn_tau_dw(dphi.data_ptr(), taus.data_ptr<float>(), dwiqn.data_ptr(),
+                (int)taus.numel(), cur_stream());
+}
+
+void conv_bwd_data(torch::Tensor dfeat, torch::Tensor feat, torch::Tensor c1g,
+                   torch::Tensor c2g, torch::Tensor w3tp, torch::Tensor w2tp,
+                   torch::Tensor dpre3g, torch::Tensor dc2g,
+                   torch::Tensor dc1g, torch::Tensor db1s, torch::Tensor db2s,
+                   torch::Tensor db3s) {
+    CHECK_T(dfeat, torch::kFloat32);
+    int B = (int)dfeat.size(0);
+    TORCH_CHECK(dpre3g.numel() == (int64_t)64 * B * 49);
+    TORCH_CHECK(dc2g.numel() == (int64_t)64 * B * 81);
+    TORCH_CHECK(dc1g.numel() == (int64_t)32 * B * 400);
+    riqn_conv_bwd_data(dfeat.data_ptr<float>(), feat.data_ptr(),
+                       c1g.data_ptr(), c2g.data_ptr(), w3tp.data_ptr(),
+                       w2tp.data_ptr(), dpre3g.data_ptr(), dc2g.data_ptr(),
+                       dc1g.data_ptr(), db1s.data_ptr<float>(),
+                       db2s.data_ptr<float>(), db3s.data_ptr<float>(), B,
+                       cur_stream());
+}
+
+void conv_bwd_dw(int64_t layer, torch::Tensor dY, torch::Tensor act,
+                 torch::Tensor dWs, int64_t B, int64_t splitk) {
+    CHECK_BF16(dY); CHECK_BF16(act);
+    CHECK_T(dWs, torch::kFloat32);
+    riqn_conv_bwd_dw((int)layer, dY.data_ptr(), act.data_ptr(),
+                     dWs.data_ptr<float>(), (int)B, (int)splitk,
+                     cur_stream());
+}
+
+void conv_finalize(std::vector<torch::Tensor> scr,
+                   std::vector<torch::Tensor> grads) {
+    TORCH_CHECK(scr.size() == 6 && grads.size() == 6);
+    riqn_conv_finalize(scr[0].data_ptr<float>(), scr[1].data_ptr<float>(),
+                       scr[2].data_ptr<float>(), scr[3].data_ptr<float>(),
+                       scr[4].data_ptr<float>(), scr[5].data_ptr<float>(),
+                       grads[0].data_ptr(), grads[1].data_ptr(),
+                       grads[2].data_ptr(), grads[3].data_ptr(),
+                       grads[4].data_ptr(), grads[5].data_ptr(),
+                       cur_stream());
+}
+
+void pack_bwd_weights(torch::Tensor w3, torch::Tensor w2, torch::Tensor w3tp,
+                      torch::Tensor w2tp) {
+    CHECK_BF16(w3); CHECK_BF16(w2); CHECK_BF16(w3tp); CHECK_BF16(w2tp);
+    TORCH_CHECK(w3tp.numel() == 64 * 1024 && w2tp.numel() == 32 * 1024);
+    riqn_pack_bwd_weights(w3.data_ptr(), w2.data_ptr(), w3tp.data_ptr(),
+                          w2tp.data_ptr(), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -342,8 +477,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("noisy_material4", &noisy_material4);
     m.def("stack_gather", &stack_gather);
     m.def("conv_pack3", &conv_pack3);
-    m.def("conv_encode", &conv_encode);
+    m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
     m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt);
     m.def("heads_gemm", &heads_gemm);
     m.def("heads_finish", &heads_finish);
+    m.def("finish_bwd_dhh", &finish_bwd_dhh);
+    m.def("zgrad", &zgrad);
+    m.def("gemm_dx", &gemm_dx);
+    m.def("gemm_dwcat", &gemm_dwcat);
+    m.def("bias_grad_heads", &bias_grad_heads);
+    m.def("bias_grad_cols", &bias_grad_cols);
+    m.def("tau_dw", &tau_dw);
+    m.def("conv_bwd_data", &conv_bwd_data);
+    m.def("conv_bwd_dw", &conv_bwd_dw);
+    m.def("conv_finalize", &conv_finalize);
+    m.def("pack_bwd_weights", &pack_bwd_weights);
 }

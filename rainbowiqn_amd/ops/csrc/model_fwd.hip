@@ -64,7 +64,9 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
     const unsigned short* __restrict__ b2,   // [64]
     const unsigned short* __restrict__ w3p,  // [64,1024] bf16 (9->16 padded)
     const unsigned short* __restrict__ b3,   // [64]
-    unsigned short* __restrict__ feat) {     // [B,3136]  bf16
+    unsigned short* __restrict__ feat,       // [B,3136]  bf16
+    unsigned short* __restrict__ c1g,        // [B,32*400] or nullptr (bwd)
+    unsigned short* __restrict__ c2g) {      // [B,64*81]  or nullptr (bwd)
     __shared__ unsigned short lds[LDS_TOTAL];
     const int b = blockIdx.x;
     const int t = threadIdx.x;
@@ -125,9 +127,10 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
             for (int j = 0; j < 4; ++j) {
                 const int mm = mt * 16 + l4 * 4 + j;
                 if (mm < 400) {
-                    float v = fmaxf(acc[nt][j] + bias, 0.0f);
-                    lds[LDS_C1 + (n * 20 + mm / 20) * C1_STRIDE + mm % 20] =
-                        f32_to_bf16(v);
+                    unsigned short o = f32_to_bf16(fmaxf(acc[nt][j] + bias, 0.0f));
+                    lds[LDS_C1 + (n * 20 + mm / 20) * C1_STRIDE + mm % 20] = o;
+                    if (c1g != nullptr)
+                        c1g[(int64_t)b * 12800 + n * 400 + mm] = o;
                 }
             }
         }
@@ -183,9 +186,10 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
             for (int j = 0; j < 4; ++j) {
                 const int mm = mt * 16 + l4 * 4 + j;
                 if (mm < 81) {
-                    float v = fmaxf(acc[nt][j] + bias, 0.0f);
-                    lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] =
-                        f32_to_bf16(v);
+                    unsigned short o = f32_to_bf16(fmaxf(acc[nt][j] + bias, 0.0f));
+                    lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] = o;
+                    if (c2g != nullptr)
+                        c2g[(int64_t)b * 5184 + n * 81 + mm] = o;
                 }
             }
         }
@@ -575,13 +579,15 @@ void riqn_conv_pack3(const void* w3, void* w3p, void* stream) {
 
 void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* w2, const void* b2, const void* w3p,
-                      const void* b3, void* feat, int B, void* stream) {
+                      const void* b3, void* feat, void* c1g, void* c2g, int B,
+                      void* stream) {
     hipLaunchKernelGGL(conv_encode_kernel, dim3(B), dim3(512), 0,
                        (hipStream_t)stream, (const unsigned short*)x,
                        (const unsigned short*)w1, (const unsigned short*)b1,
                        (const unsigned short*)w2, (const unsigned short*)b2,
                        (const unsigned short*)w3p, (const unsigned short*)b3,
-                       (unsigned short*)feat);
+                       (unsigned short*)feat, (unsigned short*)c1g,
+                       (unsigned short*)c2g);
 }
 
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,

@@ -87,7 +87,8 @@ void riqn_noisy_material(float* eps_in, int64_t n_in, float* eps_out,
 void riqn_conv_pack3(const void* w3, void* w3p, void* stream);
 void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* w2, const void* b2, const void* w3p,
-                      const void* b3, void* feat, int B, void* stream);
+                      const void* b3, void* feat, void* c1g, void* c2g, int B,
+                      void* stream);
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     const void* feat, void* h, void* phi_out, int R, int nq,
                     void* stream);
@@ -97,6 +98,41 @@ void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
                        int A, int nq, int mode, float* q_out, int64_t* act_io,
                        void* stream);
 
+
+// ---- hand-written learn-step backward (model_bwd.hip) -------
+void riqn_finish_bwd_dhh(const float* dtheta, const int64_t* actions,
+                         const void* hh, const void* wz, void* dhh, int R,
+                         int A, int nq, void* stream);
+void riqn_zgrad(const float* dtheta, const int64_t* actions, const void* hh,
+                int R, int A, int nq, const void* weps_v, const void* weps_a,
+                const void* beps_v, const void* beps_a, void* dw_mu_v,
+                void* dw_sg_v, void* db_mu_v, void* db_sg_v, void* dw_mu_a,
+                void* dw_sg_a, void* db_mu_a, void* db_sg_a, void* stream);
+void riqn_gemm_dx(const void* dhh, const void* wcat, const void* feat,
+                  const void* phi, void* dphi, float* dfeat, int R, int nq,
+                  void* stream);
+void riqn_gemm_dwcat(const void* dhh, const void* h, const void* weps_v,
+                     const void* weps_a, void* dw_mu_v, void* dw_sg_v,
+                     void* dw_mu_a, void* dw_sg_a, int R, void* stream);
+void riqn_bias_grad_heads(const void* dhh, int R, const void* beps_v,
+                          const void* beps_a, void* db_mu_v, void* db_sg_v,
+                          void* db_mu_a, void* db_sg_a, void* stream);
+void riqn_bias_grad_cols(const void* X, int R, int C, void* db_mu,
+                         void* stream);
+void riqn_tau_dw(const void* dphi, const float* taus, void* dwiqn, int R,
+                 void* stream);
+void riqn_conv_bwd_data(const float* dfeat, const void* feat, const void* c1g,
+                        const void* c2g, const void* w3tp, const void* w2tp,
+                        void* dpre3g, void* dc2g, void* dc1g, float* db1s,
+                        float* db2s, float* db3s, int B, void* stream);
+void riqn_conv_bwd_dw(int layer, const void* dY, const void* act, float* dWs,
+                      int B, int splitk, void* stream);
+void riqn_conv_finalize(float* dW1s, float* dW2s, float* dW3s, float* db1s,
+                        float* db2s, float* db3s, void* dw1, void* dw2,
+                        void* dw3, void* db1, void* db2, void* db3,
+                        void* stream);
+void riqn_pack_bwd_weights(const void* w3, const void* w2, void* w3tp,
+                           void* w2tp, void* stream);
 
 // batched 4-layer NoisyLinear regeneration (rng.hip)
 struct NoisyDesc {

@@ -89,6 +89,7 @@ class GpuLearner:
         # fused no-grad forwards (double-Q select + target quantiles): 4
         # hand-written kernels each instead of the eager module stack
         self.fused_online = self.fused_target = None
+        self.manual_bwd = False
         if self.bf16 and getattr(cfg, "fused_forward", True):
             from .fused_forward import FusedForward
 
@@ -96,6 +97,120 @@ class GpuLearner:
             self.fused_target = FusedForward(self.target, self.device)
             self.a_star = torch.zeros(b, dtype=torch.int64, device=self.device)
             self.theta_p_buf = torch.zeros(b, self.Np, device=self.device)
+            if getattr(cfg, "fused_backward", True):
+                self._init_manual_bwd()
+
+    def _init_manual_bwd(self) -> None:
+        """Buffers for the fully hand-written learn step (model_bwd.hip):
+        no torch.autograd anywhere in the hot loop."""
+        d, bf = self.device, torch.bfloat16
+        B, N, R = self.B, self.N, self.B * self.N
+        z = lambda *s, dt=bf: torch.zeros(*s, dtype=dt, device=d)
+        self.w3tp = z(64, 1024)
+        self.w2tp = z(32, 1024)
+        self.c1g = z(B, 12800)
+        self.c2g = z(B, 5184)
+        self.phi = z(R, 3136)
+        self.dhh = z(R, 1024)
+        self.dphi = z(R, 3136)
+        self.dfeat = z(B, 3136, dt=torch.float32)
+        self.dpre3g = z(64, B * 49)
+        self.dc2g = z(64, B * 81)
+        self.dc1g = z(32, B * 400)
+        self.dW1s = z(32, 256, dt=torch.float32)
+        self.dW2s = z(64, 512, dt=torch.float32)
+        self.dW3s = z(64, 576, dt=torch.float32)
+        self.db1s = z(32, dt=torch.float32)
+        self.db2s = z(64, dt=torch.float32)
+        self.db3s = z(64, dt=torch.float32)
+        self.theta_buf = z(B, N, dt=torch.float32)
+        self.dtheta = z(B, N, dt=torch.float32)
+        self.loss_buf = z(1, dt=torch.float32)
+        self.per_sample = z(B, dt=torch.float32)
+        # grad views (slices of the flat bf16 grad buffer, written in-place)
+        net = self.online
+        g = lambda p: p.grad
+        zv, za = net.fc_z_v, net.fc_z_a
+        hv, ha = net.fc_h_v, net.fc_h_a
+        self._z_eps = [zv.weight_epsilon, za.weight_epsilon,
+                       zv.bias_epsilon, za.bias_epsilon]
+        self._z_grads = [g(zv.weight_mu), g(zv.weight_sigma), g(zv.bias_mu),
+                         g(zv.bias_sigma), g(za.weight_mu), g(za.weight_sigma),
+                         g(za.bias_mu), g(za.bias_sigma)]
+        self._h_wgrads = [g(hv.weight_mu), g(hv.weight_sigma),
+                          g(ha.weight_mu), g(ha.weight_sigma)]
+        self._h_bgrads = [g(hv.bias_mu), g(hv.bias_sigma),
+                          g(ha.bias_mu), g(ha.bias_sigma)]
+        self._h_weps = (hv.weight_epsilon, ha.weight_epsilon)
+        self._h_beps = (hv.bias_epsilon, ha.bias_epsilon)
+        cw = net.convs
+        self._conv_scr = [self.dW1s, self.dW2s, self.dW3s,
+                          self.db1s, self.db2s, self.db3s]
+        self._conv_grads = [g(cw[0].weight).view(32, 256),
+                            g(cw[2].weight).view(64, 512),
+                            g(cw[4].weight).view(64, 576),
+                            g(cw[0].bias), g(cw[2].bias), g(cw[4].bias)]
+        self._iqn_wgrad = g(net.iqn_fc.weight)
+        self._iqn_bgrad = g(net.iqn_fc.bias)
+        self.manual_bwd = True
+
+    def _step_body_manual(self):
+        """One grad step, every op a hand-written HIP kernel (K1-K16)."""
+        cfg = self.cfg
+        ext = self.ext
+        b, N, A = self.B, self.N, self.action_space
+        fo, ft = self.fused_online, self.fused_target
+        ext.rand_uniform(self.taus_online, self.rng_state)
+        ext.rand_uniform(self.taus_target, self.rng_state)
+        ext.rand_uniform(self.taus_select, self.rng_state)
+        self._reset_noise_device(self.online)
+        self._reset_noise_device(self.target)
+        fo.pack_weights()
+        ft.pack_weights()
+        ext.pack_bwd_weights(fo.w3, fo.w2, self.w3tp, self.w2tp)
+        idx, s, a, r, s2, nt, w = self.replay.sample()
+        # ---- grad forward (online net, s, N taus), saving activations ----
+        feat_g, h_g, hh_g = fo._bufs(b, N)
+        ext.conv_encode(s, fo.w1, fo.b1, fo.w2, fo.b2, fo.w3p, fo.b3, feat_g,
+                        self.c1g, self.c2g)
+        ext.tau_embed(self.taus_online.view(-1), fo.wiqn, fo.biqn, feat_g,
+                      h_g, N, phi_out=self.phi)
+        ext.heads_gemm(h_g, fo.wcat, fo.bcat, hh_g)
+        ext.heads_finish(hh_g, fo.wz, fo.bz, A, N, 2,
+                         self.theta_buf.view(-1), a)
+        # ---- double-Q select (online) + target quantiles (no-grad) -------
+        fo.act(s2, self.taus_select.view(-1), self.K, self.a_star)
+        ft.theta(s2, self.taus_target.view(-1), self.Np, self.a_star,
+                 self.theta_p_buf)
+        # ---- fused loss: loss, per-sample priorities, dtheta -------------
+        ext.quantile_huber(self.theta_buf, self.theta_p_buf,
+                           self.taus_online.view(b, N), r, nt, w,
+                           cfg.discount ** cfg.multi_step, cfg.kappa,
+                           self.loss_buf, self.per_sample, self.dtheta)
+        # ---- backward chain ----------------------------------------------
+        dt_flat = self.dtheta.view(-1)
+        ext.finish_bwd_dhh(dt_flat, a, hh_g, fo.wz, self.dhh, A, N)
+        ext.zgrad(dt_flat, a, hh_g, A, N, self._z_eps, self._z_grads)
+        self.dfeat.zero_()
+        ext.gemm_dx(self.dhh, fo.wcat, feat_g, self.phi, self.dphi,
+                    self.dfeat, N)
+        ext.gemm_dwcat(self.dhh, h_g, self._h_weps[0], self._h_weps[1],
+                       self._h_wgrads)
+        ext.bias_grad_heads(self.dhh, self._h_beps[0], self._h_beps[1],
+                            self._h_bgrads)
+        ext.tau_dw(self.dphi, self.taus_online.view(-1), self._iqn_wgrad)
+        ext.bias_grad_cols(self.dphi, self._iqn_bgrad)
+        ext.conv_bwd_data(self.dfeat, feat_g, self.c1g, self.c2g, self.w3tp,
+                          self.w2tp, self.dpre3g, self.dc2g, self.dc1g,
+                          self.db1s, self.db2s, self.db3s)
+        ext.conv_bwd_dw(3, self.dpre3g, self.c2g, self.dW3s, b, 2)
+        ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 3)
+        ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 8)
+        ext.conv_finalize(self._conv_scr, self._conv_grads)
+        # ---- optimizer + priorities --------------------------------------
+        self.opt.step()
+        self.replay.update_priorities(idx, self.per_sample)
+        self.loss_static.copy_(self.loss_buf)
 
     # -- pieces -------------------------------------------------------------
     def _reset_noise_device(self, net: RainbowIQN) -> None:
@@ -112,6 +227,8 @@ class GpuLearner:
         self.ext.noisy_material4(packs[key], self.rng_state)
 
     def _step_body(self):
+        if self.manual_bwd:
+            return self._step_body_manual()
         cfg = self.cfg
         b = self.B
         ext = self.ext

@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "rng.hip"),
         os.path.join(CSRC, "env.hip"),
         os.path.join(CSRC, "model_fwd.hip"),
+        os.path.join(CSRC, "model_bwd.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

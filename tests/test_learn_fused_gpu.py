@@ -1,0 +1,186 @@
+"""Parity: hand-written learn-step backward (model_bwd.hip) against
+torch.autograd on the same bf16 network, weights, noise, taus and batch.
+
+Asserts are per-parameter-group so a failure localizes to one kernel.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from rainbowiqn_amd.config import Config
+
+pytestmark = pytest.mark.gpu
+
+
+def cos_sim(x, y):
+    x, y = x.float().flatten(), y.float().flatten()
+    n = x.norm() * y.norm()
+    if n == 0:
+        return 1.0 if x.norm() == y.norm() else 0.0
+    return float((x @ y) / n)
+
+
+@pytest.fixture(scope="module")
+def setup():
+    from rainbowiqn_amd.models.model import RainbowIQN
+    from rainbowiqn_amd.ops import get_extension
+    from rainbowiqn_amd.runtime.fused_forward import FusedForward
+
+    torch.manual_seed(5)
+    ext = get_extension()
+    cfg = Config(dtype="bf16", device="cuda")
+    A, B, N = 6, 32, 8
+    R = B * N
+    net = RainbowIQN(cfg, A).cuda().to(torch.bfloat16).train()
+    fw = FusedForward(net, "cuda")
+    for layer in (net.fc_h_v, net.fc_h_a, net.fc_z_v, net.fc_z_a):
+        with torch.no_grad():
+            layer.weight_noisy.copy_(
+                layer.weight_mu + layer.weight_sigma * layer.weight_epsilon)
+            layer.bias_noisy.copy_(
+                layer.bias_mu + layer.bias_sigma * layer.bias_epsilon)
+
+    s = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
+    a = torch.randint(0, A, (B,), device="cuda")
+    taus = torch.rand(R, device="cuda")
+    theta_p = torch.randn(B, 8, device="cuda")
+    returns = torch.rand(B, device="cuda")
+    nonterm = (torch.rand(B, device="cuda") > 0.1).float()
+    weights = torch.rand(B, device="cuda") + 0.5
+    gamma_n = cfg.discount ** cfg.multi_step
+
+    # ---- manual chain -----------------------------------------------------
+    d, bf = "cuda", torch.bfloat16
+    c1g = torch.zeros(B, 12800, dtype=bf, device=d)
+    c2g = torch.zeros(B, 5184, dtype=bf, device=d)
+    feat = torch.zeros(B, 3136, dtype=bf, device=d)
+    h = torch.zeros(R, 3136, dtype=bf, device=d)
+    hh = torch.zeros(R, 1024, dtype=bf, device=d)
+    phi = torch.zeros(R, 3136, dtype=bf, device=d)
+    theta = torch.zeros(B, N, device=d)
+    ext.conv_encode(s, fw.w1, fw.b1, fw.w2, fw.b2, fw.w3p, fw.b3, feat,
+                    c1g, c2g)
+    ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, N, phi_out=phi)
+    ext.heads_gemm(h, fw.wcat, fw.bcat, hh)
+    ext.heads_finish(hh, fw.wz, fw.bz, A, N, 2, theta.view(-1), a)
+
+    loss = torch.zeros(1, device=d)
+    per_sample = torch.zeros(B, device=d)
+    dtheta = torch.zeros(B, N, device=d)
+    ext.quantile_huber(theta, theta_p, taus.view(B, N), returns, nonterm,
+                       weights, gamma_n, cfg.kappa, loss, per_sample, dtheta)
+
+    dhh = torch.zeros(R, 1024, dtype=bf, device=d)
+    dphi = torch.zeros(R, 3136, dtype=bf, device=d)
+    dfeat = torch.zeros(B, 3136, device=d)
+    grads = {}
+
+    def gz(*shape):
+        return torch.zeros(*shape, dtype=bf, device=d)
+
+    zv, za = net.fc_z_v, net.fc_z_a
+    hv, ha = net.fc_h_v, net.fc_h_a
+    z_grads = [gz(1, 512), gz(1, 512), gz(1), gz(1),
+               gz(A, 512), gz(A, 512), gz(A), gz(A)]
+    ext.finish_bwd_dhh(dtheta.view(-1), a, hh, fw.wz, dhh, A, N)
+    ext.zgrad(dtheta.view(-1), a, hh, A, N,
+              [zv.weight_epsilon, za.weight_epsilon, zv.bias_epsilon,
+               za.bias_epsilon], z_grads)
+    ext.gemm_dx(dhh, fw.wcat, feat, phi, dphi, dfeat, N)
+    h_wgrads = [gz(512, 3136), gz(512, 3136), gz(512, 3136), gz(512, 3136)]
+    ext.gemm_dwcat(dhh, h, hv.weight_epsilon, ha.weight_epsilon, h_wgrads)
+    h_bgrads = [gz(512), gz(512), gz(512), gz(512)]
+    ext.bias_grad_heads(dhh, hv.bias_epsilon, ha.bias_epsilon, h_bgrads)
+    iqn_wg = gz(3136, 64)
+    iqn_bg = gz(3136)
+    ext.tau_dw(dphi, taus, iqn_wg)
+    ext.bias_grad_cols(dphi, iqn_bg)
+
+    w3tp = gz(64, 1024)
+    w2tp = gz(32, 1024)
+    ext.pack_bwd_weights(fw.w3, fw.w2, w3tp, w2tp)
+    dpre3g = gz(64, B * 49)
+    dc2g = gz(64, B * 81)
+    dc1g = gz(32, B * 400)
+    f32 = lambda *sh: torch.zeros(*sh, device=d)
+    scr = [f32(32, 256), f32(64, 512), f32(64, 576), f32(32), f32(64), f32(64)]
+    ext.conv_bwd_data(dfeat, feat, c1g, c2g, w3tp, w2tp, dpre3g, dc2g, dc1g,
+                      scr[3], scr[4], scr[5])
+    ext.conv_bwd_dw(3, dpre3g, c2g, scr[2], B, 2)
+    ext.conv_bwd_dw(2, dc2g, c1g, scr[1], B, 3)
+    ext.conv_bwd_dw(1, dc1g, s, scr[0], B, 8)
+    conv_g = [gz(32, 256), gz(64, 512), gz(64, 576), gz(32), gz(64), gz(64)]
+    ext.conv_finalize(scr, conv_g)
+    torch.cuda.synchronize()
+
+    manual = {
+        "theta": theta.clone(), "loss": loss.clone(),
+        "zv.w_mu": z_grads[0], "zv.w_sg": z_grads[1],
+        "zv.b_mu": z_grads[2], "zv.b_sg": z_grads[3],
+        "za.w_mu": z_grads[4], "za.w_sg": z_grads[5],
+        "za.b_mu": z_grads[6], "za.b_sg": z_grads[7],
+        "hv.w_mu": h_wgrads[0], "hv.w_sg": h_wgrads[1],
+        "ha.w_mu": h_wgrads[2], "ha.w_sg": h_wgrads[3],
+        "hv.b_mu": h_bgrads[0], "hv.b_sg": h_bgrads[1],
+        "ha.b_mu": h_bgrads[2], "ha.b_sg": h_bgrads[3],
+        "iqn.w": iqn_wg, "iqn.b": iqn_bg,
+        "conv1.w": conv_g[0], "conv2.w": conv_g[1], "conv3.w": conv_g[2],
+        "conv1.b": conv_g[3], "conv2.b": conv_g[4], "conv3.b": conv_g[5],
+    }
+
+    # ---- torch.autograd reference on the same bf16 net ---------------------
+    from rainbowiqn_amd.ops import torch_ref
+
+    for p in net.parameters():
+        p.grad = None
+    z, _ = net(s, N, taus=taus.view(R, 1))
+    th = z.gather(2, a.view(B, 1, 1).expand(B, N, 1)).squeeze(2)
+    t_theta = torch_ref.n_step_target(theta_p, returns, nonterm,
+                                      cfg.discount, cfg.multi_step)
+    loss_ref, _ = torch_ref.quantile_huber_loss(
+        th.float(), t_theta, taus.view(B, N), weights, cfg.kappa)
+    loss_ref.backward()
+    ref = {
+        "theta": th.detach().float(), "loss": loss_ref.detach().view(1),
+        "zv.w_mu": zv.weight_mu.grad, "zv.w_sg": zv.weight_sigma.grad,
+        "zv.b_mu": zv.bias_mu.grad, "zv.b_sg": zv.bias_sigma.grad,
+        "za.w_mu": za.weight_mu.grad, "za.w_sg": za.weight_sigma.grad,
+        "za.b_mu": za.bias_mu.grad, "za.b_sg": za.bias_sigma.grad,
+        "hv.w_mu": hv.weight_mu.grad, "hv.w_sg": hv.weight_sigma.grad,
+        "ha.w_mu": ha.weight_mu.grad, "ha.w_sg": ha.weight_sigma.grad,
+        "hv.b_mu": hv.bias_mu.grad, "hv.b_sg": hv.bias_sigma.grad,
+        "ha.b_mu": ha.bias_mu.grad, "ha.b_sg": ha.bias_sigma.grad,
+        "iqn.w": net.iqn_fc.weight.grad, "iqn.b": net.iqn_fc.bias.grad,
+        "conv1.w": net.convs[0].weight.grad.view(32, 256),
+        "conv2.w": net.convs[2].weight.grad.view(64, 512),
+        "conv3.w": net.convs[4].weight.grad.view(64, 576),
+        "conv1.b": net.convs[0].bias.grad,
+        "conv2.b": net.convs[2].bias.grad,
+        "conv3.b": net.convs[4].bias.grad,
+    }
+    return manual, ref
+
+
+def test_theta_and_loss(setup):
+    manual, ref = setup
+    err = (manual["theta"] - ref["theta"]).abs().max().item()
+    scale = ref["theta"].abs().max().item()
+    assert err < 0.05 * max(scale, 1.0) + 0.02, (err, scale)
+    lerr = abs(float(manual["loss"][0]) - float(ref["loss"][0]))
+    assert lerr < 0.05 * abs(float(ref["loss"][0])) + 0.02, lerr
+
+
+@pytest.mark.parametrize("key", [
+    "zv.w_mu", "zv.w_sg", "zv.b_mu", "za.w_mu", "za.w_sg", "za.b_mu",
+    "hv.w_mu", "hv.w_sg", "ha.w_mu", "ha.w_sg", "hv.b_mu", "ha.b_mu",
+    "iqn.w", "iqn.b", "conv1.w", "conv2.w", "conv3.w",
+    "conv1.b", "conv2.b", "conv3.b",
+])
+def test_grad_parity(setup, key):
+    manual, ref = setup
+    m, r = manual[key].float(), ref[key].float()
+    c = cos_sim(m, r)
+    rel = float((m - r).norm() / (r.norm() + 1e-12))
+    assert c > 0.97, f"{key}: cos {c:.4f} rel {rel:.4f}"
+    assert rel < 0.35, f"{key}: rel {rel:.4f} cos {c:.4f}"
