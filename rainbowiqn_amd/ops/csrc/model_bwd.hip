@@ -373,7 +373,7 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
             for (int u = 0; u < 8; ++u) {
                 float tau = (kk + u < R) ? taus[kk + u] : 0.0f;
                 float i_val = (float)(it * 16 + l15);
-                bv[u] = f32_to_bf16(__cosf(3.14159265358979f * i_val * tau));
+                bv[u] = f32_to_bf16(cosf(3.14159265358979f * i_val * tau));
             }
             acc[it] = mfma16(a, as_bf16x8(bv), acc[it]);
         }

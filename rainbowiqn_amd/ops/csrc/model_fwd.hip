@@ -308,7 +308,7 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
             #pragma unroll
             for (int j = 0; j < 8; ++j) {
                 float k = (float)(ks * 32 + l4 * 8 + j);
-                av[j] = f32_to_bf16(__cosf(3.14159265358979f * k * tau));
+                av[j] = f32_to_bf16(cosf(3.14159265358979f * k * tau));
             }
             afrag[mi][ks] = as_bf16x8(av);
         }
