@@ -155,7 +155,15 @@ class RainbowIQN(nn.Module):
 
     def embed_taus(self, taus: torch.Tensor) -> torch.Tensor:
         """taus: [R, 1] -> [R, 3136] cosine embedding through iqn_fc."""
-        cos = torch.cos(taus.float() * self.cos_basis.float())  # [R, 64]
+        basis = self.cos_basis
+        if basis.dtype != torch.float32:
+            # .to(bf16) on the module must not degrade the fp32 basis: a
+            # 0.5% rounding of pi*i is ~0.5 rad of phase at i=50
+            basis = math.pi * torch.arange(
+                self.quantile_embedding_dim, dtype=torch.float32,
+                device=taus.device)
+            self.cos_basis = basis  # cache fp32 again
+        cos = torch.cos(taus.float() * basis)  # [R, 64]
         return F.relu(self.iqn_fc(cos.to(self.iqn_fc.weight.dtype)))
 
     def head(self, h: torch.Tensor) -> torch.Tensor:
