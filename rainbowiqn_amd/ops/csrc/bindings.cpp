@@ -288,7 +288,8 @@ void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
 
 void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
                torch::Tensor feat, torch::Tensor h, int64_t nq,
-               c10::optional<torch::Tensor> phi_out = c10::nullopt) {
+               c10::optional<torch::Tensor> phi_out = c10::nullopt,
+               c10::optional<torch::Tensor> cosb_out = c10::nullopt) {
     CHECK_T(taus, torch::kFloat32);
     CHECK_BF16(w); CHECK_BF16(bias); CHECK_BF16(feat); CHECK_BF16(h);
     int R = (int)taus.numel();
@@ -296,12 +297,17 @@ void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
     TORCH_CHECK(w.numel() == 3136 * 64);
     TORCH_CHECK(feat.numel() * nq == h.numel(), "feat rows * nq != R");
     void* pp = nullptr;
+    void* cb = nullptr;
     if (phi_out.has_value()) {
         TORCH_CHECK(phi_out->numel() == h.numel());
         pp = phi_out->data_ptr();
     }
+    if (cosb_out.has_value()) {
+        TORCH_CHECK(cosb_out->numel() == (int64_t)R * 64);
+        cb = cosb_out->data_ptr();
+    }
     riqn_tau_embed(taus.data_ptr<float>(), w.data_ptr(), bias.data_ptr(),
-                   feat.data_ptr(), h.data_ptr(), pp, R, (int)nq,
+                   feat.data_ptr(), h.data_ptr(), pp, cb, R, (int)nq,
                    cur_stream());
 }
 
@@ -404,12 +410,13 @@ void bias_grad_cols(torch::Tensor X, torch::Tensor db) {
                         db.data_ptr(), cur_stream());
 }
 
-void tau_dw(torch::Tensor dphi, torch::Tensor taus, torch::Tensor dwiqn) {
-    CHECK_BF16(dphi); CHECK_BF16(dwiqn);
-    CHECK_T(taus, torch::kFloat32);
+void tau_dw(torch::Tensor dphi, torch::Tensor cosb, torch::Tensor dwiqn) {
+    CHECK_BF16(dphi); CHECK_BF16(dwiqn); CHECK_BF16(cosb);
     TORCH_CHECK(dwiqn.numel() == 3136 * 64);
-    riqn_tau_dw(dphi.data_ptr(), taus.data_ptr<float>(), dwiqn.data_ptr(),
-                (int)taus.numel(), cur_stream());
+    int R = (int)cosb.size(0);
+    TORCH_CHECK(R <= 256 && cosb.size(1) == 64, "tau_dw: R <= 256");
+    riqn_tau_dw(dphi.data_ptr(), cosb.data_ptr(), dwiqn.data_ptr(), R,
+                cur_stream());
 }
 
 void conv_bwd_data(torch::Tensor dfeat, torch::Tensor feat, torch::Tensor c1g,
@@ -478,7 +485,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("stack_gather", &stack_gather);
     m.def("conv_pack3", &conv_pack3);
     m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
-    m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt);
+    m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt);
     m.def("heads_gemm", &heads_gemm);
     m.def("heads_finish", &heads_finish);
     m.def("finish_bwd_dhh", &finish_bwd_dhh);

@@ -100,23 +100,39 @@ __global__ __launch_bounds__(256) void zgrad_kernel(
     unsigned short* __restrict__ db_mu_v, unsigned short* __restrict__ db_sg_v,
     unsigned short* __restrict__ dw_mu_a, unsigned short* __restrict__ dw_sg_a,
     unsigned short* __restrict__ db_mu_a, unsigned short* __restrict__ db_sg_a) {
+    __shared__ float part[4][512];
+    __shared__ float bpart[4];
     const int row = blockIdx.x;  // 0 = V, 1..A = advantage rows
     const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;
     const float invA = 1.0f / (float)A;
-    float acc[2] = {0.0f, 0.0f};  // thread covers cols t*2, t*2+1
-    float bacc = 0.0f;
     const int off = (row == 0) ? 0 : 512;
-    for (int r = 0; r < R; ++r) {
+    // each wave accumulates rows wave::4, lane covers 8 coalesced columns
+    float wacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float wb = 0.0f;
+    for (int r = wave; r < R; r += 4) {
         float c = dtheta[r];
         if (row > 0) {
             int a = (int)actions[r / nq];
             c *= ((a == row - 1) ? 1.0f : 0.0f) - invA;
         }
-        const unsigned short* hrow = hh + (int64_t)r * 1024 + off;
-        acc[0] += c * bf(hrow[t * 2]);
-        acc[1] += c * bf(hrow[t * 2 + 1]);
-        bacc += c;
+        const unsigned short* hrow = hh + (int64_t)r * 1024 + off + lane * 8;
+        u16x8 hv8 = *(const u16x8*)hrow;
+        #pragma unroll
+        for (int u = 0; u < 8; ++u) wacc[u] += c * bf(hv8[u]);
+        wb += c;
     }
+    #pragma unroll
+    for (int u = 0; u < 8; ++u) part[wave][lane * 8 + u] = wacc[u];
+    if (lane == 0) bpart[wave] = wb;
+    __syncthreads();
+    float acc[2];
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+        int j = t * 2 + u;
+        acc[u] = part[0][j] + part[1][j] + part[2][j] + part[3][j];
+    }
+    float bacc = bpart[0] + bpart[1] + bpart[2] + bpart[3];
     if (row == 0) {
         #pragma unroll
         for (int u = 0; u < 2; ++u) {
@@ -312,12 +328,25 @@ __global__ void bias_grad_cols_kernel(
     const unsigned short* __restrict__ beps,  // [C] or nullptr
     unsigned short* __restrict__ db_mu,       // [C]
     unsigned short* __restrict__ db_sg) {     // [C] or nullptr
-    int c = threadIdx.x + blockIdx.x * blockDim.x;
-    if (c >= C) return;
+    // WG covers 64 columns; thread (c_local, row-partition) layout so the
+    // 4 waves split the serial row loop 4-ways; LDS-combined.
+    __shared__ float part[64];
+    const int t = threadIdx.x;
+    const int cl = t & 63, rp = t >> 6;
+    const int c = blockIdx.x * 64 + cl;
     float s = 0.0f;
-    for (int r = 0; r < R; ++r) s += bf(X[(int64_t)r * C + c]);
-    db_mu[c] = f32_to_bf16(s);
-    if (db_sg != nullptr) db_sg[c] = f32_to_bf16(s * bf(beps[c]));
+    if (c < C) {
+        for (int r = rp; r < R; r += 4) s += bf(X[(int64_t)r * C + c]);
+    }
+    if (t < 64) part[t] = 0.0f;
+    __syncthreads();
+    atomicAdd(&part[cl], s);
+    __syncthreads();
+    if (t < 64 && c < C) {
+        float v = part[t];
+        db_mu[c] = f32_to_bf16(v);
+        if (db_sg != nullptr) db_sg[c] = f32_to_bf16(v * bf(beps[c]));
+    }
 }
 
 // split-column variant for the concat dhh buffer: cols 0..511 -> layer V,
@@ -328,10 +357,18 @@ __global__ void bias_grad_heads_kernel(
     const unsigned short* __restrict__ beps_a,
     unsigned short* __restrict__ db_mu_v, unsigned short* __restrict__ db_sg_v,
     unsigned short* __restrict__ db_mu_a, unsigned short* __restrict__ db_sg_a) {
-    int c = threadIdx.x + blockIdx.x * blockDim.x;
-    if (c >= 1024) return;
+    __shared__ float part[64];
+    const int t = threadIdx.x;
+    const int cl = t & 63, rp = t >> 6;
+    const int c = blockIdx.x * 64 + cl;
     float s = 0.0f;
-    for (int r = 0; r < R; ++r) s += bf(dhh[(int64_t)r * 1024 + c]);
+    for (int r = rp; r < R; r += 4) s += bf(dhh[(int64_t)r * 1024 + c]);
+    if (t < 64) part[t] = 0.0f;
+    __syncthreads();
+    atomicAdd(&part[cl], s);
+    __syncthreads();
+    if (t >= 64) return;
+    s = part[t];
     if (c < 512) {
         db_mu_v[c] = f32_to_bf16(s);
         db_sg_v[c] = f32_to_bf16(s * bf(beps_v[c]));
@@ -348,14 +385,18 @@ __global__ void bias_grad_heads_kernel(
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void tau_dw_kernel(
     const unsigned short* __restrict__ dphi,  // [R,3136]
-    const float* __restrict__ taus,           // [R]
+    const unsigned short* __restrict__ cosb,  // [R,64] bf16 (from tau_embed)
     unsigned short* __restrict__ dwiqn,       // [3136,64] bf16 out
     int R) {
+    __shared__ unsigned short cs[256 * 64];  // staged cos table (R <= 256)
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int l15 = lane & 15, l4 = lane >> 4;
     const int mt = blockIdx.x * 4 + wave;  // n-tile per wave
     const int n = mt * 16 + l15;
+    for (int i = t * 4; i < R * 64; i += 256 * 4)
+        *(u16x4*)&cs[i] = *(const u16x4*)(cosb + i);
+    __syncthreads();
 
     f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
                     f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
@@ -368,13 +409,10 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
         bf16x8 a = as_bf16x8(av);
         #pragma unroll
         for (int it = 0; it < 4; ++it) {  // i-tiles: 64 cols
-            u16x8 bv;
+            u16x8 bv = {0, 0, 0, 0, 0, 0, 0, 0};
             #pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                float tau = (kk + u < R) ? taus[kk + u] : 0.0f;
-                float i_val = (float)(it * 16 + l15);
-                bv[u] = f32_to_bf16(cosf(3.14159265358979f * i_val * tau));
-            }
+            for (int u = 0; u < 8; ++u)
+                if (kk + u < R) bv[u] = cs[(kk + u) * 64 + it * 16 + l15];
             acc[it] = mfma16(a, as_bf16x8(bv), acc[it]);
         }
     }
@@ -457,8 +495,21 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
         const int m = mt * 16 + l15;
         const int iy = m / 9, ix = m % 9;
         const bool mv = m < 81;
+        bf16x8 bbuf[2][2];
+        #pragma unroll
+        for (int nt = 0; nt < 2; ++nt)
+            bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
+                w3tp + ((nh * 2 + nt) * 16 + l15) * 1024 + l4 * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
+            const int cur = ks & 1;
+            if (ks + 1 < 32) {
+                const int k1 = (ks + 1) * 32 + l4 * 8;
+                #pragma unroll
+                for (int nt = 0; nt < 2; ++nt)
+                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                        w3tp + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
+            }
             const int k0 = ks * 32 + l4 * 8;
             u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
             if (mv) {
@@ -474,13 +525,8 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
                 }
             }
             bf16x8 a = as_bf16x8(av);
-            #pragma unroll
-            for (int nt = 0; nt < 2; ++nt) {
-                const int c = (nh * 2 + nt) * 16 + l15;
-                bf16x8 bfr =
-                    as_bf16x8(*(const u16x8*)(w3tp + c * 1024 + k0));
-                acc[nt] = mfma16(a, bfr, acc[nt]);
-            }
+            acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
+            acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
         }
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt) {
@@ -514,8 +560,21 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
         const int m = mt * 16 + l15;
         const int iy = m / 20, ix = m % 20;
         const bool mv = m < 400;
+        bf16x8 bbuf[2][2];
+        #pragma unroll
+        for (int nt = 0; nt < 2; ++nt)
+            bbuf[0][nt] = as_bf16x8(
+                *(const u16x8*)(w2tp + (nt * 16 + l15) * 1024 + l4 * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
+            const int cur = ks & 1;
+            if (ks + 1 < 32) {
+                const int k1 = (ks + 1) * 32 + l4 * 8;
+                #pragma unroll
+                for (int nt = 0; nt < 2; ++nt)
+                    bbuf[cur ^ 1][nt] = as_bf16x8(
+                        *(const u16x8*)(w2tp + (nt * 16 + l15) * 1024 + k1));
+            }
             const int k0 = ks * 32 + l4 * 8;
             u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
             if (mv) {
@@ -532,13 +591,8 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
                 }
             }
             bf16x8 a = as_bf16x8(av);
-            #pragma unroll
-            for (int nt = 0; nt < 2; ++nt) {
-                const int c = nt * 16 + l15;
-                bf16x8 bfr =
-                    as_bf16x8(*(const u16x8*)(w2tp + c * 1024 + k0));
-                acc[nt] = mfma16(a, bfr, acc[nt]);
-            }
+            acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
+            acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
         }
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt) {
@@ -734,7 +788,7 @@ void riqn_gemm_dwcat(const void* dhh, const void* h, const void* weps_v,
 void riqn_bias_grad_heads(const void* dhh, int R, const void* beps_v,
                           const void* beps_a, void* db_mu_v, void* db_sg_v,
                           void* db_mu_a, void* db_sg_a, void* stream) {
-    hipLaunchKernelGGL(bias_grad_heads_kernel, dim3(4), dim3(256), 0,
+    hipLaunchKernelGGL(bias_grad_heads_kernel, dim3(16), dim3(256), 0,
                        (hipStream_t)stream, (const unsigned short*)dhh, R,
                        (const unsigned short*)beps_v,
                        (const unsigned short*)beps_a,
@@ -744,17 +798,17 @@ void riqn_bias_grad_heads(const void* dhh, int R, const void* beps_v,
 
 void riqn_bias_grad_cols(const void* X, int R, int C, void* db_mu,
                          void* stream) {
-    hipLaunchKernelGGL(bias_grad_cols_kernel, dim3((C + 255) / 256), dim3(256),
+    hipLaunchKernelGGL(bias_grad_cols_kernel, dim3((C + 63) / 64), dim3(256),
                        0, (hipStream_t)stream, (const unsigned short*)X, R, C,
                        (const unsigned short*)nullptr, (unsigned short*)db_mu,
                        (unsigned short*)nullptr);
 }
 
-void riqn_tau_dw(const void* dphi, const float* taus, void* dwiqn, int R,
+void riqn_tau_dw(const void* dphi, const void* cosb, void* dwiqn, int R,
                  void* stream) {
     hipLaunchKernelGGL(tau_dw_kernel, dim3(3136 / 64), dim3(256), 0,
-                       (hipStream_t)stream, (const unsigned short*)dphi, taus,
-                       (unsigned short*)dwiqn, R);
+                       (hipStream_t)stream, (const unsigned short*)dphi,
+                       (const unsigned short*)cosb, (unsigned short*)dwiqn, R);
 }
 
 void riqn_conv_bwd_data(const float* dfeat, const void* feat, const void* c1g,

@@ -276,6 +276,7 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
     const unsigned short* __restrict__ feat,// [B,3136] bf16
     unsigned short* __restrict__ h,         // [R,3136] bf16
     unsigned short* __restrict__ phi_out,   // [R,3136] bf16 or nullptr
+    unsigned short* __restrict__ cosb_out,  // [R,64] bf16 or nullptr (bwd)
     int R, int nq) {
     __shared__ unsigned short tile[2 * 16 * 784];  // two m-tiles per WG
     __shared__ unsigned short ftile[4 * 784];
@@ -310,6 +311,8 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
                 float k = (float)(ks * 32 + l4 * 8 + j);
                 av[j] = f32_to_bf16(cosf(3.14159265358979f * k * tau));
             }
+            if (cosb_out != nullptr && blockIdx.y == 0 && r_a < R)
+                *(u16x8*)(cosb_out + (int64_t)r_a * 64 + ks * 32 + l4 * 8) = av;
             afrag[mi][ks] = as_bf16x8(av);
         }
     }
@@ -591,14 +594,15 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
 }
 
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
-                    const void* feat, void* h, void* phi_out, int R, int nq,
-                    void* stream) {
+                    const void* feat, void* h, void* phi_out, void* cosb_out,
+                    int R, int nq, void* stream) {
     dim3 grid((R + 31) / 32, 4);
     hipLaunchKernelGGL(tau_embed_kernel, grid, dim3(256), 0,
                        (hipStream_t)stream, taus, (const unsigned short*)w,
                        (const unsigned short*)bias,
                        (const unsigned short*)feat, (unsigned short*)h,
-                       (unsigned short*)phi_out, R, nq);
+                       (unsigned short*)phi_out, (unsigned short*)cosb_out,
+                       R, nq);
 }
 
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,

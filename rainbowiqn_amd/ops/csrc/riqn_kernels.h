@@ -90,8 +90,8 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* b3, void* feat, void* c1g, void* c2g, int B,
                       void* stream);
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
-                    const void* feat, void* h, void* phi_out, int R, int nq,
-                    void* stream);
+                    const void* feat, void* h, void* phi_out, void* cosb_out,
+                    int R, int nq, void* stream);
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
                      void* out, int R, void* stream);
 void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
@@ -119,7 +119,7 @@ void riqn_bias_grad_heads(const void* dhh, int R, const void* beps_v,
                           void* db_mu_a, void* db_sg_a, void* stream);
 void riqn_bias_grad_cols(const void* X, int R, int C, void* db_mu,
                          void* stream);
-void riqn_tau_dw(const void* dphi, const float* taus, void* dwiqn, int R,
+void riqn_tau_dw(const void* dphi, const void* cosb, void* dwiqn, int R,
                  void* stream);
 void riqn_conv_bwd_data(const float* dfeat, const void* feat, const void* c1g,
                         const void* c2g, const void* w3tp, const void* w2tp,

@@ -63,8 +63,12 @@ class GpuReplay:
         # static output buffers (graph-capture targets)
         self.out_idx = torch.zeros(b, dtype=torch.int32, device=d)
         # bf16 output feeds the native-bf16 model path with no cast kernels
-        self.out_states = torch.zeros(b, self.history, 84, 84, dtype=out_dtype, device=d)
-        self.out_next_states = torch.zeros(b, self.history, 84, 84, dtype=out_dtype, device=d)
+        # states and next_states are halves of ONE contiguous buffer so the
+        # learner can run a single conv_encode over [s; s'] (B -> 2B batch)
+        self.states_cat = torch.zeros(2 * b, self.history, 84, 84,
+                                      dtype=out_dtype, device=d)
+        self.out_states = self.states_cat[:b]
+        self.out_next_states = self.states_cat[b:]
         self.out_actions = torch.zeros(b, dtype=torch.int64, device=d)
         self.out_returns = torch.zeros(b, device=d)
         self.out_nonterminals = torch.zeros(b, device=d)

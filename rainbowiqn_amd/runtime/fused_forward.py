@@ -124,6 +124,20 @@ class FusedForward:
                               actions)
         return actions
 
+    def act_from_feat(self, featv: torch.Tensor, taus, nq: int,
+                      actions: torch.Tensor) -> torch.Tensor:
+        """Greedy action from precomputed conv features (lets the learner
+        batch s and s' through one conv_encode call)."""
+        B = featv.shape[0]
+        _, h, hh = self._bufs(B, nq)
+        self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn,
+                           featv.contiguous(), h, nq)
+        self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
+        dummy = torch.zeros(1, dtype=torch.float32, device=self.device)
+        self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 1, dummy,
+                              actions)
+        return actions
+
     def theta(self, x, taus, nq: int, actions: torch.Tensor,
               out: torch.Tensor) -> torch.Tensor:
         """Z_tau(x, actions[b]) -> out [B, nq] (target-net path, K10 feed)."""

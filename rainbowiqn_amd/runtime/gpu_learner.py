@@ -108,12 +108,14 @@ class GpuLearner:
         z = lambda *s, dt=bf: torch.zeros(*s, dtype=dt, device=d)
         self.w3tp = z(64, 1024)
         self.w2tp = z(32, 1024)
-        self.c1g = z(B, 12800)
-        self.c2g = z(B, 5184)
+        self.c1g = z(2 * B, 12800)  # [s; s'] batch; backward uses [:B]
+        self.c2g = z(2 * B, 5184)
+        self.feat_cat = z(2 * B, 3136)
         self.phi = z(R, 3136)
         self.dhh = z(R, 1024)
         self.dphi = z(R, 3136)
         self.dfeat = z(B, 3136, dt=torch.float32)
+        self.cosb = z(R, 64)
         self.dpre3g = z(64, B * 49)
         self.dc2g = z(64, B * 81)
         self.dc1g = z(32, B * 400)
@@ -169,17 +171,20 @@ class GpuLearner:
         ft.pack_weights()
         ext.pack_bwd_weights(fo.w3, fo.w2, self.w3tp, self.w2tp)
         idx, s, a, r, s2, nt, w = self.replay.sample()
-        # ---- grad forward (online net, s, N taus), saving activations ----
-        feat_g, h_g, hh_g = fo._bufs(b, N)
-        ext.conv_encode(s, fo.w1, fo.b1, fo.w2, fo.b2, fo.w3p, fo.b3, feat_g,
-                        self.c1g, self.c2g)
+        # ---- online conv over [s; s'] in ONE call (activations saved for
+        # the backward; the s' half feeds the double-Q select) -------------
+        _, h_g, hh_g = fo._bufs(b, N)
+        ext.conv_encode(self.replay.states_cat, fo.w1, fo.b1, fo.w2, fo.b2,
+                        fo.w3p, fo.b3, self.feat_cat, self.c1g, self.c2g)
+        feat_g = self.feat_cat[:b]
         ext.tau_embed(self.taus_online.view(-1), fo.wiqn, fo.biqn, feat_g,
-                      h_g, N, phi_out=self.phi)
+                      h_g, N, phi_out=self.phi, cosb_out=self.cosb)
         ext.heads_gemm(h_g, fo.wcat, fo.bcat, hh_g)
         ext.heads_finish(hh_g, fo.wz, fo.bz, A, N, 2,
                          self.theta_buf.view(-1), a)
-        # ---- double-Q select (online) + target quantiles (no-grad) -------
-        fo.act(s2, self.taus_select.view(-1), self.K, self.a_star)
+        # ---- double-Q select (online, from the s' conv half) + target ----
+        fo.act_from_feat(self.feat_cat[b:], self.taus_select.view(-1),
+                         self.K, self.a_star)
         ft.theta(s2, self.taus_target.view(-1), self.Np, self.a_star,
                  self.theta_p_buf)
         # ---- fused loss: loss, per-sample priorities, dtheta -------------
@@ -198,14 +203,14 @@ class GpuLearner:
                        self._h_wgrads)
         ext.bias_grad_heads(self.dhh, self._h_beps[0], self._h_beps[1],
                             self._h_bgrads)
-        ext.tau_dw(self.dphi, self.taus_online.view(-1), self._iqn_wgrad)
+        ext.tau_dw(self.dphi, self.cosb, self._iqn_wgrad)
         ext.bias_grad_cols(self.dphi, self._iqn_bgrad)
         ext.conv_bwd_data(self.dfeat, feat_g, self.c1g, self.c2g, self.w3tp,
                           self.w2tp, self.dpre3g, self.dc2g, self.dc1g,
                           self.db1s, self.db2s, self.db3s)
-        ext.conv_bwd_dw(3, self.dpre3g, self.c2g, self.dW3s, b, 2)
-        ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 3)
-        ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 8)
+        ext.conv_bwd_dw(3, self.dpre3g, self.c2g, self.dW3s, b, 8)
+        ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 12)
+        ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 32)
         ext.conv_finalize(self._conv_scr, self._conv_grads)
         # ---- optimizer + priorities --------------------------------------
         self.opt.step()

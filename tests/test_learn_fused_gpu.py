@@ -61,7 +61,9 @@ def setup():
     theta = torch.zeros(B, N, device=d)
     ext.conv_encode(s, fw.w1, fw.b1, fw.w2, fw.b2, fw.w3p, fw.b3, feat,
                     c1g, c2g)
-    ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, N, phi_out=phi)
+    cosb = torch.zeros(R, 64, dtype=bf, device=d)
+    ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, N, phi_out=phi,
+                  cosb_out=cosb)
     ext.heads_gemm(h, fw.wcat, fw.bcat, hh)
     ext.heads_finish(hh, fw.wz, fw.bz, A, N, 2, theta.view(-1), a)
 
@@ -94,7 +96,7 @@ def setup():
     ext.bias_grad_heads(dhh, hv.bias_epsilon, ha.bias_epsilon, h_bgrads)
     iqn_wg = gz(3136, 64)
     iqn_bg = gz(3136)
-    ext.tau_dw(dphi, taus, iqn_wg)
+    ext.tau_dw(dphi, cosb, iqn_wg)
     ext.bias_grad_cols(dphi, iqn_bg)
 
     w3tp = gz(64, 1024)
@@ -107,9 +109,9 @@ def setup():
     scr = [f32(32, 256), f32(64, 512), f32(64, 576), f32(32), f32(64), f32(64)]
     ext.conv_bwd_data(dfeat, feat, c1g, c2g, w3tp, w2tp, dpre3g, dc2g, dc1g,
                       scr[3], scr[4], scr[5])
-    ext.conv_bwd_dw(3, dpre3g, c2g, scr[2], B, 2)
-    ext.conv_bwd_dw(2, dc2g, c1g, scr[1], B, 3)
-    ext.conv_bwd_dw(1, dc1g, s, scr[0], B, 8)
+    ext.conv_bwd_dw(3, dpre3g, c2g, scr[2], B, 8)
+    ext.conv_bwd_dw(2, dc2g, c1g, scr[1], B, 12)
+    ext.conv_bwd_dw(1, dc1g, s, scr[0], B, 32)
     conv_g = [gz(32, 256), gz(64, 512), gz(64, 576), gz(32), gz(64), gz(64)]
     ext.conv_finalize(scr, conv_g)
     torch.cuda.synchronize()
