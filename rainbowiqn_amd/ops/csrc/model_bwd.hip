@@ -187,6 +187,7 @@ __global__ __launch_bounds__(256) void gemm_dx_kernel(
                        {{0, 0, 0, 0}, {0, 0, 0, 0}}};
     for (int k0 = 0; k0 < 1024; k0 += DX_BK) {
         // stage B = Wcat[k0..+64][bn..+64] as [k][n] (rows coalesced)
+        #pragma unroll 2
         for (int i = t * 8; i < DX_BK * 64; i += 256 * 8) {
             int kk = i / 64, nn = i % 64;
             *(u16x8*)&Bs[kk * 64 + nn] =
@@ -264,6 +265,7 @@ __global__ __launch_bounds__(256) void gemm_dwcat_kernel(
                        {{0, 0, 0, 0}, {0, 0, 0, 0}}};
     for (int k0 = 0; k0 < R; k0 += 32) {
         // stage A^T: As[k][m] = dhh[k0+k][bm+m], k<32, m<64 (rows coalesced)
+        // (single iteration for 256 threads; kept as loop for clarity)
         for (int i = t * 8; i < 32 * 64; i += 256 * 8) {
             int kk = i / 64, mm = i % 64;
             u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -394,6 +396,7 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
     const int l15 = lane & 15, l4 = lane >> 4;
     const int mt = blockIdx.x * 4 + wave;  // n-tile per wave
     const int n = mt * 16 + l15;
+    #pragma unroll 8
     for (int i = t * 4; i < R * 64; i += 256 * 4)
         *(u16x4*)&cs[i] = *(const u16x4*)(cosb + i);
     __syncthreads();
@@ -465,10 +468,13 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
     const int l15 = lane & 15, l4 = lane >> 4;
 
     // stage activations + dpre3
+    #pragma unroll 8
     for (int i = t * 4; i < BW_C1; i += 512 * 4)
         *(u16x4*)&c1s[i] = *(const u16x4*)(c1g + (int64_t)b * BW_C1 + i);
+    #pragma unroll 4
     for (int i = t * 4; i < BW_C2; i += 512 * 4)
         *(u16x4*)&c2s[i] = *(const u16x4*)(c2g + (int64_t)b * BW_C2 + i);
+    #pragma unroll 4
     for (int i = t; i < BW_P3; i += 512) {
         float g = dfeat[(int64_t)b * 3136 + i];
         float m = bf(feat[(int64_t)b * 3136 + i]) > 0.0f ? 1.0f : 0.0f;

@@ -77,11 +77,23 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
 
     // ---- stage input state into LDS (row stride 88) ----------------------
     {
+        // two-phase: issue ALL loads, then all LDS stores -- a fused
+        // load->store loop serializes at one HBM latency per iteration
         const unsigned short* src = x + (int64_t)b * (4 * 84 * 84);
-        for (int e = t * 4; e < 4 * 84 * 84; e += 512 * 4) {
-            u16x4 v = *(const u16x4*)(src + e);
-            int c = e / 7056, r = (e % 7056) / 84, cc = e % 84;
-            *(u16x4*)&lds[LDS_IN + (c * 84 + r) * IN_STRIDE + cc] = v;
+        u16x4 tmp[14];
+        #pragma unroll
+        for (int i = 0; i < 14; ++i) {
+            int e = t * 4 + i * 2048;
+            tmp[i] = (e < 4 * 84 * 84) ? *(const u16x4*)(src + e)
+                                       : u16x4{0, 0, 0, 0};
+        }
+        #pragma unroll
+        for (int i = 0; i < 14; ++i) {
+            int e = t * 4 + i * 2048;
+            if (e < 4 * 84 * 84) {
+                int c = e / 7056, r = (e % 7056) / 84, cc = e % 84;
+                *(u16x4*)&lds[LDS_IN + (c * 84 + r) * IN_STRIDE + cc] = tmp[i];
+            }
         }
     }
     __syncthreads();
@@ -291,6 +303,7 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
     // stage the feat rows of the states this tile touches (<= 4 for nq >= 8)
     const int s0 = (mt0 * 16) / nq;
     const int span = min((mt0 * 16 + 31) / nq - s0 + 1, 4);
+    #pragma unroll 4
     for (int i = t * 4; i < span * 784; i += 256 * 4) {
         int si = i / 784, cc = i % 784;
         int64_t src = (int64_t)min(s0 + si, (R - 1) / nq) * 3136 + nt0 * 16 + cc;
@@ -346,6 +359,7 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
     }
     __syncthreads();
     // coalesced row stores: 32 rows x 784 cols
+    #pragma unroll 8
     for (int i = t * 4; i < 32 * 784; i += 256 * 4) {
         int rr = i / 784, cc = i % 784;
         int r = mt0 * 16 + rr;
@@ -476,6 +490,7 @@ __global__ __launch_bounds__(256) void heads_finish_kernel(
     __shared__ float sv[4][20];
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
+    #pragma unroll 8
     for (int i = t * 4; i < (1 + A) * 512; i += 256 * 4)
         *(u16x4*)&wzs[i] = *(const u16x4*)(wz + i);
     __syncthreads();
