@@ -517,18 +517,17 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
                         w3tp + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
             }
             const int k0 = ks * 32 + l4 * 8;
-            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (mv) {
-                #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int k = k0 + j;
-                    const int n = k >> 4, tt = k & 15;
-                    const int ky = tt >> 2, kx = tt & 3;
-                    const int oy = iy - ky, ox = ix - kx;
-                    if (ky < 3 && kx < 3 && oy >= 0 && oy < 7 && ox >= 0 &&
-                        ox < 7)
-                        av[j] = p3s[n * 49 + oy * 7 + ox];
-                }
+            u16x8 av;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int k = k0 + j;
+                const int n = k >> 4, tt = k & 15;
+                const int ky = tt >> 2, kx = tt & 3;
+                const int oy = iy - ky, ox = ix - kx;
+                const bool ok = mv && ky < 3 && kx < 3 && oy >= 0 && oy < 7 &&
+                                ox >= 0 && ox < 7;
+                unsigned short v = p3s[n * 49 + (ok ? oy * 7 + ox : 0)];
+                av[j] = ok ? v : (unsigned short)0;
             }
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
@@ -582,19 +581,18 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
                         *(const u16x8*)(w2tp + (nt * 16 + l15) * 1024 + k1));
             }
             const int k0 = ks * 32 + l4 * 8;
-            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (mv) {
-                #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int k = k0 + j;
-                    const int n = k >> 4, tt = k & 15;
-                    const int ky = tt >> 2, kx = tt & 3;
-                    const int ry = iy - ky, rx = ix - kx;
-                    if (ry >= 0 && rx >= 0 && (ry & 1) == 0 && (rx & 1) == 0) {
-                        const int oy = ry >> 1, ox = rx >> 1;
-                        if (oy < 9 && ox < 9) av[j] = d2s[n * 81 + oy * 9 + ox];
-                    }
-                }
+            u16x8 av;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int k = k0 + j;
+                const int n = k >> 4, tt = k & 15;
+                const int ky = tt >> 2, kx = tt & 3;
+                const int ry = iy - ky, rx = ix - kx;
+                const int oy = ry >> 1, ox = rx >> 1;
+                const bool ok = mv && ry >= 0 && rx >= 0 && (ry & 1) == 0 &&
+                                (rx & 1) == 0 && oy < 9 && ox < 9;
+                unsigned short v = d2s[n * 81 + (ok ? oy * 9 + ox : 0)];
+                av[j] = ok ? v : (unsigned short)0;
             }
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
@@ -676,16 +674,19 @@ __global__ __launch_bounds__(256) void conv_bwd_dw_kernel(
                 if (kk + u < kend) av[u] = dY[(int64_t)mrow * K + kk + u];
         }
         // B-frag: im2col tap for 8 (state,px) values, fixed tap n
-        u16x8 bv = {0, 0, 0, 0, 0, 0, 0, 0};
+        u16x8 bv;
         #pragma unroll
         for (int u = 0; u < 8; ++u) {
             const int k = kk + u;
-            if (k < kend && n < N) {
-                const int st = k / PX, px = k % PX;
-                const int oy = px / OW, ox = px % OW;
-                const int iy = oy * STR + ky, ix = ox * STR + kx;
-                bv[u] = act[((int64_t)st * CIN + c) * (IW * IW) + iy * IW + ix];
-            }
+            const bool ok = k < kend && n < N;
+            const int kc = ok ? k : kbeg;
+            const int st = kc / PX, px = kc % PX;
+            const int oy = px / OW, ox = px % OW;
+            const int iy = oy * STR + ky, ix = ox * STR + kx;
+            unsigned short v =
+                act[((int64_t)st * CIN + min(c, CIN - 1)) * (IW * IW) +
+                    iy * IW + ix];
+            bv[u] = ok ? v : (unsigned short)0;
         }
         acc = mfma16(as_bf16x8(av), as_bf16x8(bv), acc);
     }

@@ -177,10 +177,12 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
             const int k0 = ks * 32 + l4 * 8;
             const int c = k0 >> 4, t16 = k0 & 15;  // t16 in {0,8}
             const int ky0 = t16 >> 2;
-            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (mv) {
-                const int base =
-                    LDS_C1 + (c * 20 + oy * 2 + ky0) * C1_STRIDE + ox * 2;
+            // branchless gather: clamp the base for invalid rows (their
+            // MFMA output is discarded) so the LDS reads are unconditional
+            u16x8 av;
+            {
+                const int base = mv ? LDS_C1 +
+                    (c * 20 + oy * 2 + ky0) * C1_STRIDE + ox * 2 : LDS_C1;
                 #pragma unroll
                 for (int j = 0; j < 4; ++j) av[j] = lds[base + j];
                 #pragma unroll
@@ -233,15 +235,16 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
             }
             const int k0 = ks * 32 + l4 * 8;
             const int c = k0 >> 4, t16 = k0 & 15;
-            u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (mv) {
-                #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
-                    if (ky < 3 && kx < 3)
-                        av[j] = lds[LDS_C2 + (c * 9 + oy + ky) * C2_STRIDE +
-                                    ox + kx];
-                }
+            u16x8 av;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+                const bool ok = mv && ky < 3 && kx < 3;
+                const int idx =
+                    LDS_C2 + (c * 9 + (ok ? oy + ky : 0)) * C2_STRIDE +
+                    (ok ? ox + kx : 0);
+                unsigned short v = lds[idx];  // unconditional (clamped) read
+                av[j] = ok ? v : (unsigned short)0;
             }
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
