@@ -501,6 +501,23 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
         const int m = mt * 16 + l15;
         const int iy = m / 9, ix = m % 9;
         const bool mv = m < 81;
+        // tap geometry is fixed per lane; only the n-plane advances (+2 per
+        // k-step): precomputed incremental gather addresses + zero masks
+        int ga[8];
+        unsigned short okm[8];
+        {
+            const int n0 = l4 >> 1;
+            const int t16 = (l4 & 1) * 8;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+                const int oy = iy - ky, ox = ix - kx;
+                const bool ok = mv && ky < 3 && kx < 3 && oy >= 0 && oy < 7 &&
+                                ox >= 0 && ox < 7;
+                ga[j] = n0 * 49 + (ok ? oy * 7 + ox : 0);
+                okm[j] = ok ? 0xFFFFu : 0;
+            }
+        }
         bf16x8 bbuf[2][2];
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
@@ -516,19 +533,11 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
                     bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
                         w3tp + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
             }
-            const int k0 = ks * 32 + l4 * 8;
             u16x8 av;
             #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int k = k0 + j;
-                const int n = k >> 4, tt = k & 15;
-                const int ky = tt >> 2, kx = tt & 3;
-                const int oy = iy - ky, ox = ix - kx;
-                const bool ok = mv && ky < 3 && kx < 3 && oy >= 0 && oy < 7 &&
-                                ox >= 0 && ox < 7;
-                unsigned short v = p3s[n * 49 + (ok ? oy * 7 + ox : 0)];
-                av[j] = ok ? v : (unsigned short)0;
-            }
+            for (int j = 0; j < 8; ++j) av[j] = p3s[ga[j]] & okm[j];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) ga[j] += 2 * 49;
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
             acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
@@ -565,6 +574,22 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
         const int m = mt * 16 + l15;
         const int iy = m / 20, ix = m % 20;
         const bool mv = m < 400;
+        int ga[8];
+        unsigned short okm[8];
+        {
+            const int n0 = l4 >> 1;
+            const int t16 = (l4 & 1) * 8;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+                const int ry = iy - ky, rx = ix - kx;
+                const int oy = ry >> 1, ox = rx >> 1;
+                const bool ok = mv && ry >= 0 && rx >= 0 && (ry & 1) == 0 &&
+                                (rx & 1) == 0 && oy < 9 && ox < 9;
+                ga[j] = n0 * 81 + (ok ? oy * 9 + ox : 0);
+                okm[j] = ok ? 0xFFFFu : 0;
+            }
+        }
         bf16x8 bbuf[2][2];
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
@@ -580,20 +605,11 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
                     bbuf[cur ^ 1][nt] = as_bf16x8(
                         *(const u16x8*)(w2tp + (nt * 16 + l15) * 1024 + k1));
             }
-            const int k0 = ks * 32 + l4 * 8;
             u16x8 av;
             #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int k = k0 + j;
-                const int n = k >> 4, tt = k & 15;
-                const int ky = tt >> 2, kx = tt & 3;
-                const int ry = iy - ky, rx = ix - kx;
-                const int oy = ry >> 1, ox = rx >> 1;
-                const bool ok = mv && ry >= 0 && rx >= 0 && (ry & 1) == 0 &&
-                                (rx & 1) == 0 && oy < 9 && ox < 9;
-                unsigned short v = d2s[n * 81 + (ok ? oy * 9 + ox : 0)];
-                av[j] = ok ? v : (unsigned short)0;
-            }
+            for (int j = 0; j < 8; ++j) av[j] = d2s[ga[j]] & okm[j];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) ga[j] += 2 * 81;
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
             acc[1] = mfma16(a, bbuf[cur][1], acc[1]);

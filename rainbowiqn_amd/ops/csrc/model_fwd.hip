@@ -159,6 +159,21 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
         const int m = mt * 16 + l15;
         const int oy = m / 9, ox = m % 9;
         const bool mv = m < 81;
+        // gather addresses are fixed per lane except the channel plane,
+        // which advances +2 per k-step (+2*20*C1_STRIDE ushorts) -- the
+        // per-element index arithmetic dominated this loop otherwise
+        int ga[8];
+        {
+            const int c0 = l4 >> 1;
+            const int ky0 = (l4 & 1) * 2;  // t16 = (l4*8)&15 in {0,8}
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int row = ky0 + (j >> 2), kx = j & 3;
+                ga[j] = mv ? LDS_C1 + (c0 * 20 + oy * 2 + row) * C1_STRIDE +
+                                 ox * 2 + kx
+                           : LDS_C1 + c0 * 20 * C1_STRIDE;
+            }
+        }
         bf16x8 bbuf[2][2];
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
@@ -174,20 +189,11 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
                     bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
                         w2 + ((nh * 2 + nt) * 16 + l15) * 512 + k1));
             }
-            const int k0 = ks * 32 + l4 * 8;
-            const int c = k0 >> 4, t16 = k0 & 15;  // t16 in {0,8}
-            const int ky0 = t16 >> 2;
-            // branchless gather: clamp the base for invalid rows (their
-            // MFMA output is discarded) so the LDS reads are unconditional
             u16x8 av;
-            {
-                const int base = mv ? LDS_C1 +
-                    (c * 20 + oy * 2 + ky0) * C1_STRIDE + ox * 2 : LDS_C1;
-                #pragma unroll
-                for (int j = 0; j < 4; ++j) av[j] = lds[base + j];
-                #pragma unroll
-                for (int j = 0; j < 4; ++j) av[4 + j] = lds[base + C1_STRIDE + j];
-            }
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) av[j] = lds[ga[j]];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) ga[j] += 2 * 20 * C1_STRIDE;
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
             acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
@@ -218,6 +224,20 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
         const int m = mt * 16 + l15;
         const int oy = m / 7, ox = m % 7;
         const bool mv = m < 49;
+        int ga[8];
+        unsigned short okm[8];
+        {
+            const int c0 = l4 >> 1;
+            const int t16 = (l4 & 1) * 8;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+                const bool ok = mv && ky < 3 && kx < 3;
+                ga[j] = LDS_C2 + (c0 * 9 + (ok ? oy + ky : 0)) * C2_STRIDE +
+                        (ok ? ox + kx : 0);
+                okm[j] = ok ? 0xFFFFu : 0;
+            }
+        }
         bf16x8 bbuf[2][2];
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
@@ -233,19 +253,12 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
                     bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
                         w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
             }
-            const int k0 = ks * 32 + l4 * 8;
-            const int c = k0 >> 4, t16 = k0 & 15;
             u16x8 av;
             #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
-                const bool ok = mv && ky < 3 && kx < 3;
-                const int idx =
-                    LDS_C2 + (c * 9 + (ok ? oy + ky : 0)) * C2_STRIDE +
-                    (ok ? ox + kx : 0);
-                unsigned short v = lds[idx];  // unconditional (clamped) read
-                av[j] = ok ? v : (unsigned short)0;
-            }
+            for (int j = 0; j < 8; ++j)
+                av[j] = lds[ga[j]] & okm[j];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) ga[j] += 2 * 9 * C2_STRIDE;
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
             acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
