@@ -87,17 +87,23 @@ class Communicator:
         dummy of the same shape).  Returns the list of actor chunks on the
         learner (entries 1..W-1), None on actors.
 
-        On RCCL this is an all-gather over xGMI: each actor's chunk rides its
-        own direct xGMI link to the learner GPU concurrently; chunks are a few
-        hundred KB so the redundant actor-side copies are immaterial, and
-        all_gather is supported by both RCCL and gloo (NCCL gather is not).
+        Implemented as batched point-to-point sends (RCCL/gloo both support
+        send/recv): each actor's chunk rides its own direct xGMI link to the
+        learner GPU concurrently (7 links x ~153 GB/s -- star-shaped gather,
+        exactly the reference's Redis traffic shape).  An all-gather would
+        broadcast every chunk to every rank: W x the bytes for no reason.
         """
         if not self.initialized:
             return [chunk]
-        out = [torch.empty_like(chunk) for _ in range(self.world_size)]
-        dist.all_gather(out, chunk.contiguous())
         if self.is_learner:
-            return out[1:]
+            out = [torch.empty_like(chunk) for _ in range(self.n_actors)]
+            ops = [dist.P2POp(dist.irecv, out[a], a + 1)
+                   for a in range(self.n_actors)]
+            for w in dist.batch_isend_irecv(ops):
+                w.wait()
+            return out
+        dist.batch_isend_irecv(
+            [dist.P2POp(dist.isend, chunk.contiguous(), 0)])[0].wait()
         return None
 
     def all_reduce_sum(self, t: torch.Tensor) -> torch.Tensor:
