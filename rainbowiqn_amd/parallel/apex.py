@@ -149,6 +149,17 @@ class ActorState:
             self.total_steps += self.n_envs
 
 
+class _ActorSteps:
+    """total_steps across the rank's local actor loops."""
+
+    def __init__(self, loops):
+        self._loops = loops
+
+    @property
+    def total_steps(self) -> int:
+        return sum(l.total_steps for l in self._loops)
+
+
 class ApexRunner:
     """Fixed-cadence Ape-X round loop shared by the launch scripts and bench.
 
@@ -161,18 +172,25 @@ class ApexRunner:
 
     def __init__(self, cfg, comm: Communicator, n_envs: int = 16,
                  chunk_steps: Optional[int] = None, replay_capacity: Optional[int] = None,
-                 device: Optional[str] = None):
+                 device: Optional[str] = None, n_local_actors: Optional[int] = None):
         self.cfg = cfg
         self.comm = comm
         dev = device or str(comm.device)
         self.n_envs = n_envs
         self.chunk_steps = chunk_steps or max(1, cfg.transition_chunk_size // n_envs)
+        # BASELINE config 3: several independent actors (own nets, own noise,
+        # own env streams, own replay rings) colocated on each actor rank.
+        # Defaults to the reference's --nb-actor flag in single-process mode.
+        if n_local_actors is None:
+            n_local_actors = max(1, cfg.nb_actor) if comm.world_size == 1 else 1
+        self.n_local_actors = n_local_actors
         self.grad_steps = 0
-        self.losses: List[float] = []
+        self.losses = []
         self.actor = None
         self.replay = None
         self.glearner = None
         self.gactor = None
+        self.gactors = []
 
         from ..env import n_actions_for
 
@@ -187,6 +205,8 @@ class ApexRunner:
 
                 get_extension()  # raises: native path is mandatory on GPU
 
+        if not self.fast:
+            self.n_local_actors = 1  # multi-local-actor is a fast-path feature
         if self.fast:
             self._init_fast(cfg, comm, dev, n_act, replay_capacity)
         else:
@@ -206,25 +226,36 @@ class ApexRunner:
         # chunk_steps must be a multiple of history for the actor ring phase
         h = cfg.history_length
         self.chunk_steps = max(h, (self.chunk_steps // h) * h)
+        M = self.n_local_actors
         if comm.is_learner:
             if replay_capacity is not None:
                 cfg_cap = cfg
                 cfg_cap.memory_capacity = replay_capacity
-            self.glearner = GpuLearner(cfg, n_act, n_rings=comm.n_actors,
+            self.glearner = GpuLearner(cfg, n_act, n_rings=comm.n_actors * M,
                                        device=dev, use_graph=cfg.hipgraph)
             self.replay = self.glearner.replay
             self.flat_weights = self.glearner.flat_online
         if comm.is_actor:
-            if comm.world_size == 1:
-                net = self.glearner.online  # colocated: share the online net
-            else:
-                dt = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
-                net = RainbowIQN(cfg, n_act).to(dev).to(dt)
-                self.flat_weights = flatten_module(net)
-            self.gactor = GpuActorLoop(cfg, net, self.n_envs, self.chunk_steps,
-                                       device=dev, use_graph=cfg.hipgraph,
-                                       seed_offset=comm.rank * 7717)
-            self.actor = self.gactor  # exposes .total_steps
+            dt = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
+            self._actor_flats = []
+            for j in range(M):
+                if comm.world_size == 1 and M == 1:
+                    net = self.glearner.online  # colocated: share the online net
+                    flat = None
+                else:
+                    # independent actor: own net copy (own NoisyNet noise),
+                    # refreshed from the learner's flat weights each round
+                    net = RainbowIQN(cfg, n_act).to(dev).to(dt)
+                    flat = flatten_module(net)
+                    if comm.world_size > 1 and j == 0:
+                        self.flat_weights = flat
+                self._actor_flats.append(flat)
+                self.gactors.append(
+                    GpuActorLoop(cfg, net, self.n_envs, self.chunk_steps,
+                                 device=dev, use_graph=cfg.hipgraph,
+                                 seed_offset=comm.rank * 7717 + j * 131071))
+            self.gactor = self.gactors[0]
+            self.actor = _ActorSteps(self.gactors)
 
     # -- one Ape-X round ----------------------------------------------------
     def run_round(self, learn: bool = True) -> None:
@@ -236,23 +267,45 @@ class ApexRunner:
     def _run_round_fast(self, learn: bool) -> None:
         cfg = self.cfg
         comm = self.comm
-        if self.gactor is not None:
-            # NoisyNet noise reset happens inside the captured round graph
-            self.gactor.run_round()
-            chunk_f, chunk_m = self.gactor.chunk_frames, self.gactor.chunk_meta
+        M = self.n_local_actors
+        local_f, local_m = [], []
+        if self.gactors:
+            for j, loop in enumerate(self.gactors):
+                if self._actor_flats[j] is not None:
+                    # refresh this actor's net from the latest weights
+                    src = (self.glearner.flat_online if comm.world_size == 1
+                           else self.flat_weights)
+                    self._actor_flats[j].copy_(src)
+                loop.run_round()
+                local_f.append(loop.chunk_frames)
+                local_m.append(loop.chunk_meta)
         else:
             c = self.n_envs * self.chunk_steps
-            chunk_f = torch.zeros(c, 84, 84, dtype=torch.uint8, device=comm.device)
-            chunk_m = torch.zeros(c, 3, device=comm.device)
+            for _ in range(M):
+                local_f.append(torch.zeros(c, 84, 84, dtype=torch.uint8,
+                                           device=comm.device))
+                local_m.append(torch.zeros(c, 3, device=comm.device))
 
+        frames_list, meta_list = [], []
         if comm.world_size > 1:
-            frames_list = comm.gather_chunks(chunk_f)
-            meta_list = comm.gather_chunks(chunk_m)
+            for j in range(M):
+                fl = comm.gather_chunks(local_f[j])
+                ml = comm.gather_chunks(local_m[j])
+                if comm.is_learner:
+                    frames_list.extend(fl)
+                    meta_list.extend(ml)
         else:
-            frames_list, meta_list = [chunk_f], [chunk_m]
+            frames_list, meta_list = local_f, local_m
         if comm.is_learner:
-            for a, (f, m) in enumerate(zip(frames_list, meta_list)):
-                self.glearner.append_actor_chunk(a, f, meta=m)
+            # ring id = actor_rank * M + local actor (interleaved per j-gather:
+            # frames_list order is [j0: ranks 1..W-1][j1: ...] when W > 1)
+            for i, (f, m) in enumerate(zip(frames_list, meta_list)):
+                if comm.world_size > 1:
+                    j, a = divmod(i, comm.n_actors)
+                    ring = a * M + j
+                else:
+                    ring = i
+                self.glearner.append_actor_chunk(ring, f, meta=m)
             if learn and len(self.glearner.replay) >= cfg.learn_start:
                 self.glearner.learn_steps(cfg.learn_per_round)
                 self.grad_steps = self.glearner.grad_steps

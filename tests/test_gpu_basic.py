@@ -77,3 +77,28 @@ def test_bf16_autocast_matches_fp32_roughly(gcfg):
         z32, _ = a32.online_net(x, 8, taus=taus)
     # same taus + same weights: values close at bf16 tolerance
     assert (z16.float() - z32).abs().max() < 0.15
+
+
+@pytest.mark.gpu
+def test_apex_multi_actor_one_gpu():
+    """BASELINE config 3: several independent actors (own nets/noise/rings)
+    plus the learner colocated on ONE GPU (reference: 4 actor processes)."""
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    cfg = Config(game="breakout", device="cuda", dtype="bf16",
+                 learn_start=512, memory_capacity=16384, nb_actor=4,
+                 learn_per_round=2)
+    comm = Communicator(backend="auto")
+    runner = ApexRunner(cfg, comm, n_envs=16, chunk_steps=8)
+    assert len(runner.gactors) == 4
+    assert runner.replay.n_rings == 4
+    for _ in range(8):
+        runner.run_round()
+    torch.cuda.synchronize()
+    assert runner.grad_steps > 0
+    # every actor's ring received transitions
+    counts = runner.replay.ring_state[:, 1].cpu()
+    assert (counts > 0).all(), counts
+    assert runner.actor.total_steps == 4 * 8 * 16 * 8  # M * rounds * E * steps
