@@ -102,3 +102,35 @@ def test_apex_multi_actor_one_gpu():
     counts = runner.replay.ring_state[:, 1].cpu()
     assert (counts > 0).all(), counts
     assert runner.actor.total_steps == 4 * 8 * 16 * 8  # M * rounds * E * steps
+
+
+@pytest.mark.gpu
+def test_replay_large_capacity_hbm():
+    """BASELINE config 5 (capacity leg): a multi-million-transition PER ring
+    resident in HBM3E -- 10M transitions is ~71 GB of frames on one GPU;
+    append and stratified sampling stay kernel-bound."""
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.replay.gpu_replay import GpuReplay
+
+    free, total = torch.cuda.mem_get_info()
+    if free < 90 * 2**30:
+        pytest.skip("needs ~90 GB free HBM")
+    cfg = Config(device="cuda", dtype="bf16")
+    rep = GpuReplay(cfg, n_rings=8, capacity=10_000_000, device="cuda",
+                    out_dtype=torch.bfloat16)
+    assert rep.capacity >= 10_000_000 - 8
+    cf = torch.randint(0, 255, (4096, 84, 84), dtype=torch.uint8, device="cuda")
+    cm = torch.rand(4096, 3, device="cuda")
+    for ring in range(8):
+        for _ in range(2):
+            rep.append_actor_chunk(ring, cf, meta=cm)
+    torch.cuda.synchronize()
+    assert len(rep) == 8 * 2 * 4096
+    idx, s, a, r, s2, nt, w = rep.sample()
+    torch.cuda.synchronize()
+    assert torch.isfinite(w).all() and (w > 0).all()
+    assert s.shape == (cfg.batch_size, 4, 84, 84)
+    tot = rep.tree_total()
+    assert abs(tot - len(rep)) / len(rep) < 0.01  # all priorities 1.0
+    del rep
+    torch.cuda.empty_cache()
