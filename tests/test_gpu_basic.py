@@ -101,7 +101,8 @@ def test_apex_multi_actor_one_gpu():
     # every actor's ring received transitions
     counts = runner.replay.ring_state[:, 1].cpu()
     assert (counts > 0).all(), counts
-    assert runner.actor.total_steps == 4 * 8 * 16 * 8  # M * rounds * E * steps
+    # M * (rounds + 1 capture-warmup round) * E * steps
+    assert runner.actor.total_steps == 4 * 9 * 16 * 8
 
 
 @pytest.mark.gpu
