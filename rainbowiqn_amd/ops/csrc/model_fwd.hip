@@ -403,7 +403,7 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
     const unsigned short* __restrict__ bcat, // [1024]
     unsigned short* __restrict__ out,        // [R,1024]
     int R) {
-    __shared__ unsigned short smem[2 * 2 * HG_BM * HG_BK];  // 2 bufs x (A|B)
+    __shared__ unsigned short smem[3 * 2 * HG_BM * HG_BK];  // 3 bufs x (A|B)
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid of 32x32
@@ -441,12 +441,17 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
 
     f32x4 acc[2][2] = {{{0, 0, 0, 0}, {0, 0, 0, 0}},
                        {{0, 0, 0, 0}, {0, 0, 0, 0}}};
+    // 3-buffer glds pipeline with raw barriers + counted vmcnt: one tile
+    // always in flight across each barrier (__syncthreads would emit a
+    // vmcnt(0) and drain it -- cdna_hip_programming.md section 5 glds table)
     stage(0, 0);
-    __syncthreads();
+    stage(1, HG_BK);
     for (int kt = 0; kt < 3136 / HG_BK; ++kt) {
-        const int cur = kt & 1;
-        if (kt + 1 < 3136 / HG_BK) stage(cur ^ 1, (kt + 1) * HG_BK);
-        const unsigned short* As = smem + cur * (2 * HG_BM * HG_BK);
+        // own 4 glds for buf[kt] landed; buf[kt+1]'s 4 stay in flight
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + 2 < 3136 / HG_BK) stage((kt + 2) % 3, (kt + 2) * HG_BK);
+        const unsigned short* As = smem + (kt % 3) * (2 * HG_BM * HG_BK);
         const unsigned short* Bs = As + HG_BM * HG_BK;
         #pragma unroll
         for (int ksub = 0; ksub < 2; ++ksub) {
@@ -464,8 +469,8 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
                 }
             }
         }
-        __syncthreads();  // drains the in-flight glds for the next buffer
     }
+    __builtin_amdgcn_s_barrier();  // epilogue reads nothing from LDS; order only
     #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
         #pragma unroll
