@@ -447,8 +447,12 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
     stage(0, 0);
     stage(1, HG_BK);
     for (int kt = 0; kt < 3136 / HG_BK; ++kt) {
-        // own 4 glds for buf[kt] landed; buf[kt+1]'s 4 stay in flight
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        // own 4 glds for buf[kt] landed; buf[kt+1]'s 4 stay in flight.
+        // Last iteration has only buf[kt]'s 4 outstanding -> wait to 0.
+        if (kt + 1 < 3136 / HG_BK)
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         if (kt + 2 < 3136 / HG_BK) stage((kt + 2) % 3, (kt + 2) * HG_BK);
         const unsigned short* As = smem + (kt % 3) * (2 * HG_BM * HG_BK);
