@@ -397,13 +397,14 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
 // ds_read_b128 fragment reads don't hit row-stride bank conflicts
 // (cdna_hip_programming.md section 5 glds recipe).  The next K-tile's DMA is
 // in flight while the current one computes; __syncthreads drains it.
+template <int NBUF>  // 2 at large R (occupancy), 3 at small R (deeper pipe)
 __global__ __launch_bounds__(256) void heads_gemm_kernel(
     const unsigned short* __restrict__ hin,  // [R,3136]
     const unsigned short* __restrict__ wcat, // [1024,3136]
     const unsigned short* __restrict__ bcat, // [1024]
     unsigned short* __restrict__ out,        // [R,1024]
     int R) {
-    __shared__ unsigned short smem[3 * 2 * HG_BM * HG_BK];  // 3 bufs x (A|B)
+    __shared__ unsigned short smem[NBUF * 2 * HG_BM * HG_BK];
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid of 32x32
@@ -445,17 +446,18 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
     // always in flight across each barrier (__syncthreads would emit a
     // vmcnt(0) and drain it -- cdna_hip_programming.md section 5 glds table)
     stage(0, 0);
-    stage(1, HG_BK);
+    if (NBUF == 3) stage(1, HG_BK);
     for (int kt = 0; kt < 3136 / HG_BK; ++kt) {
-        // own 4 glds for buf[kt] landed; buf[kt+1]'s 4 stay in flight.
-        // Last iteration has only buf[kt]'s 4 outstanding -> wait to 0.
-        if (kt + 1 < 3136 / HG_BK)
+        // wait own glds for buf[kt]; with NBUF=3 buf[kt+1]'s 4 stay in
+        // flight (except at the tail, where only buf[kt] is outstanding)
+        if (NBUF == 3 && kt + 1 < 3136 / HG_BK)
             asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
         else
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
-        if (kt + 2 < 3136 / HG_BK) stage((kt + 2) % 3, (kt + 2) * HG_BK);
-        const unsigned short* As = smem + (kt % 3) * (2 * HG_BM * HG_BK);
+        if (kt + NBUF - 1 < 3136 / HG_BK)
+            stage((kt + NBUF - 1) % NBUF, (kt + NBUF - 1) * HG_BK);
+        const unsigned short* As = smem + (kt % NBUF) * (2 * HG_BM * HG_BK);
         const unsigned short* Bs = As + HG_BM * HG_BK;
         #pragma unroll
         for (int ksub = 0; ksub < 2; ++ksub) {
@@ -648,10 +650,18 @@ void riqn_tau_embed(const float* taus, const void* w, const void* bias,
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
                      void* out, int R, void* stream) {
     dim3 grid((R + HG_BM - 1) / HG_BM, 1024 / HG_BN);
-    hipLaunchKernelGGL(heads_gemm_kernel, grid, dim3(256), 0,
-                       (hipStream_t)stream, (const unsigned short*)h,
-                       (const unsigned short*)wcat,
-                       (const unsigned short*)bcat, (unsigned short*)out, R);
+    if (R >= 4096)
+        hipLaunchKernelGGL(heads_gemm_kernel<2>, grid, dim3(256), 0,
+                           (hipStream_t)stream, (const unsigned short*)h,
+                           (const unsigned short*)wcat,
+                           (const unsigned short*)bcat, (unsigned short*)out,
+                           R);
+    else
+        hipLaunchKernelGGL(heads_gemm_kernel<3>, grid, dim3(256), 0,
+                           (hipStream_t)stream, (const unsigned short*)h,
+                           (const unsigned short*)wcat,
+                           (const unsigned short*)bcat, (unsigned short*)out,
+                           R);
 }
 
 void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
