@@ -275,7 +275,8 @@ class ApexRunner:
                     # refresh this actor's net from the latest weights
                     src = (self.glearner.flat_online if comm.world_size == 1
                            else self.flat_weights)
-                    self._actor_flats[j].copy_(src)
+                    if self._actor_flats[j] is not src:
+                        self._actor_flats[j].copy_(src)
                 loop.run_round()
                 local_f.append(loop.chunk_frames)
                 local_m.append(loop.chunk_meta)
