@@ -265,5 +265,116 @@ int main(int argc, char** argv) {
     printf("stage  c1 only           %7.2f us\n", run<2>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
     printf("stage  c2 only           %7.2f us\n", run<4>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
     printf("stage  c3 only           %7.2f us\n", run<8>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("c3 iso: mfma only        %7.2f us\n", runc3<0>(B, w3p, feat, it));
+    printf("c3 iso: +gather          %7.2f us\n", runc3<1>(B, w3p, feat, it));
+    printf("c3 iso: +wload           %7.2f us\n", runc3<2>(B, w3p, feat, it));
+    printf("c3 iso: gather+wload     %7.2f us\n", runc3<3>(B, w3p, feat, it));
+    printf("c3 iso: gather+ldsw      %7.2f us\n", runc3<7>(B, w3p, feat, it));
     return 0;
+}
+
+// ---------------------------------------------------------------------------
+// conv3 cost isolation: V bit0 = real A-gathers, bit1 = real B weight loads,
+// bit2 = stage weights in LDS first and read B-frags from LDS
+// ---------------------------------------------------------------------------
+template <int V>
+__global__ __launch_bounds__(512, 1) void c3probe_kernel(
+    const unsigned short* __restrict__ w3p,
+    unsigned short* __restrict__ feat) {
+    __shared__ unsigned short lds[LDS_TOTAL];
+    const int b = blockIdx.x;
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wave = t >> 6;
+    const int l15 = lane & 15;
+    const int l4 = lane >> 4;
+    // fill c2out region with something
+    for (int i = t; i < 6912; i += 512) lds[LDS_C2 + i] = (unsigned short)(i);
+    unsigned short* wlds = lds;  // reuse input region for staged weights
+    if (V & 4) {
+        // stage HALF of w3p (the 64 KB this wave-group needs) into LDS
+        for (int i = t * 8; i < 32 * 1024; i += 512 * 8)
+            *(u16x8*)&wlds[i] = *(const u16x8*)(w3p + i);
+    }
+    __syncthreads();
+
+    const int mt = wave >> 1, nh = wave & 1;
+    f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+    const int m = mt * 16 + l15;
+    const int oy = m / 7, ox = m % 7;
+    const bool mv = m < 49;
+    int ga[8];
+    unsigned short okm[8];
+    {
+        const int c0 = l4 >> 1;
+        const int t16 = (l4 & 1) * 8;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+            const bool ok = mv && ky < 3 && kx < 3;
+            ga[j] = LDS_C2 + (c0 * 9 + (ok ? oy + ky : 0)) * C2_STRIDE +
+                    (ok ? ox + kx : 0);
+            okm[j] = ok ? 0xFFFFu : 0;
+        }
+    }
+    bf16x8 bbuf[2][2];
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+        bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
+            w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + l4 * 8));
+    #pragma unroll 2
+    for (int ks = 0; ks < 32; ++ks) {
+        const int cur = ks & 1;
+        if ((V & 2) && ks + 1 < 32) {
+            const int k1 = (ks + 1) * 32 + l4 * 8;
+            #pragma unroll
+            for (int nt = 0; nt < 2; ++nt) {
+                if (V & 4)
+                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                        &wlds[(((nh * 2 + nt) * 16 + l15) & 31) * 1024 + k1]));
+                else
+                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                        w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
+            }
+        }
+        u16x8 av;
+        if (V & 1) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) av[j] = lds[ga[j]] & okm[j];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) ga[j] += 2 * 9 * C2_STRIDE;
+        } else {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) av[j] = okm[j];
+        }
+        bf16x8 a = as_bf16x8(av);
+        acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
+        acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
+    }
+    unsigned short* outp = feat + (int64_t)b * 3136;
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+        const int n = (nh * 2 + nt) * 16 + l15;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const int mm = mt * 16 + l4 * 4 + j;
+            if (mm < 49)
+                outp[n * 49 + mm] = f32_to_bf16(fmaxf(acc[nt][j], 0.0f));
+        }
+    }
+}
+
+template <int V>
+float runc3(int B, unsigned short* w3p, unsigned short* feat, int iters) {
+    for (int i = 0; i < 20; ++i)
+        hipLaunchKernelGGL(c3probe_kernel<V>, dim3(B), dim3(512), 0, 0, w3p, feat);
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0); hipEventCreate(&e1);
+    hipEventRecord(e0);
+    for (int i = 0; i < iters; ++i)
+        hipLaunchKernelGGL(c3probe_kernel<V>, dim3(B), dim3(512), 0, 0, w3p, feat);
+    hipEventRecord(e1);
+    hipEventSynchronize(e1);
+    float ms; hipEventElapsedTime(&ms, e0, e1);
+    return ms * 1000.0f / iters;
 }
