@@ -240,38 +240,6 @@ float run(int B, unsigned short* x, unsigned short* w1, unsigned short* b1,
     return ms * 1000.0f / iters;
 }
 
-int main(int argc, char** argv) {
-    int B = argc > 1 ? atoi(argv[1]) : 256;
-    auto alloc = [](size_t n) {
-        void* p;
-        HIP_CHECK(hipMalloc(&p, n * 2));
-        HIP_CHECK(hipMemset(p, 0x3c, n * 2));
-        return (unsigned short*)p;
-    };
-    auto* x = alloc((size_t)B * 4 * 84 * 84);
-    auto* w1 = alloc(32 * 256);
-    auto* b1 = alloc(32);
-    auto* w2 = alloc(64 * 512);
-    auto* b2 = alloc(64);
-    auto* w3p = alloc(64 * 1024);
-    auto* b3 = alloc(64);
-    auto* feat = alloc((size_t)B * 3136);
-    int it = 300;
-    printf("B=%d\n", B);
-    printf("stage  in                %7.2f us\n", run<1>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
-    printf("stage  in+c1             %7.2f us\n", run<3>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
-    printf("stage  in+c1+c2          %7.2f us\n", run<7>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
-    printf("stage  full              %7.2f us\n", run<15>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
-    printf("stage  c1 only           %7.2f us\n", run<2>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
-    printf("stage  c2 only           %7.2f us\n", run<4>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
-    printf("stage  c3 only           %7.2f us\n", run<8>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
-    printf("c3 iso: mfma only        %7.2f us\n", runc3<0>(B, w3p, feat, it));
-    printf("c3 iso: +gather          %7.2f us\n", runc3<1>(B, w3p, feat, it));
-    printf("c3 iso: +wload           %7.2f us\n", runc3<2>(B, w3p, feat, it));
-    printf("c3 iso: gather+wload     %7.2f us\n", runc3<3>(B, w3p, feat, it));
-    printf("c3 iso: gather+ldsw      %7.2f us\n", runc3<7>(B, w3p, feat, it));
-    return 0;
-}
 
 // ---------------------------------------------------------------------------
 // conv3 cost isolation: V bit0 = real A-gathers, bit1 = real B weight loads,
@@ -377,4 +345,37 @@ float runc3(int B, unsigned short* w3p, unsigned short* feat, int iters) {
     hipEventSynchronize(e1);
     float ms; hipEventElapsedTime(&ms, e0, e1);
     return ms * 1000.0f / iters;
+}
+
+int main(int argc, char** argv) {
+    int B = argc > 1 ? atoi(argv[1]) : 256;
+    auto alloc = [](size_t n) {
+        void* p;
+        HIP_CHECK(hipMalloc(&p, n * 2));
+        HIP_CHECK(hipMemset(p, 0x3c, n * 2));
+        return (unsigned short*)p;
+    };
+    auto* x = alloc((size_t)B * 4 * 84 * 84);
+    auto* w1 = alloc(32 * 256);
+    auto* b1 = alloc(32);
+    auto* w2 = alloc(64 * 512);
+    auto* b2 = alloc(64);
+    auto* w3p = alloc(64 * 1024);
+    auto* b3 = alloc(64);
+    auto* feat = alloc((size_t)B * 3136);
+    int it = 300;
+    printf("B=%d\n", B);
+    printf("stage  in                %7.2f us\n", run<1>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("stage  in+c1             %7.2f us\n", run<3>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("stage  in+c1+c2          %7.2f us\n", run<7>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("stage  full              %7.2f us\n", run<15>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("stage  c1 only           %7.2f us\n", run<2>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("stage  c2 only           %7.2f us\n", run<4>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("stage  c3 only           %7.2f us\n", run<8>(B, x, w1, b1, w2, b2, w3p, b3, feat, it));
+    printf("c3 iso: mfma only        %7.2f us\n", runc3<0>(B, w3p, feat, it));
+    printf("c3 iso: +gather          %7.2f us\n", runc3<1>(B, w3p, feat, it));
+    printf("c3 iso: +wload           %7.2f us\n", runc3<2>(B, w3p, feat, it));
+    printf("c3 iso: gather+wload     %7.2f us\n", runc3<3>(B, w3p, feat, it));
+    printf("c3 iso: gather+ldsw      %7.2f us\n", runc3<7>(B, w3p, feat, it));
+    return 0;
 }
