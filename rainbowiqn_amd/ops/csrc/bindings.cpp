@@ -254,11 +254,18 @@ void noisy_material4(std::vector<torch::Tensor> ts, torch::Tensor rng_state) {
 
 #define CHECK_BF16(t) CHECK_T(t, torch::kBFloat16)
 
-void conv_pack3(torch::Tensor w3, torch::Tensor w3p) {
-    CHECK_BF16(w3);
-    CHECK_BF16(w3p);
-    TORCH_CHECK(w3.numel() == 64 * 64 * 9 && w3p.numel() == 64 * 64 * 16);
-    riqn_conv_pack3(w3.data_ptr(), w3p.data_ptr(), cur_stream());
+void pack_fwd_frag(torch::Tensor w1, torch::Tensor w2, torch::Tensor w3,
+                   torch::Tensor wiqn, torch::Tensor w1f, torch::Tensor w2f,
+                   torch::Tensor w3f, torch::Tensor wiqnf) {
+    CHECK_BF16(w1); CHECK_BF16(w2); CHECK_BF16(w3); CHECK_BF16(wiqn);
+    CHECK_BF16(w1f); CHECK_BF16(w2f); CHECK_BF16(w3f); CHECK_BF16(wiqnf);
+    TORCH_CHECK(w1.numel() == 32 * 256 && w1f.numel() == 16 * 512);
+    TORCH_CHECK(w2.numel() == 64 * 512 && w2f.numel() == 64 * 512);
+    TORCH_CHECK(w3.numel() == 64 * 576 && w3f.numel() == 128 * 512);
+    TORCH_CHECK(wiqn.numel() == 3136 * 64 && wiqnf.numel() == 392 * 512);
+    riqn_pack_fwd_frag(w1.data_ptr(), w2.data_ptr(), w3.data_ptr(),
+                       wiqn.data_ptr(), w1f.data_ptr(), w2f.data_ptr(),
+                       w3f.data_ptr(), wiqnf.data_ptr(), cur_stream());
 }
 
 void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
@@ -272,7 +279,8 @@ void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
     CHECK_BF16(feat);
     int B = (int)x.size(0);
     TORCH_CHECK(x.numel() == (int64_t)B * 4 * 84 * 84, "history must be 4");
-    TORCH_CHECK(w1.numel() == 32 * 256 && w2.numel() == 64 * 512);
+    TORCH_CHECK(w1.numel() == 16 * 512 && w2.numel() == 64 * 512);
+    TORCH_CHECK(w3p.numel() == 128 * 512);
     TORCH_CHECK(feat.numel() == (int64_t)B * 3136);
     void *p1 = nullptr, *p2 = nullptr;
     if (c1g.has_value()) {
@@ -294,7 +302,7 @@ void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
     CHECK_BF16(w); CHECK_BF16(bias); CHECK_BF16(feat); CHECK_BF16(h);
     int R = (int)taus.numel();
     TORCH_CHECK(h.numel() == (int64_t)R * 3136);
-    TORCH_CHECK(w.numel() == 3136 * 64);
+    TORCH_CHECK(w.numel() == 392 * 512);
     TORCH_CHECK(feat.numel() * nq == h.numel(), "feat rows * nq != R");
     void* pp = nullptr;
     void* cb = nullptr;
@@ -483,7 +491,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("noisy_material", &noisy_material);
     m.def("noisy_material4", &noisy_material4);
     m.def("stack_gather", &stack_gather);
-    m.def("conv_pack3", &conv_pack3);
+    m.def("pack_fwd_frag", &pack_fwd_frag);
     m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
     m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt);
     m.def("heads_gemm", &heads_gemm);

@@ -522,16 +522,15 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
             bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
-                w3tp + ((nh * 2 + nt) * 16 + l15) * 1024 + l4 * 8));
+                w3tp + ((nh * 2 + nt) << 9) + lane * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
             const int cur = ks & 1;
             if (ks + 1 < 32) {
-                const int k1 = (ks + 1) * 32 + l4 * 8;
                 #pragma unroll
                 for (int nt = 0; nt < 2; ++nt)
                     bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                        w3tp + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
+                        w3tp + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
             }
             u16x8 av;
             #pragma unroll
@@ -594,16 +593,15 @@ __global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
             bbuf[0][nt] = as_bf16x8(
-                *(const u16x8*)(w2tp + (nt * 16 + l15) * 1024 + l4 * 8));
+                *(const u16x8*)(w2tp + (nt << 9) + lane * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
             const int cur = ks & 1;
             if (ks + 1 < 32) {
-                const int k1 = (ks + 1) * 32 + l4 * 8;
                 #pragma unroll
                 for (int nt = 0; nt < 2; ++nt)
-                    bbuf[cur ^ 1][nt] = as_bf16x8(
-                        *(const u16x8*)(w2tp + (nt * 16 + l15) * 1024 + k1));
+                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                        w2tp + (((ks + 1) * 2 + nt) << 9) + lane * 8));
             }
             u16x8 av;
             #pragma unroll
@@ -735,23 +733,35 @@ __global__ void conv_finalize_kernel(
     }
 }
 
-// pack transposed conv weights for the data backward:
-//   w3tp[c][n*16 + ky*4 + kx] = w3[n][c*9 + ky*3 + kx]   (3x3 -> 4x4 pad)
-//   w2tp[c][n*16 + ky*4 + kx] = w2[n][c*16 + ky*4 + kx]
-__global__ void pack_bwd_weights_kernel(const unsigned short* __restrict__ w3,
-                                        const unsigned short* __restrict__ w2,
-                                        unsigned short* __restrict__ w3tp,
-                                        unsigned short* __restrict__ w2tp) {
-    int i = threadIdx.x + blockIdx.x * blockDim.x;
-    if (i < 64 * 1024) {  // w3tp
-        int c = i >> 10, rem = i & 1023, n = rem >> 4, tt = rem & 15;
-        int ky = tt >> 2, kx = tt & 3;
-        w3tp[i] = (ky < 3 && kx < 3) ? w3[n * 576 + c * 9 + ky * 3 + kx]
-                                     : (unsigned short)0;
+// Transposed conv weights for the data backward, packed FRAGMENT-MAJOR
+// ([frag][lane][8], frag = ks*CT + ct) so each B-fragment load is one
+// contiguous 1 KB block (see pack_fwd_frag_kernel in model_fwd.hip).
+// Transposed element (c-row, k = n*16 + ky*4 + kx):
+//   conv3: w3[n][c*9 + ky*3 + kx] (0 on the padded taps)
+//   conv2: w2[n][c*16 + ky*4 + kx]
+__global__ void pack_bwd_frag_kernel(const unsigned short* __restrict__ w3,
+                                     const unsigned short* __restrict__ w2,
+                                     unsigned short* __restrict__ w3tf,
+                                     unsigned short* __restrict__ w2tf) {
+    const int i = threadIdx.x + blockIdx.x * blockDim.x;
+    const int lane = (i >> 3) & 63, j = i & 7;
+    const int l15 = lane & 15, l4 = (lane >> 4) & 3;
+    const int frag = i >> 9;
+    if (i < 128 * 512) {  // w3tf: frag = ks*4 + ct, ks in [0,32), ct in [0,4)
+        const int ks = frag >> 2, ct = frag & 3;
+        const int c = ct * 16 + l15;
+        const int k = ks * 32 + l4 * 8 + j;
+        const int n = k >> 4, tt = k & 15, ky = tt >> 2, kx = tt & 3;
+        w3tf[i] = (ky < 3 && kx < 3)
+                      ? w3[n * 576 + c * 9 + ky * 3 + kx]
+                      : (unsigned short)0;
     }
-    if (i < 32 * 1024) {  // w2tp
-        int c = i >> 10, rem = i & 1023, n = rem >> 4, tt = rem & 15;
-        w2tp[i] = w2[n * 512 + c * 16 + tt];
+    if (i < 64 * 512) {  // w2tf: frag = ks*2 + ct, ks in [0,32), ct in [0,2)
+        const int ks = frag >> 1, ct = frag & 1;
+        const int c = ct * 16 + l15;
+        const int k = ks * 32 + l4 * 8 + j;
+        const int n = k >> 4, tt = k & 15;
+        w2tf[i] = w2[n * 512 + c * 16 + tt];
     }
 }
 
@@ -883,8 +893,8 @@ void riqn_conv_finalize(float* dW1s, float* dW2s, float* dW3s, float* db1s,
 
 void riqn_pack_bwd_weights(const void* w3, const void* w2, void* w3tp,
                            void* w2tp, void* stream) {
-    hipLaunchKernelGGL(pack_bwd_weights_kernel, dim3(64), dim3(1024), 0,
-                       (hipStream_t)stream, (const unsigned short*)w3,
+    hipLaunchKernelGGL(pack_bwd_frag_kernel, dim3(128 * 512 / 256), dim3(256),
+                       0, (hipStream_t)stream, (const unsigned short*)w3,
                        (const unsigned short*)w2, (unsigned short*)w3tp,
                        (unsigned short*)w2tp);
 }

@@ -108,7 +108,7 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
         #pragma unroll
         for (int ks = 0; ks < 8; ++ks)
             b1f[nt][ks] = as_bf16x8(
-                *(const u16x8*)(w1 + (nt * 16 + l15) * 256 + ks * 32 + l4 * 8));
+                *(const u16x8*)(w1 + ((ks * 2 + nt) << 9) + lane * 8));
 
     for (int mt = wave; mt < 25; mt += 8) {
         f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
@@ -178,16 +178,15 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
             bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
-                w2 + ((nh * 2 + nt) * 16 + l15) * 512 + l4 * 8));
+                w2 + ((nh * 2 + nt) << 9) + lane * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 16; ++ks) {
             const int cur = ks & 1;
             if (ks + 1 < 16) {
-                const int k1 = (ks + 1) * 32 + l4 * 8;
                 #pragma unroll
                 for (int nt = 0; nt < 2; ++nt)
                     bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                        w2 + ((nh * 2 + nt) * 16 + l15) * 512 + k1));
+                        w2 + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
             }
             u16x8 av;
             #pragma unroll
@@ -242,16 +241,15 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
             bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
-                w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + l4 * 8));
+                w3p + ((nh * 2 + nt) << 9) + lane * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
             const int cur = ks & 1;
             if (ks + 1 < 32) {
-                const int k1 = (ks + 1) * 32 + l4 * 8;
                 #pragma unroll
                 for (int nt = 0; nt < 2; ++nt)
                     bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                        w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
+                        w3p + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
             }
             u16x8 av;
             #pragma unroll
@@ -278,15 +276,47 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
     }
 }
 
-// pack conv3 weight [64][64*3*3] -> [64][64*16] (tap grid padded 3x3 -> 4x4)
-__global__ void pack_conv3_kernel(const unsigned short* __restrict__ w3,
-                                  unsigned short* __restrict__ w3p) {
-    int i = threadIdx.x + blockIdx.x * blockDim.x;
-    if (i >= 64 * 1024) return;
-    int n = i >> 10, rem = i & 1023, c = rem >> 4, t16 = rem & 15;
-    int ky = t16 >> 2, kx = t16 & 3;
-    w3p[i] = (ky < 3 && kx < 3) ? w3[n * 576 + c * 9 + ky * 3 + kx]
-                                : (unsigned short)0;
+// Fragment-major weight packing: MFMA B-fragments are 8 contiguous k per
+// lane at a ROW decided by lane&15 -- loading them from the natural [n][k]
+// layout touches 16 cache lines per instruction (the dominant cost of the
+// conv/tau k-loops, measured by scripts/conv_probe).  Packing weights as
+// [frag][lane][8] makes every fragment load one contiguous 1 KB block.
+// frag(W, ks, nt) element (lane, j) = W[nt*16 + (lane&15)][ks*32 + (lane>>4)*8 + j]
+DEV unsigned short w3_tap(const unsigned short* w3, int n, int k) {
+    int c = k >> 4, t = k & 15, ky = t >> 2, kx = t & 3;
+    return (ky < 3 && kx < 3) ? w3[n * 576 + c * 9 + ky * 3 + kx]
+                              : (unsigned short)0;
+}
+
+__global__ void pack_fwd_frag_kernel(
+    const unsigned short* __restrict__ w1,    // [32,256]
+    const unsigned short* __restrict__ w2,    // [64,512]
+    const unsigned short* __restrict__ w3,    // [64,576]
+    const unsigned short* __restrict__ wiqn,  // [3136,64]
+    unsigned short* __restrict__ w1f,         // [16][512]  frag = ks*2+nt
+    unsigned short* __restrict__ w2f,         // [64][512]  frag = ks*4+nt
+    unsigned short* __restrict__ w3f,         // [128][512] frag = ks*4+nt
+    unsigned short* __restrict__ wiqnf) {     // [392][512] frag = nt*2+ks
+    const int i = threadIdx.x + blockIdx.x * blockDim.x;
+    const int lane = (i >> 3) & 63, j = i & 7;
+    const int l15 = lane & 15, l4 = (lane >> 4) & 3;
+    const int frag = i >> 9;
+    if (i < 16 * 512) {  // w1f: ks in [0,8), nt in [0,2)
+        const int ks = frag >> 1, nt = frag & 1;
+        w1f[i] = w1[(nt * 16 + l15) * 256 + ks * 32 + l4 * 8 + j];
+    }
+    if (i < 64 * 512) {  // w2f: ks in [0,16), nt in [0,4)
+        const int ks = frag >> 2, nt = frag & 3;
+        w2f[i] = w2[(nt * 16 + l15) * 512 + ks * 32 + l4 * 8 + j];
+    }
+    if (i < 128 * 512) {  // w3f: ks in [0,32), nt in [0,4)
+        const int ks = frag >> 2, nt = frag & 3;
+        w3f[i] = w3_tap(w3, nt * 16 + l15, ks * 32 + l4 * 8 + j);
+    }
+    if (i < 392 * 512) {  // wiqnf: nt in [0,196), ks in [0,2)
+        const int nt = frag >> 1, ks = frag & 1;
+        wiqnf[i] = wiqn[(nt * 16 + l15) * 64 + ks * 32 + l4 * 8 + j];
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -352,7 +382,8 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
         f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-            bf16x8 bfr = as_bf16x8(*(const u16x8*)(w + n * 64 + ks * 32 + l4 * 8));
+            bf16x8 bfr = as_bf16x8(
+                *(const u16x8*)(w + ((nt * 2 + ks) << 9) + lane * 8));
             acc[0] = mfma16(afrag[0][ks], bfr, acc[0]);
             acc[1] = mfma16(afrag[1][ks], bfr, acc[1]);
         }
@@ -616,10 +647,15 @@ __global__ __launch_bounds__(256) void heads_act_kernel(
 
 extern "C" {
 
-void riqn_conv_pack3(const void* w3, void* w3p, void* stream) {
-    hipLaunchKernelGGL(pack_conv3_kernel, dim3(64), dim3(1024), 0,
-                       (hipStream_t)stream, (const unsigned short*)w3,
-                       (unsigned short*)w3p);
+void riqn_pack_fwd_frag(const void* w1, const void* w2, const void* w3,
+                        const void* wiqn, void* w1f, void* w2f, void* w3f,
+                        void* wiqnf, void* stream) {
+    hipLaunchKernelGGL(pack_fwd_frag_kernel, dim3(392 * 512 / 256), dim3(256),
+                       0, (hipStream_t)stream, (const unsigned short*)w1,
+                       (const unsigned short*)w2, (const unsigned short*)w3,
+                       (const unsigned short*)wiqn, (unsigned short*)w1f,
+                       (unsigned short*)w2f, (unsigned short*)w3f,
+                       (unsigned short*)wiqnf);
 }
 
 void riqn_conv_encode(const void* x, const void* w1, const void* b1,

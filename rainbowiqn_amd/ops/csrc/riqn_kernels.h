@@ -84,7 +84,9 @@ void riqn_noisy_material(float* eps_in, int64_t n_in, float* eps_out,
 
 // ---- fused RainbowIQN forward (model_fwd.hip) -------------
 // All tensors bf16 unless noted.  Layouts documented in model_fwd.hip.
-void riqn_conv_pack3(const void* w3, void* w3p, void* stream);
+void riqn_pack_fwd_frag(const void* w1, const void* w2, const void* w3,
+                        const void* wiqn, void* w1f, void* w2f, void* w3f,
+                        void* wiqnf, void* stream);
 void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* w2, const void* b2, const void* w3p,
                       const void* b3, void* feat, void* c1g, void* c2g, int B,

@@ -60,24 +60,31 @@ class FusedForward:
             layer.weight_noisy = wslice.view(layer.weight_noisy.shape)
             layer.bias_noisy = bslice.view(layer.bias_noisy.shape)
 
-        # conv weights in natural layout (views into the flat param buffer)
-        self.w1 = net.convs[0].weight.detach().view(32, 256)
+        # raw conv/iqn weights (views into the flat param buffer) ...
+        self.w1raw = net.convs[0].weight.detach().view(32, 256)
         self.b1 = net.convs[0].bias.detach()
-        self.w2 = net.convs[2].weight.detach().view(64, 512)
+        self.w2raw = net.convs[2].weight.detach().view(64, 512)
         self.b2 = net.convs[2].bias.detach()
-        self.w3 = net.convs[4].weight.detach().view(64, 576)
+        self.w3raw = net.convs[4].weight.detach().view(64, 576)
         self.b3 = net.convs[4].bias.detach()
-        self.w3p = torch.empty(64, 1024, dtype=bf, device=d)
-        self.wiqn = net.iqn_fc.weight.detach()  # [3136, 64]
+        self.wiqnraw = net.iqn_fc.weight.detach()  # [3136, 64]
         self.biqn = net.iqn_fc.bias.detach()
+        # ... and their fragment-major packs (one contiguous 1 KB block per
+        # MFMA B-fragment load -- see pack_fwd_frag_kernel)
+        self.w1 = torch.empty(16 * 512, dtype=bf, device=d)
+        self.w2 = torch.empty(64 * 512, dtype=bf, device=d)
+        self.w3p = torch.empty(128 * 512, dtype=bf, device=d)
+        self.wiqn = torch.empty(392 * 512, dtype=bf, device=d)
         self.pack_weights()
 
         self._scratch: Dict[Tuple[int, int], tuple] = {}
 
     def pack_weights(self) -> None:
-        """Refresh the conv3 padded layout (call after conv weights change;
-        graph-capturable)."""
-        self.ext.conv_pack3(self.w3.contiguous(), self.w3p)
+        """Refresh the fragment-major weight packs (call after weight
+        changes; graph-capturable)."""
+        self.ext.pack_fwd_frag(self.w1raw, self.w2raw, self.w3raw,
+                               self.wiqnraw, self.w1, self.w2, self.w3p,
+                               self.wiqn)
 
     def _bufs(self, B: int, nq: int):
         key = (B, nq)

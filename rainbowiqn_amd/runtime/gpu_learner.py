@@ -106,8 +106,8 @@ class GpuLearner:
         d, bf = self.device, torch.bfloat16
         B, N, R = self.B, self.N, self.B * self.N
         z = lambda *s, dt=bf: torch.zeros(*s, dtype=dt, device=d)
-        self.w3tp = z(64, 1024)
-        self.w2tp = z(32, 1024)
+        self.w3tp = z(128 * 512)  # fragment-major transposed conv3
+        self.w2tp = z(64 * 512)   # fragment-major transposed conv2
         self.c1g = z(2 * B, 12800)  # [s; s'] batch; backward uses [:B]
         self.c2g = z(2 * B, 5184)
         self.feat_cat = z(2 * B, 3136)
@@ -169,7 +169,7 @@ class GpuLearner:
         self._reset_noise_device(self.target)
         fo.pack_weights()
         ft.pack_weights()
-        ext.pack_bwd_weights(fo.w3, fo.w2, self.w3tp, self.w2tp)
+        ext.pack_bwd_weights(fo.w3raw, fo.w2raw, self.w3tp, self.w2tp)
         idx, s, a, r, s2, nt, w = self.replay.sample()
         # ---- online conv over [s; s'] in ONE call (activations saved for
         # the backward; the s' half feeds the double-Q select) -------------

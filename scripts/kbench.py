@@ -71,8 +71,8 @@ def main():
 
     us = timeit(lambda: ext.noisy_material4(pack, rng), iters)
     print(f"{'noisy_material4':<28} {'4 layers':<18} {us:9.2f}")
-    us = timeit(lambda: ext.conv_pack3(fw.w3, fw.w3p), iters)
-    print(f"{'conv_pack3':<28} {'':<18} {us:9.2f}")
+    us = timeit(lambda: fw.pack_weights(), iters)
+    print(f"{'pack_fwd_frag':<28} {'':<18} {us:9.2f}")
 
     # replay append at bench chunk size
     from rainbowiqn_amd.replay.gpu_replay import GpuReplay

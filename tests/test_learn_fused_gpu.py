@@ -99,9 +99,9 @@ def setup():
     ext.tau_dw(dphi, cosb, iqn_wg)
     ext.bias_grad_cols(dphi, iqn_bg)
 
-    w3tp = gz(64, 1024)
-    w2tp = gz(32, 1024)
-    ext.pack_bwd_weights(fw.w3, fw.w2, w3tp, w2tp)
+    w3tp = gz(128 * 512)
+    w2tp = gz(64 * 512)
+    ext.pack_bwd_weights(fw.w3raw, fw.w2raw, w3tp, w2tp)
     dpre3g = gz(64, B * 49)
     dc2g = gz(64, B * 81)
     dc1g = gz(32, B * 400)
