@@ -356,19 +356,41 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
         *(u16x4*)&ftile[si * 784 + cc] = *(const u16x4*)(feat + src);
     }
 
-    // A-fragments for both m-tiles x both k-steps (k = l4*8 + j + 32*ks)
+    // A-fragments for both m-tiles x both k-steps (k = l4*8 + j + 32*ks).
+    // cos(k*pi*tau) for the 16 consecutive k of a lane comes from the
+    // Chebyshev recurrence cos((k+1)t) = 2cos(t)cos(kt) - cos((k-1)t):
+    // 3 accurate cosf per lane instead of 16 (this kernel was ALU-bound).
     bf16x8 afrag[2][2];
     #pragma unroll
     for (int mi = 0; mi < 2; ++mi) {
         const int r_a = (mt0 + mi) * 16 + l15;
         const float tau = (r_a < R) ? taus[r_a] : 0.0f;
+        const float th = 3.14159265358979f * tau;
+        const int kbase = l4 * 8;
+        float cm1 = cosf((float)kbase * th);        // cos(k0 * t)
+        float c0 = cosf((float)(kbase + 1) * th);   // cos((k0+1) * t)
+        const float two_ct = 2.0f * cosf(th);
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
             u16x8 av;
             #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                float k = (float)(ks * 32 + l4 * 8 + j);
-                av[j] = f32_to_bf16(cosf(3.14159265358979f * k * tau));
+                if (j == 0) {
+                    av[j] = f32_to_bf16(cm1);
+                } else if (j == 1) {
+                    av[j] = f32_to_bf16(c0);
+                } else {
+                    float c1 = two_ct * c0 - cm1;
+                    cm1 = c0;
+                    c0 = c1;
+                    av[j] = f32_to_bf16(c1);
+                }
+            }
+            if (ks == 0) {
+                // jump from k0+7 to k0+32: re-seed the recurrence so the
+                // ks=1 fragment starts at cos((k0+32)t), cos((k0+33)t)
+                cm1 = cosf((float)(kbase + 32) * th);
+                c0 = two_ct * cm1 - cosf((float)(kbase + 31) * th);
             }
             if (cosb_out != nullptr && blockIdx.y == 0 && r_a < R)
                 *(u16x8*)(cosb_out + (int64_t)r_a * 64 + ks * 32 + l4 * 8) = av;
