@@ -555,6 +555,97 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
 // mode 1: actions[e] = argmax_i mean_{k<nq} q[e*nq+k][i]   (one WG per env)
 // mode 2: theta[r] = q[r][act[r/nq]]                        (f32 [R])
 // ---------------------------------------------------------------------------
+// 128x128-tile variant for large R (act path): 4x the MFMA per barrier and
+// half the on-chip traffic per FLOP vs the 64x64 tile.  Each of 4 waves owns
+// a 64x64 quadrant (4x4 fragments); 2 glds buffers, raw barrier + counted
+// waits, same XOR source swizzle as the small-tile kernel.
+__global__ __launch_bounds__(256) void heads_gemm_big_kernel(
+    const unsigned short* __restrict__ hin,  // [R,3136]
+    const unsigned short* __restrict__ wcat, // [1024,3136]
+    const unsigned short* __restrict__ bcat, // [1024]
+    unsigned short* __restrict__ out,        // [R,1024]
+    int R) {
+    __shared__ unsigned short smem[2 * 2 * 128 * 64];  // 2 bufs x (A|B 16KB)
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;
+    const int wr = wave >> 1, wc = wave & 1;
+    const int l15 = lane & 15, l4 = lane >> 4;
+    const int bm = blockIdx.x * 128, bn = blockIdx.y * 128;
+
+    // 16 glds insts cover a 128x64 operand; wave w issues insts [4w, 4w+4)
+    // for A and for B (8 per wave per buffer).
+    auto stage = [&](int buf, int k0) {
+        unsigned short* base = smem + buf * (2 * 128 * 64);
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int inst = wave * 4 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            {
+                const int gr = min(bm + row, R - 1);
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (hin + (int64_t)gr * 3136 + k0 + piece * 8);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + inst * 512);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+            }
+            {
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (wcat + (int64_t)(bn + row) * 3136 + k0 + piece * 8);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + 128 * 64 + inst * 512);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+            }
+        }
+    };
+
+    f32x4 acc[4][4];
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stage(0, 0);
+    for (int kt = 0; kt < 3136 / 64; ++kt) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + 1 < 3136 / 64) stage((kt + 1) & 1, (kt + 1) * 64);
+        const unsigned short* As = smem + (kt & 1) * (2 * 128 * 64);
+        const unsigned short* Bs = As + 128 * 64;
+        #pragma unroll
+        for (int ksub = 0; ksub < 2; ++ksub) {
+            #pragma unroll
+            for (int fm = 0; fm < 4; ++fm) {
+                const int row = wr * 64 + fm * 16 + l15;
+                const int piece = (ksub * 4 + l4) ^ (row & 7);
+                bf16x8 a = as_bf16x8(*(const u16x8*)&As[row * 64 + piece * 8]);
+                #pragma unroll
+                for (int fn = 0; fn < 4; ++fn) {
+                    const int nrow = wc * 64 + fn * 16 + l15;
+                    const int bp = (ksub * 4 + l4) ^ (nrow & 7);
+                    bf16x8 b = as_bf16x8(*(const u16x8*)&Bs[nrow * 64 + bp * 8]);
+                    acc[fm][fn] = mfma16(a, b, acc[fm][fn]);
+                }
+            }
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+            const int n = bn + wc * 64 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * 64 + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
 // MODE 0: q [R,A] f32 out.  MODE 2: theta[r] = q[r][act[r/nq]] out.
 // Two-pass per row: advantage sums stashed in LDS (sv) -- no dynamically
 // indexed register array (which spills to scratch).
@@ -707,6 +798,15 @@ void riqn_tau_embed(const float* taus, const void* w, const void* bias,
 
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
                      void* out, int R, void* stream) {
+    if (R >= 8192) {
+        dim3 gridb((R + 127) / 128, 1024 / 128);
+        hipLaunchKernelGGL(heads_gemm_big_kernel, gridb, dim3(256), 0,
+                           (hipStream_t)stream, (const unsigned short*)h,
+                           (const unsigned short*)wcat,
+                           (const unsigned short*)bcat, (unsigned short*)out,
+                           R);
+        return;
+    }
     dim3 grid((R + HG_BM - 1) / HG_BM, 1024 / HG_BN);
     if (R >= 4096)
         hipLaunchKernelGGL(heads_gemm_kernel<2>, grid, dim3(256), 0,
