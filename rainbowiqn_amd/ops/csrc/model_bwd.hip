@@ -110,6 +110,7 @@ __global__ __launch_bounds__(256) void zgrad_kernel(
     // each wave accumulates rows wave::4, lane covers 8 coalesced columns
     float wacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     float wb = 0.0f;
+    #pragma unroll 4
     for (int r = wave; r < R; r += 4) {
         float c = dtheta[r];
         if (row > 0) {
@@ -403,6 +404,7 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
 
     f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
                     f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+    #pragma unroll 2
     for (int k0 = 0; k0 < R; k0 += 32) {
         const int kk = k0 + l4 * 8;
         u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};

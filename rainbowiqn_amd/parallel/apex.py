@@ -281,11 +281,14 @@ class ApexRunner:
                 local_f.append(loop.chunk_frames)
                 local_m.append(loop.chunk_meta)
         else:
-            c = self.n_envs * self.chunk_steps
-            for _ in range(M):
-                local_f.append(torch.zeros(c, 84, 84, dtype=torch.uint8,
-                                           device=comm.device))
-                local_m.append(torch.zeros(c, 3, device=comm.device))
+            # learner-only rank: shape templates for the P2P receives
+            if not hasattr(self, "_dummy_chunks"):
+                c = self.n_envs * self.chunk_steps
+                self._dummy_chunks = (
+                    [torch.zeros(c, 84, 84, dtype=torch.uint8,
+                                 device=comm.device) for _ in range(M)],
+                    [torch.zeros(c, 3, device=comm.device) for _ in range(M)])
+            local_f, local_m = self._dummy_chunks
 
         frames_list, meta_list = [], []
         if comm.world_size > 1:

@@ -56,8 +56,6 @@ class GpuActorLoop:
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.total_steps = 0
         self._newest = 0  # ring slot of the most recent frame (round-invariant)
-        self._eps_in = torch.empty(4096, device=d)
-        self._eps_out = torch.empty(4096, device=d)
         # hand-written fused forward (4 kernels/step) on the bf16 path
         self.fused = None
         if (self.dtype == torch.bfloat16

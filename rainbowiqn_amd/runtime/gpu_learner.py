@@ -78,9 +78,6 @@ class GpuLearner:
         self.taus_online = torch.empty(b * self.N, 1, device=self.device)
         self.taus_target = torch.empty(b * self.Np, 1, device=self.device)
         self.taus_select = torch.empty(b * self.K, 1, device=self.device)
-        # scratch for noisy eps vectors (max fan 3136)
-        self._eps_in = torch.empty(4096, device=self.device)
-        self._eps_out = torch.empty(4096, device=self.device)
         self.loss_static = torch.zeros(1, device=self.device)
         self.use_graph = use_graph
         self.graph: Optional[torch.cuda.CUDAGraph] = None
