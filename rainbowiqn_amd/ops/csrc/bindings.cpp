@@ -474,6 +474,31 @@ void pack_bwd_weights(torch::Tensor w3, torch::Tensor w2, torch::Tensor w3tp,
                           w2tp.data_ptr(), cur_stream());
 }
 
+
+void quant_fp8(torch::Tensor src, torch::Tensor dst, torch::Tensor scales) {
+    CHECK_BF16(src);
+    CHECK_T(dst, torch::kUInt8);
+    CHECK_T(scales, torch::kUInt8);
+    int M = (int)src.size(0);
+    TORCH_CHECK(src.size(1) == 3136 && dst.numel() == (int64_t)M * 3200 &&
+                scales.numel() == (int64_t)M * 100);
+    riqn_quant_fp8(src.data_ptr(), dst.data_ptr(), scales.data_ptr(), M,
+                   cur_stream());
+}
+
+void heads_gemm_fp8(torch::Tensor hq, torch::Tensor hsc, torch::Tensor wq,
+                    torch::Tensor wsc, torch::Tensor bcat, torch::Tensor out) {
+    CHECK_T(hq, torch::kUInt8);
+    CHECK_T(wq, torch::kUInt8);
+    CHECK_BF16(bcat); CHECK_BF16(out);
+    int R = (int)hq.size(0);
+    TORCH_CHECK(hq.size(1) == 3200 && wq.numel() == 1024 * 3200);
+    TORCH_CHECK(out.numel() == (int64_t)R * 1024);
+    riqn_heads_gemm_fp8(hq.data_ptr(), hsc.data_ptr(), wq.data_ptr(),
+                        wsc.data_ptr(), bcat.data_ptr(), out.data_ptr(), R,
+                        cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -492,6 +517,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("noisy_material4", &noisy_material4);
     m.def("stack_gather", &stack_gather);
     m.def("pack_fwd_frag", &pack_fwd_frag);
+    m.def("quant_fp8", &quant_fp8);
+    m.def("heads_gemm_fp8", &heads_gemm_fp8);
     m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
     m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt);
     m.def("heads_gemm", &heads_gemm);

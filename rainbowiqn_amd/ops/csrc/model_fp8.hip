@@ -1,0 +1,189 @@
+// fp8 NoisyLinear MFMA path (BASELINE config 5) for gfx950.
+//
+// The heads GEMM (h [R,3136] x Wcat^T -> [R,1024], ~80% of forward FLOPs)
+// runs on the block-scaled MX-fp8 matrix core
+// (v_mfma_scale_f32_16x16x128_f8f6f4, ~2x the bf16 MFMA rate): operands are
+// OCP e4m3 with one e8m0 scale per 32-element k-block, dequantized by the
+// hardware inside the MFMA.  Quantization:
+//   quant_rows_fp8: bf16 [M,3136] -> fp8 [M,3200] (K padded to 25 blocks of
+//   128) + e8m0 scales [M,100]; per-32-block scale 2^e with
+//   e = ilogb(amax) - 7 so |q| < 256 <= 448 (e4m3 max).
+// Wcat is quantized once per noise reset (with pack_weights); h once per
+// forward.  Everything is bf16 elsewhere -- this is the opt-in
+// cfg.fp8_heads path, off by default (the bf16 path is the benchmark dtype).
+#include "common.h"
+#include "riqn_kernels.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+typedef __attribute__((ext_vector_type(4))) unsigned short u16x4;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+typedef __attribute__((ext_vector_type(4))) int i32x4;
+
+DEV float bf(const unsigned short u) { return bf16_to_f32(u); }
+
+// one thread = one 32-element k-block of one row
+__global__ void quant_rows_fp8_kernel(
+    const unsigned short* __restrict__ src,  // [M,3136] bf16
+    unsigned char* __restrict__ dst,         // [M,3200] fp8 e4m3
+    unsigned char* __restrict__ scales,      // [M,100] e8m0
+    int M) {
+    const int64_t tid = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+    if (tid >= (int64_t)M * 100) return;
+    const int row = (int)(tid / 100), blk = (int)(tid % 100);
+    unsigned char* out = dst + (int64_t)row * 3200 + blk * 32;
+    if (blk >= 98) {  // K padding: zero payload, zero scale
+        for (int i = 0; i < 32; i += 4) *(int*)(out + i) = 0;
+        scales[(int64_t)row * 100 + blk] = 0;
+        return;
+    }
+    const unsigned short* s = src + (int64_t)row * 3136 + blk * 32;
+    float v[32];
+    float amax = 0.0f;
+    #pragma unroll
+    for (int i = 0; i < 32; i += 4) {
+        u16x4 x = *(const u16x4*)(s + i);
+        v[i] = bf(x[0]); v[i + 1] = bf(x[1]);
+        v[i + 2] = bf(x[2]); v[i + 3] = bf(x[3]);
+        amax = fmaxf(amax, fmaxf(fmaxf(fabsf(v[i]), fabsf(v[i + 1])),
+                                 fmaxf(fabsf(v[i + 2]), fabsf(v[i + 3]))));
+    }
+    int e = 0;
+    if (amax > 0.0f) e = max(ilogbf(amax) - 7, -127);
+    const float inv = exp2f((float)-e);
+    #pragma unroll
+    for (int i = 0; i < 32; i += 2) {
+        int packed = 0;
+        packed = __builtin_amdgcn_cvt_pk_fp8_f32(v[i] * inv, v[i + 1] * inv,
+                                                 packed, false);
+        *(unsigned short*)(out + i) = (unsigned short)(packed & 0xFFFF);
+    }
+    scales[(int64_t)row * 100 + blk] = (unsigned char)(e + 127);
+}
+
+// 128x128 tile, BK=128 (one scaled MFMA per fragment pair), 2 glds buffers.
+__global__ __launch_bounds__(256) void heads_gemm_fp8_kernel(
+    const unsigned char* __restrict__ hq,   // [R,3200] fp8
+    const unsigned char* __restrict__ hsc,  // [R,100] e8m0
+    const unsigned char* __restrict__ wq,   // [1024,3200] fp8
+    const unsigned char* __restrict__ wsc,  // [1024,100] e8m0
+    const unsigned short* __restrict__ bcat,// [1024] bf16
+    unsigned short* __restrict__ out,       // [R,1024] bf16
+    int R) {
+    __shared__ unsigned char smem[2 * 2 * 128 * 128];  // 2 bufs x (A|B 16KB)
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;
+    const int wr = wave >> 1, wc = wave & 1;
+    const int l15 = lane & 15, l4 = lane >> 4;
+    const int bm = blockIdx.x * 128, bn = blockIdx.y * 128;
+
+    // 16 glds insts per 128x128-byte operand; wave w issues [4w, 4w+4).
+    // Source XOR swizzle on the 16-byte piece index (8 pieces per row).
+    auto stage = [&](int buf, int k0) {
+        unsigned char* base = smem + buf * (2 * 128 * 128);
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int inst = wave * 4 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            {
+                const int gr = min(bm + row, R - 1);
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (hq + (int64_t)gr * 3200 + k0 + piece * 16);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + inst * 1024);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+            }
+            {
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (wq + (int64_t)(bn + row) * 3200 + k0 + piece * 16);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + 128 * 128 + inst * 1024);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+            }
+        }
+    };
+
+    f32x4 acc[4][4];
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stage(0, 0);
+    for (int kt = 0; kt < 25; ++kt) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + 1 < 25) stage((kt + 1) & 1, (kt + 1) * 128);
+        const unsigned char* As = smem + (kt & 1) * (2 * 128 * 128);
+        const unsigned char* Bs = As + 128 * 128;
+        // lane's 32-byte A fragment: [row][l4*32 .. +32) = pieces 2*l4, 2*l4+1
+        #pragma unroll
+        for (int fm = 0; fm < 4; ++fm) {
+            const int row = wr * 64 + fm * 16 + l15;
+            const int pa0 = (l4 * 2) ^ (row & 7), pa1 = (l4 * 2 + 1) ^ (row & 7);
+            i32x4 alo = *(const i32x4*)&As[row * 128 + pa0 * 16];
+            i32x4 ahi = *(const i32x4*)&As[row * 128 + pa1 * 16];
+            i32x8 a = {alo[0], alo[1], alo[2], alo[3],
+                       ahi[0], ahi[1], ahi[2], ahi[3]};
+            const int sa = hsc[(int64_t)min(bm + row, R - 1) * 100 + kt * 4 + l4];
+            #pragma unroll
+            for (int fn = 0; fn < 4; ++fn) {
+                const int nrow = wc * 64 + fn * 16 + l15;
+                const int pb0 = (l4 * 2) ^ (nrow & 7),
+                          pb1 = (l4 * 2 + 1) ^ (nrow & 7);
+                i32x4 blo = *(const i32x4*)&Bs[nrow * 128 + pb0 * 16];
+                i32x4 bhi = *(const i32x4*)&Bs[nrow * 128 + pb1 * 16];
+                i32x8 b = {blo[0], blo[1], blo[2], blo[3],
+                           bhi[0], bhi[1], bhi[2], bhi[3]};
+                const int sb = wsc[(int64_t)(bn + nrow) * 100 + kt * 4 + l4];
+                acc[fm][fn] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                    a, b, acc[fm][fn], 0, 0, 0, sa, 0, sb);
+            }
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+            const int n = bn + wc * 64 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * 64 + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
+}  // namespace
+
+extern "C" {
+
+void riqn_quant_fp8(const void* src, void* dst, void* scales, int M,
+                    void* stream) {
+    int64_t n = (int64_t)M * 100;
+    hipLaunchKernelGGL(quant_rows_fp8_kernel,
+                       dim3((uint32_t)((n + 255) / 256)), dim3(256), 0,
+                       (hipStream_t)stream, (const unsigned short*)src,
+                       (unsigned char*)dst, (unsigned char*)scales, M);
+}
+
+void riqn_heads_gemm_fp8(const void* hq, const void* hsc, const void* wq,
+                         const void* wsc, const void* bcat, void* out, int R,
+                         void* stream) {
+    dim3 grid((R + 127) / 128, 1024 / 128);
+    hipLaunchKernelGGL(heads_gemm_fp8_kernel, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const unsigned char*)hq,
+                       (const unsigned char*)hsc, (const unsigned char*)wq,
+                       (const unsigned char*)wsc,
+                       (const unsigned short*)bcat, (unsigned short*)out, R);
+}
+
+}  // extern "C"

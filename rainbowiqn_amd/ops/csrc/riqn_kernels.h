@@ -136,6 +136,13 @@ void riqn_conv_finalize(float* dW1s, float* dW2s, float* dW3s, float* db1s,
 void riqn_pack_bwd_weights(const void* w3, const void* w2, void* w3tp,
                            void* w2tp, void* stream);
 
+// ---- fp8 MX heads path (model_fp8.hip; BASELINE config 5) ------------------
+void riqn_quant_fp8(const void* src, void* dst, void* scales, int M,
+                    void* stream);
+void riqn_heads_gemm_fp8(const void* hq, const void* hsc, const void* wq,
+                         const void* wsc, const void* bcat, void* out, int R,
+                         void* stream);
+
 // batched 4-layer NoisyLinear regeneration (rng.hip)
 struct NoisyDesc {
     const void *w_mu, *w_sigma, *b_mu, *b_sigma;

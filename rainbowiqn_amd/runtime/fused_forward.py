@@ -30,9 +30,10 @@ from ..ops import get_extension
 
 
 class FusedForward:
-    def __init__(self, net: RainbowIQN, device):
+    def __init__(self, net: RainbowIQN, device, fp8_heads: bool = False):
         self.ext = get_extension()
         self.net = net
+        self.fp8_heads = fp8_heads
         self.device = torch.device(device)
         p = net.fc_h_v.weight_mu
         assert p.dtype == torch.bfloat16, "fused forward is the bf16 path"
@@ -75,6 +76,10 @@ class FusedForward:
         self.w2 = torch.empty(64 * 512, dtype=bf, device=d)
         self.w3p = torch.empty(128 * 512, dtype=bf, device=d)
         self.wiqn = torch.empty(392 * 512, dtype=bf, device=d)
+        # fp8 MX heads path (BASELINE config 5): e4m3 + e8m0 block scales
+        if fp8_heads:
+            self.wq = torch.empty(1024, 3200, dtype=torch.uint8, device=d)
+            self.wsc = torch.empty(1024, 100, dtype=torch.uint8, device=d)
         self.pack_weights()
 
         self._scratch: Dict[Tuple[int, int], tuple] = {}
@@ -85,17 +90,23 @@ class FusedForward:
         self.ext.pack_fwd_frag(self.w1raw, self.w2raw, self.w3raw,
                                self.wiqnraw, self.w1, self.w2, self.w3p,
                                self.wiqn)
+        if self.fp8_heads:
+            self.ext.quant_fp8(self.wcat, self.wq, self.wsc)
 
     def _bufs(self, B: int, nq: int):
         key = (B, nq)
         if key not in self._scratch:
             d, bf = self.device, torch.bfloat16
             R = B * nq
+            extra = ()
+            if self.fp8_heads:
+                extra = (torch.empty(R, 3200, dtype=torch.uint8, device=d),
+                         torch.empty(R, 100, dtype=torch.uint8, device=d))
             self._scratch[key] = (
                 torch.empty(B, 3136, dtype=bf, device=d),     # feat
                 torch.empty(R, 3136, dtype=bf, device=d),     # h
                 torch.empty(R, 1024, dtype=bf, device=d),     # hh
-            )
+            ) + extra
         return self._scratch[key]
 
     def encode(self, x: torch.Tensor) -> torch.Tensor:
@@ -108,11 +119,17 @@ class FusedForward:
     def _trunk(self, x: torch.Tensor, taus: torch.Tensor, nq: int):
         """x [B,4,84,84] bf16, taus [B*nq] f32 -> hh [B*nq, 1024]."""
         B = x.shape[0]
-        feat, h, hh = self._bufs(B, nq)
+        bufs = self._bufs(B, nq)
+        feat, h, hh = bufs[:3]
         self.ext.conv_encode(x, self.w1, self.b1, self.w2, self.b2, self.w3p,
                              self.b3, feat)
         self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h, nq)
-        self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
+        if self.fp8_heads:
+            hq, hsc = bufs[3], bufs[4]
+            self.ext.quant_fp8(h, hq, hsc)
+            self.ext.heads_gemm_fp8(hq, hsc, self.wq, self.wsc, self.bcat, hh)
+        else:
+            self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
         return hh
 
     def quantiles(self, x, taus, nq: int, out: torch.Tensor) -> torch.Tensor:

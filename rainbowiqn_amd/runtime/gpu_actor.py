@@ -63,7 +63,7 @@ class GpuActorLoop:
                 and getattr(cfg, "fused_forward", True)):
             from .fused_forward import FusedForward
 
-            self.fused = FusedForward(net, d)
+            self.fused = FusedForward(net, d, fp8_heads=getattr(cfg, 'fp8_heads', False))
 
     def _round_body(self) -> None:
         cfg = self.cfg

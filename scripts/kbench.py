@@ -69,6 +69,20 @@ def main():
         print(f"{'heads_finish theta':<28} {f'R={R}':<18} {us:9.2f}")
         print()
 
+    # fp8 MX heads path
+    fw8 = FusedForward(RainbowIQN(cfg, A).cuda().to(torch.bfloat16).train(),
+                       "cuda", fp8_heads=True)
+    for B, nq in [(256, 32), (32, 32)]:
+        R = B * nq
+        _, h8, hh8, hq, hsc = fw8._bufs(B, nq)
+        us = timeit(lambda: ext.quant_fp8(h8, hq, hsc), iters)
+        print(f"{'quant_fp8':<28} {f'R={R}':<18} {us:9.2f}")
+        us = timeit(lambda: ext.heads_gemm_fp8(hq, hsc, fw8.wq, fw8.wsc,
+                                               fw8.bcat, hh8), iters)
+        gf = R * 1024 * 3200 * 2 / 1e9
+        print(f"{'heads_gemm_fp8':<28} {f'R={R}':<18} {us:9.2f}  "
+              f"({gf / us * 1e6 / 1e3:.2f} TF)")
+
     us = timeit(lambda: ext.noisy_material4(pack, rng), iters)
     print(f"{'noisy_material4':<28} {'4 layers':<18} {us:9.2f}")
     us = timeit(lambda: fw.pack_weights(), iters)

@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "env.hip"),
         os.path.join(CSRC, "model_fwd.hip"),
         os.path.join(CSRC, "model_bwd.hip"),
+        os.path.join(CSRC, "model_fp8.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

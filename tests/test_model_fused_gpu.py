@@ -118,3 +118,37 @@ def test_fused_matches_eager_bf16_closely():
     err = (out - z16.float()).abs().max().item()
     scale = z16.float().abs().max().item()
     assert err < 0.04 * max(scale, 1.0) + 0.02, (err, scale)
+
+
+def test_fp8_heads_parity():
+    """MX-fp8 heads GEMM (config 5 path): block-scaled e4m3 quantiles within
+    fp8 tolerance of the bf16 eager module; act/theta modes stay consistent."""
+    A = 6
+    _, net32, net16 = make_nets(A, seed=21)
+    from rainbowiqn_amd.runtime.fused_forward import FusedForward
+
+    fw = FusedForward(net16, "cuda", fp8_heads=True)
+    B, nq = 32, 8
+    x = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
+    taus = torch.rand(B * nq, 1, device="cuda")
+    z16, _ = net16(x, nq, taus=taus)
+    out = torch.empty(B, nq, A, device="cuda")
+    fw.quantiles(x, taus.view(-1), nq, out)
+    torch.cuda.synchronize()
+    err = (out - z16.float()).abs().max().item()
+    scale = z16.float().abs().max().item()
+    assert torch.isfinite(out).all()
+    # e4m3 has ~2 significant digits; the K=3136 reduction averages it out
+    assert err < 0.10 * max(scale, 1.0) + 0.05, (err, scale)
+    # internal consistency: act (mode 1) == argmax of mean quantiles (mode 0)
+    E, K = 64, 32
+    xa = torch.rand(E, 4, 84, 84, device="cuda").to(torch.bfloat16)
+    ta = torch.rand(E * K, device="cuda")
+    q = torch.empty(E, K, A, device="cuda")
+    fw.quantiles(xa, ta, K, q)
+    actions = torch.zeros(E, dtype=torch.int64, device="cuda")
+    fw.act(xa, ta, K, actions)
+    torch.cuda.synchronize()
+    expect = q.mean(dim=1).argmax(dim=1)
+    agree = (actions == expect).float().mean().item()
+    assert agree == 1.0, agree
