@@ -119,11 +119,12 @@ __global__ __launch_bounds__(256) void heads_gemm_fp8_kernel(
         if (kt + 1 < 25) stage((kt + 1) & 1, (kt + 1) * 128);
         const unsigned char* As = smem + (kt & 1) * (2 * 128 * 128);
         const unsigned char* Bs = As + 128 * 128;
-        // lane's 32-byte A fragment: [row][l4*32 .. +32) = pieces 2*l4, 2*l4+1
+        // lane's 32-byte fragment = chunks at k 16*l4 and 64+16*l4 (probed
+        // hardware layout, scripts/mx_probe.hip): pieces l4 and 4+l4
         #pragma unroll
         for (int fm = 0; fm < 4; ++fm) {
             const int row = wr * 64 + fm * 16 + l15;
-            const int pa0 = (l4 * 2) ^ (row & 7), pa1 = (l4 * 2 + 1) ^ (row & 7);
+            const int pa0 = l4 ^ (row & 7), pa1 = (4 + l4) ^ (row & 7);
             i32x4 alo = *(const i32x4*)&As[row * 128 + pa0 * 16];
             i32x4 ahi = *(const i32x4*)&As[row * 128 + pa1 * 16];
             i32x8 a = {alo[0], alo[1], alo[2], alo[3],
@@ -132,8 +133,8 @@ __global__ __launch_bounds__(256) void heads_gemm_fp8_kernel(
             #pragma unroll
             for (int fn = 0; fn < 4; ++fn) {
                 const int nrow = wc * 64 + fn * 16 + l15;
-                const int pb0 = (l4 * 2) ^ (nrow & 7),
-                          pb1 = (l4 * 2 + 1) ^ (nrow & 7);
+                const int pb0 = l4 ^ (nrow & 7),
+                          pb1 = (4 + l4) ^ (nrow & 7);
                 i32x4 blo = *(const i32x4*)&Bs[nrow * 128 + pb0 * 16];
                 i32x4 bhi = *(const i32x4*)&Bs[nrow * 128 + pb1 * 16];
                 i32x8 b = {blo[0], blo[1], blo[2], blo[3],

@@ -53,46 +53,44 @@ static float e4m3_to_f32(unsigned char v) {
 
 int main() {
     unsigned char hA[16 * 128], hB[16 * 128], hsA[64], hsB[64];
-    // A[i][k] = small value depending on (i,k); B[j][k] likewise (ASYMMETRIC)
     for (int i = 0; i < 16; ++i)
         for (int k = 0; k < 128; ++k) {
             hA[i * 128 + k] = f32_to_e4m3(0.25f * ((i + 2 * k) % 5 - 2));
             hB[i * 128 + k] = f32_to_e4m3(0.125f * ((3 * i + k) % 7 - 3));
         }
-    for (int i = 0; i < 64; ++i) { hsA[i] = 127; hsB[i] = 127; }
-    hsA[5] = 128;  // row 1, block 1 of A scaled by 2 -- tests scale mapping
     unsigned char *dA, *dB, *dsA, *dsB; float* dD;
     hipMalloc(&dA, sizeof hA); hipMalloc(&dB, sizeof hB);
     hipMalloc(&dsA, 64); hipMalloc(&dsB, 64); hipMalloc(&dD, 256 * 4);
     hipMemcpy(dA, hA, sizeof hA, hipMemcpyHostToDevice);
     hipMemcpy(dB, hB, sizeof hB, hipMemcpyHostToDevice);
-    hipMemcpy(dsA, hsA, 64, hipMemcpyHostToDevice);
-    hipMemcpy(dsB, hsB, 64, hipMemcpyHostToDevice);
-    hipLaunchKernelGGL(mx_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dsA, dsB, dD);
-    float hD[256];
-    hipMemcpy(hD, dD, sizeof hD, hipMemcpyDeviceToHost);
-    // reference: D[i][j] = sum_k scaleA(i, k/32) a[i][k] * scaleB(j, k/32) b[j][k]
-    float ref[256];
-    double maxerr = 0;
-    for (int i = 0; i < 16; ++i)
-        for (int j = 0; j < 16; ++j) {
-            double scl = 0;
-            for (int k = 0; k < 128; ++k) {
-                float fa = e4m3_to_f32(hA[i * 128 + k]) *
-                           exp2f((float)(hsA[i * 4 + k / 32] - 127));
-                float fb = e4m3_to_f32(hB[j * 128 + k]) *
-                           exp2f((float)(hsB[j * 4 + k / 32] - 127));
-                scl += (double)fa * fb;
+    float hD[256], ref[256];
+
+    // one-hot B probes for each scale byte kb: which (lane, byte) operand
+    // positions (reported as my linear k = lane*32 + byte) it multiplies
+    for (int kb = 0; kb < 4; ++kb) {
+        for (int i = 0; i < 64; ++i) { hsA[i] = 127; hsB[i] = 127; }
+        hsA[1 * 4 + kb] = 128;
+        hipMemcpy(dsA, hsA, 64, hipMemcpyHostToDevice);
+        hipMemcpy(dsB, hsB, 64, hipMemcpyHostToDevice);
+        printf("scale byte kb=%d multiplies k:", kb);
+        for (int p = 0; p < 8; ++p) {
+            unsigned char oneB[16 * 128];
+            for (int j = 0; j < 16; ++j)
+                for (int k = 0; k < 128; ++k)
+                    oneB[j * 128 + k] = (k == j + 16 * p) ? f32_to_e4m3(1.0f) : 0;
+            hipMemcpy(dB, oneB, sizeof oneB, hipMemcpyHostToDevice);
+            hipLaunchKernelGGL(mx_kernel, dim3(1), dim3(64), 0, 0, dA, dB,
+                               dsA, dsB, dD);
+            hipMemcpy(hD, dD, sizeof hD, hipMemcpyDeviceToHost);
+            for (int j = 0; j < 16; ++j) {
+                int k = j + 16 * p;
+                float base = e4m3_to_f32(hA[1 * 128 + k]);
+                float got = hD[1 * 16 + j];
+                if (base != 0 && fabsf(got / base - 2.0f) < 0.01f)
+                    printf(" %d", k);
             }
-            ref[i * 16 + j] = (float)scl;
-            maxerr = fmax(maxerr, fabs(scl - hD[i * 16 + j]));
         }
-    printf("max err: %g\n", maxerr);
-    printf("D[0][0..4]:   %g %g %g %g\n", hD[0], hD[1], hD[2], hD[3]);
-    printf("ref[0][0..4]: %g %g %g %g\n", ref[0], ref[1], ref[2], ref[3]);
-    printf("D[1][0..4]:   %g %g %g %g\n", hD[16], hD[17], hD[18], hD[19]);
-    printf("ref[1][0..4]: %g %g %g %g\n", ref[16], ref[17], ref[18], ref[19]);
-    printf("D[5][5] %g ref %g | D[2][7] %g ref %g\n", hD[5 * 16 + 5],
-           ref[5 * 16 + 5], hD[2 * 16 + 7], ref[2 * 16 + 7]);
+        printf("\n");
+    }
     return 0;
 }
