@@ -67,6 +67,8 @@ def main():
     p.add_argument("--chunk-steps", type=int, default=8, help="env steps per round")
     p.add_argument("--learn-per-round", type=int, default=8)
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--fp8-heads", action="store_true",
+                   help="MX-fp8 NoisyLinear heads GEMM (BASELINE config 5)")
     args = p.parse_args()
 
     on_gpu = torch.cuda.is_available() and args.device != "cpu"
@@ -82,6 +84,7 @@ def main():
         dtype="bf16" if on_gpu else "fp32",
         device="cuda" if on_gpu else "cpu",
         seed=int(os.environ.get("RANK", 0)) + 1,
+        fp8_heads=args.fp8_heads,
     )
     comm = Communicator(backend="auto")
     torch.manual_seed(cfg.seed)
@@ -136,7 +139,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no throughput numbers
-            "dtype": cfg.dtype,
+            "dtype": "bf16+fp8-heads" if cfg.fp8_heads else cfg.dtype,
             "data": "synthetic",
             "learner_grad_steps_per_sec": round(grad_per_sec, 2),
             "env_steps_per_sec": round(env_steps / elapsed_max, 1),
