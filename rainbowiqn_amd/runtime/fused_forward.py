@@ -153,7 +153,7 @@ class FusedForward:
         """Greedy action from precomputed conv features (lets the learner
         batch s and s' through one conv_encode call)."""
         B = featv.shape[0]
-        _, h, hh = self._bufs(B, nq)
+        _, h, hh = self._bufs(B, nq)[:3]
         self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn,
                            featv.contiguous(), h, nq)
         self.ext.heads_gemm(h, self.wcat, self.bcat, hh)

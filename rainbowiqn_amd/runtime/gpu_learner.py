@@ -173,7 +173,7 @@ class GpuLearner:
         idx, s, a, r, s2, nt, w = self.replay.sample()
         # ---- online conv over [s; s'] in ONE call (activations saved for
         # the backward; the s' half feeds the double-Q select) -------------
-        _, h_g, hh_g = fo._bufs(b, N)
+        _, h_g, hh_g = fo._bufs(b, N)[:3]
         ext.conv_encode(self.replay.states_cat, fo.w1, fo.b1, fo.w2, fo.b2,
                         fo.w3p, fo.b3, self.feat_cat, self.c1g, self.c2g)
         feat_g = self.feat_cat[:b]
