@@ -99,7 +99,8 @@ class _EvalProxy:
 
 def train_apex(cfg: Config, comm: Optional[Communicator] = None,
                n_envs: int = 16, results_dir: Optional[str] = None,
-               max_rounds: Optional[int] = None, log_interval: int = 50):
+               max_rounds: Optional[int] = None, log_interval: int = 50,
+               metrics_port: Optional[int] = None):
     """Run Ape-X training to cfg.T_max agent steps.  Returns the runner."""
     device = cfg.resolve_device()
     if device == "cuda":
@@ -113,6 +114,9 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
     if cfg.memory_save_path and comm.is_learner:
         _load_memory(runner, cfg.memory_save_path)
 
+    from .utils.metrics import Metrics
+
+    metrics = Metrics((metrics_port + comm.rank) if metrics_port else None)
     evaluator = None
     proxy = None
     if comm.is_learner and cfg.evaluation_interval <= cfg.T_max:
@@ -148,6 +152,12 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
                 log(f"T={T} rounds={rounds} grad_steps={runner.grad_steps} "
                     f"loss={loss:.4f} steps/s={sps:.0f} "
                     f"frames/s={sps * 4:.0f}")
+                metrics.update(d_env_steps=T - T_log,
+                               d_grad_steps=runner.grad_steps,
+                               loss=loss if loss == loss else None,
+                               replay_size=(len(runner.replay)
+                                            if runner.replay else None),
+                               steps_per_sec=sps)
                 t_log, T_log = time.perf_counter(), T
         if max_rounds is not None and rounds >= max_rounds:
             break
@@ -179,12 +189,15 @@ def main(argv=None):
     p.add_argument("--n-envs", type=int, default=16,
                    help="vectorized env streams per actor rank")
     p.add_argument("--max-rounds", type=int, default=None)
+    p.add_argument("--metrics-port", type=int, default=None,
+                   help="Prometheus exporter port (rank offset added)")
     ns, rest = p.parse_known_args(argv)
     cfg = parse_args(rest)
     comm = Communicator(backend=cfg.backend)
     results_dir = os.path.join("results", cfg.id)
     runner = train_apex(cfg, comm, n_envs=ns.n_envs, results_dir=results_dir,
-                        max_rounds=ns.max_rounds)
+                        max_rounds=ns.max_rounds,
+                        metrics_port=ns.metrics_port)
     if comm.is_learner:
         log(f"done: grad_steps={runner.grad_steps}")
     comm.close()

@@ -1,0 +1,48 @@
+"""Optional Prometheus metrics exporter (the reference has
+stdout prints only; we add native steps/sec + loss + replay gauges).
+
+Enabled with --metrics-port; no-op if prometheus_client is missing.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+
+class Metrics:
+    def __init__(self, port: Optional[int] = None):
+        self.enabled = False
+        if not port:
+            return
+        try:
+            from prometheus_client import Counter, Gauge, start_http_server
+        except ImportError:  # pragma: no cover
+            return
+        start_http_server(port)
+        self.env_steps = Counter("riqn_env_steps_total",
+                                 "agent env steps on this rank")
+        self.grad_steps = Counter("riqn_grad_steps_total",
+                                  "learner gradient steps")
+        self.loss = Gauge("riqn_loss", "latest quantile-Huber loss")
+        self.replay_size = Gauge("riqn_replay_transitions",
+                                 "transitions resident in replay")
+        self.steps_per_sec = Gauge("riqn_env_steps_per_sec",
+                                   "recent env steps/sec (this rank)")
+        self.enabled = True
+
+    def update(self, d_env_steps: int = 0, d_grad_steps: int = 0,
+               loss: Optional[float] = None,
+               replay_size: Optional[int] = None,
+               steps_per_sec: Optional[float] = None) -> None:
+        if not self.enabled:
+            return
+        if d_env_steps:
+            self.env_steps.inc(d_env_steps)
+        if d_grad_steps:
+            self.grad_steps.inc(d_grad_steps)
+        if loss is not None:
+            self.loss.set(loss)
+        if replay_size is not None:
+            self.replay_size.set(replay_size)
+        if steps_per_sec is not None:
+            self.steps_per_sec.set(steps_per_sec)

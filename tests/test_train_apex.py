@@ -90,3 +90,15 @@ def test_svg_plot_fallback(tmp_path):
     _plot_rewards_svg([100, 200, 300], [[1, 2], [2, 3], [3, 4]], str(p), "t")
     body = p.read_text()
     assert "<svg" in body and "polyline" in body
+
+
+def test_metrics_exporter(tmp_path):
+    import urllib.request
+
+    cfg = tiny_cfg()
+    runner = train_apex(cfg, n_envs=8, results_dir=str(tmp_path),
+                        max_rounds=25, log_interval=10, metrics_port=18123)
+    body = urllib.request.urlopen(
+        "http://127.0.0.1:18123/metrics", timeout=5).read().decode()
+    assert "riqn_grad_steps_total" in body
+    assert "riqn_env_steps_per_sec" in body
