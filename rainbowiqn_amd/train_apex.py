@@ -131,6 +131,7 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
 
     t_log = time.perf_counter()
     T_log = 0
+    g_log = 0
     while T < cfg.T_max:
         runner.run_round()
         rounds += 1
@@ -153,12 +154,13 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
                     f"loss={loss:.4f} steps/s={sps:.0f} "
                     f"frames/s={sps * 4:.0f}")
                 metrics.update(d_env_steps=T - T_log,
-                               d_grad_steps=runner.grad_steps,
+                               d_grad_steps=runner.grad_steps - g_log,
                                loss=loss if loss == loss else None,
                                replay_size=(len(runner.replay)
                                             if runner.replay else None),
                                steps_per_sec=sps)
                 t_log, T_log = time.perf_counter(), T
+                g_log = runner.grad_steps
         if max_rounds is not None and rounds >= max_rounds:
             break
 
