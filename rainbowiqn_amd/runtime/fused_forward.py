@@ -30,6 +30,21 @@ from ..ops import get_extension
 
 
 class FusedForward:
+    @classmethod
+    def for_net(cls, net: RainbowIQN, device,
+                fp8_heads: bool = False) -> "FusedForward":
+        """One FusedForward per net.  Two instances on one net would each
+        re-point weight_noisy to their own concat buffer and the loser's
+        GEMM operands would never see another noise reset (colocated
+        actor+learner share the online net)."""
+        fw = getattr(net, "_fused_forward", None)
+        if fw is not None:
+            assert fw.fp8_heads == fp8_heads, "fp8 flag mismatch on shared net"
+            return fw
+        fw = cls(net, device, fp8_heads=fp8_heads)
+        net._fused_forward = fw
+        return fw
+
     def __init__(self, net: RainbowIQN, device, fp8_heads: bool = False):
         self.ext = get_extension()
         self.net = net

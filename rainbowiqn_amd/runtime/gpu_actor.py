@@ -63,7 +63,8 @@ class GpuActorLoop:
                 and getattr(cfg, "fused_forward", True)):
             from .fused_forward import FusedForward
 
-            self.fused = FusedForward(net, d, fp8_heads=getattr(cfg, 'fp8_heads', False))
+            self.fused = FusedForward.for_net(
+                net, d, fp8_heads=getattr(cfg, 'fp8_heads', False))
 
     def _round_body(self) -> None:
         cfg = self.cfg

@@ -91,10 +91,10 @@ class GpuLearner:
             from .fused_forward import FusedForward
 
             fp8 = getattr(cfg, "fp8_heads", False)
-            self.fused_online = FusedForward(self.online, self.device,
-                                             fp8_heads=fp8)
-            self.fused_target = FusedForward(self.target, self.device,
-                                             fp8_heads=fp8)
+            self.fused_online = FusedForward.for_net(self.online, self.device,
+                                                     fp8_heads=fp8)
+            self.fused_target = FusedForward.for_net(self.target, self.device,
+                                                     fp8_heads=fp8)
             self.a_star = torch.zeros(b, dtype=torch.int64, device=self.device)
             self.theta_p_buf = torch.zeros(b, self.Np, device=self.device)
             if getattr(cfg, "fused_backward", True):

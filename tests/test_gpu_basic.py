@@ -179,3 +179,26 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
     assert err < 0.05 * max(q32.abs().max().item(), 1.0) + 0.02, err
     agent.train()
     lrn.online.train()
+
+
+@pytest.mark.gpu
+def test_colocated_shares_fused_forward():
+    """Colocated actor+learner share ONE FusedForward on the online net --
+    two instances would leave the learner's GEMM weight buffers stale after
+    noise resets (regression test)."""
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    cfg = Config(device="cuda", dtype="bf16", learn_start=512,
+                 memory_capacity=8192, learn_per_round=2)
+    runner = ApexRunner(cfg, Communicator(backend="auto"), n_envs=16,
+                        chunk_steps=8)
+    assert runner.gactor.fused is runner.glearner.fused_online
+    wcat_before = runner.glearner.fused_online.wcat.clone()
+    for _ in range(6):
+        runner.run_round()
+    torch.cuda.synchronize()
+    # noise regenerated + Adam steps: the learner's GEMM operand moved
+    assert not torch.equal(wcat_before, runner.glearner.fused_online.wcat)
+    assert runner.grad_steps > 0
