@@ -250,12 +250,30 @@ class ApexRunner:
                     if comm.world_size > 1 and j == 0:
                         self.flat_weights = flat
                 self._actor_flats.append(flat)
+                shared = comm.world_size == 1 and M == 1
                 self.gactors.append(
                     GpuActorLoop(cfg, net, self.n_envs, self.chunk_steps,
                                  device=dev, use_graph=cfg.hipgraph,
-                                 seed_offset=comm.rank * 7717 + j * 131071))
+                                 seed_offset=comm.rank * 7717 + j * 131071,
+                                 external_weights=shared))
             self.gactor = self.gactors[0]
             self.actor = _ActorSteps(self.gactors)
+        # colocated overlap: actor round runs on its own HIP stream,
+        # concurrent with the learner's grad steps (one-round bounded
+        # staleness -- the same contract as the multi-GPU split).  The
+        # outgoing chunk is double-buffered so the learner appends round
+        # k-1's chunk while round k records.
+        self.overlap = (comm.world_size == 1 and M == 1 and self.gactors
+                        and self.glearner is not None
+                        and getattr(cfg, "overlap_colocated", True))
+        if self.overlap:
+            d = self.gactors[0].device
+            c = self.n_envs * self.chunk_steps
+            self._astream = torch.cuda.Stream(device=d)
+            self._staging_f = [torch.zeros(c, 84, 84, dtype=torch.uint8,
+                                           device=d) for _ in range(2)]
+            self._staging_m = [torch.zeros(c, 3, device=d) for _ in range(2)]
+            self._round_idx = 0
 
     # -- one Ape-X round ----------------------------------------------------
     def run_round(self, learn: bool = True) -> None:
@@ -267,6 +285,8 @@ class ApexRunner:
     def _run_round_fast(self, learn: bool) -> None:
         cfg = self.cfg
         comm = self.comm
+        if getattr(self, "overlap", False):
+            return self._run_round_overlap(learn)
         M = self.n_local_actors
         local_f, local_m = [], []
         if self.gactors:
@@ -321,6 +341,35 @@ class ApexRunner:
 
         if comm.world_size > 1:
             comm.broadcast_weights(self.flat_weights, src=0)
+
+    def _run_round_overlap(self, learn: bool) -> None:
+        """Colocated round with actor/learner stream concurrency."""
+        cfg = self.cfg
+        gl = self.glearner
+        loop = self.gactors[0]
+        j = self._round_idx % 2
+        prev = (self._round_idx - 1) % 2
+        # learner stream: append LAST round's chunk, then grad steps; these
+        # small-grid kernels fill the CUs the actor's big grids leave idle
+        if self._round_idx > 0:
+            gl.append_actor_chunk(0, self._staging_f[prev],
+                                  meta=self._staging_m[prev])
+        # actor stream: round + chunk stage-out, concurrent with the learner
+        self._astream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._astream):
+            loop.run_round()
+            self._staging_f[j].copy_(loop.chunk_frames, non_blocking=True)
+            self._staging_m[j].copy_(loop.chunk_meta, non_blocking=True)
+        if learn and len(gl.replay) >= cfg.learn_start:
+            gl.learn_steps(cfg.learn_per_round)
+            self.grad_steps = gl.grad_steps
+            self.losses.append(gl.last_loss())
+            tgt = max(1, cfg.target_update // max(1, cfg.replay_frequency))
+            if self.grad_steps % tgt < cfg.learn_per_round:
+                gl.update_target()
+                gl.maintenance()
+        torch.cuda.current_stream().wait_stream(self._astream)
+        self._round_idx += 1
 
     def _run_round_ref(self, learn: bool) -> None:
         cfg = self.cfg

@@ -25,13 +25,19 @@ FRAME_PX = 84 * 84
 class GpuActorLoop:
     def __init__(self, cfg, net: RainbowIQN, n_envs: int, chunk_steps: int,
                  device: str = "cuda", use_graph: bool = True,
-                 seed_offset: int = 0, p_done: float = 4.0 / 512.0):
+                 seed_offset: int = 0, p_done: float = 4.0 / 512.0,
+                 external_weights: bool = False):
+        # external_weights: the owner of `net` (the colocated learner)
+        # regenerates noise and weight packs; this loop only READS them --
+        # required when the actor graph runs CONCURRENTLY with the learner
+        # stream (writes from two streams to the shared buffers would race).
         assert chunk_steps % cfg.history_length == 0, (
             "chunk_steps must be a multiple of history_length (ring phase)"
         )
         self.cfg = cfg
         self.ext = get_extension()
         self.net = net
+        self.external_weights = external_weights
         self.E = n_envs
         self.chunk_steps = chunk_steps
         self.K = cfg.num_quantile_samples
@@ -68,9 +74,10 @@ class GpuActorLoop:
 
     def _round_body(self) -> None:
         cfg = self.cfg
-        self.reset_noise()  # in-graph: device philox, fixed launch sequence
-        if self.fused is not None:
-            self.fused.pack_weights()
+        if not self.external_weights:
+            self.reset_noise()  # in-graph: device philox noise regeneration
+            if self.fused is not None:
+                self.fused.pack_weights()
         newest = self._newest
         for s in range(self.chunk_steps):
             self.ext.stack_gather(
