@@ -56,7 +56,10 @@ def test_apex_single_gpu_rounds(gcfg):
     for _ in range(3):
         runner.run_round()
     torch.cuda.synchronize()
-    assert len(runner.replay) == 192
+    # overlap mode appends with a one-round staging lag (+ the graph-capture
+    # warmup round's chunk is recorded but round 0's append lands in round 1)
+    expect = 3 * 64 if not runner.overlap else 2 * 64
+    assert len(runner.replay) == expect, len(runner.replay)
     assert runner.grad_steps > 0
     assert all(np.isfinite(l) for l in runner.losses)
 
