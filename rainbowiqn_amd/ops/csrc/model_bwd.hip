@@ -256,6 +256,7 @@ __global__ __launch_bounds__(256) void gemm_dwcat_kernel(
     unsigned short* __restrict__ dw_mu_a, unsigned short* __restrict__ dw_sg_a,
     int R) {
     __shared__ unsigned short As[64 * 32];  // [k(32)][m(64)] transposed image
+    __shared__ unsigned short Bs2[32 * 64]; // [k(32)][n(64)] staged h slice
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;
@@ -265,14 +266,17 @@ __global__ __launch_bounds__(256) void gemm_dwcat_kernel(
     f32x4 acc[2][2] = {{{0, 0, 0, 0}, {0, 0, 0, 0}},
                        {{0, 0, 0, 0}, {0, 0, 0, 0}}};
     for (int k0 = 0; k0 < R; k0 += 32) {
-        // stage A^T: As[k][m] = dhh[k0+k][bm+m], k<32, m<64 (rows coalesced)
-        // (single iteration for 256 threads; kept as loop for clarity)
-        for (int i = t * 8; i < 32 * 64; i += 256 * 8) {
-            int kk = i / 64, mm = i % 64;
+        // stage A^T and the B slice, rows coalesced
+        {
+            int kk = (t * 8) / 64, mm = (t * 8) % 64;
             u16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
             if (k0 + kk < R)
                 v = *(const u16x8*)(dhh + (int64_t)(k0 + kk) * 1024 + bm + mm);
             *(u16x8*)&As[kk * 64 + mm] = v;
+            u16x8 w = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (k0 + kk < R)
+                w = *(const u16x8*)(hbuf + (int64_t)(k0 + kk) * 3136 + bn + mm);
+            *(u16x8*)&Bs2[kk * 64 + mm] = w;
         }
         __syncthreads();
         #pragma unroll
@@ -284,13 +288,11 @@ __global__ __launch_bounds__(256) void gemm_dwcat_kernel(
             bf16x8 a = as_bf16x8(av);
             #pragma unroll
             for (int fn = 0; fn < 2; ++fn) {
-                const int n = bn + wc * 32 + fn * 16 + l15;
-                const int kk = k0 + l4 * 8;
-                u16x8 bv = {0, 0, 0, 0, 0, 0, 0, 0};
+                const int n_l = wc * 32 + fn * 16 + l15;
+                u16x8 bv;
                 #pragma unroll
                 for (int u = 0; u < 8; ++u)
-                    if (kk + u < R)
-                        bv[u] = hbuf[(int64_t)(kk + u) * 3136 + n];
+                    bv[u] = Bs2[(l4 * 8 + u) * 64 + n_l];
                 acc[fm][fn] = mfma16(a, as_bf16x8(bv), acc[fm][fn]);
             }
         }
@@ -445,199 +447,207 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
 #define BW_C1 12800  // 32*400 ush
 #define BW_C2 5184   // 64*81
 #define BW_P3 3136   // 64*49
-__global__ __launch_bounds__(512, 1) void conv_bwd_data_kernel(
+
+// conv backward data, split for occupancy (B=32 states would otherwise fill
+// only 32 CUs):
+//   conv_bwd_dc2: grid (B, 3), unit = (m-tile, n-half) of conv3^T; also
+//     computes dpre3 = dfeat * relu-mask (y==0 writes dpre3g + db3)
+//   conv_bwd_dc1: grid (B, 7), m-tiles of conv2^T over dc2g
+// Transposed weights come fragment-major (pack_bwd_frag_kernel).
+__global__ __launch_bounds__(256) void conv_bwd_dc2_kernel(
     const float* __restrict__ dfeat,         // [B,3136] f32
     const unsigned short* __restrict__ feat, // [B,3136] bf16 (relu mask)
-    const unsigned short* __restrict__ c1g,  // [B][32*400]
-    const unsigned short* __restrict__ c2g,  // [B][64*81]
-    const unsigned short* __restrict__ w3tp, // [64c][64n*16]
-    const unsigned short* __restrict__ w2tp, // [32c][64n*16]
+    const unsigned short* __restrict__ c2g,  // [B][64*81] (relu mask)
+    const unsigned short* __restrict__ w3tp, // fragment-major [128*512]
     unsigned short* __restrict__ dpre3g,     // [64][B*49]
     unsigned short* __restrict__ dc2g,       // [64][B*81]
-    unsigned short* __restrict__ dc1g,       // [32][B*400]
-    float* __restrict__ db1s, float* __restrict__ db2s,
-    float* __restrict__ db3s,                // f32 scratch [32],[64],[64]
+    float* __restrict__ db2s, float* __restrict__ db3s,
     int B) {
-    __shared__ unsigned short lds[BW_C1 + BW_C2 + BW_P3 + BW_C2 + BW_C1];
-    unsigned short* c1s = lds;                      // c1out
-    unsigned short* c2s = lds + BW_C1;              // c2out
-    unsigned short* p3s = c2s + BW_C2;              // dpre3
-    unsigned short* d2s = p3s + BW_P3;              // dc2
-    unsigned short* d1s = d2s + BW_C2;              // dc1
+    __shared__ unsigned short p3s[BW_P3];
     const int b = blockIdx.x;
+    const int y = blockIdx.y;
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int l15 = lane & 15, l4 = lane >> 4;
 
-    // stage activations + dpre3
-    #pragma unroll 8
-    for (int i = t * 4; i < BW_C1; i += 512 * 4)
-        *(u16x4*)&c1s[i] = *(const u16x4*)(c1g + (int64_t)b * BW_C1 + i);
     #pragma unroll 4
-    for (int i = t * 4; i < BW_C2; i += 512 * 4)
-        *(u16x4*)&c2s[i] = *(const u16x4*)(c2g + (int64_t)b * BW_C2 + i);
-    #pragma unroll 4
-    for (int i = t; i < BW_P3; i += 512) {
+    for (int i = t; i < BW_P3; i += 256) {
         float g = dfeat[(int64_t)b * 3136 + i];
         float m = bf(feat[(int64_t)b * 3136 + i]) > 0.0f ? 1.0f : 0.0f;
         p3s[i] = f32_to_bf16(g * m);
     }
     __syncthreads();
-    // db3 + write dpre3g (channel-major global layout for the dW GEMM)
-    if (wave < 8) {
-        for (int n = wave; n < 64; n += 8) {
-            float s = 0.0f;
-            for (int px = lane; px < 49; px += 64) s += bf(p3s[n * 49 + px]);
-            s = wave_allreduce_sum(s);
-            if (lane == 0) atomicAdd(&db3s[n], s);
-            for (int px = lane; px < 49; px += 64)
-                dpre3g[(int64_t)n * (B * 49) + b * 49 + px] = p3s[n * 49 + px];
+    if (y == 0) {  // db3 + dpre3g once
+        if (wave < 8) {
+            for (int n = wave; n < 64; n += 4) {
+                float s = 0.0f;
+                for (int px = lane; px < 49; px += 64) s += bf(p3s[n * 49 + px]);
+                s = wave_allreduce_sum(s);
+                if (lane == 0) atomicAdd(&db3s[n], s);
+                for (int px = lane; px < 49; px += 64)
+                    dpre3g[(int64_t)n * (B * 49) + b * 49 + px] =
+                        p3s[n * 49 + px];
+            }
         }
     }
 
-    // ---- dc2: M=81 (6 mt), N=64c (4 nt), K=1024 (32 ks) ------------------
-    // unit = (mt, n-half): 12 units over 8 waves
-    for (int u = wave; u < 12; u += 8) {
-        const int mt = u >> 1, nh = u & 1;
-        f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
-        const int m = mt * 16 + l15;
-        const int iy = m / 9, ix = m % 9;
-        const bool mv = m < 81;
-        // tap geometry is fixed per lane; only the n-plane advances (+2 per
-        // k-step): precomputed incremental gather addresses + zero masks
-        int ga[8];
-        unsigned short okm[8];
-        {
-            const int n0 = l4 >> 1;
-            const int t16 = (l4 & 1) * 8;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
-                const int oy = iy - ky, ox = ix - kx;
-                const bool ok = mv && ky < 3 && kx < 3 && oy >= 0 && oy < 7 &&
-                                ox >= 0 && ox < 7;
-                ga[j] = n0 * 49 + (ok ? oy * 7 + ox : 0);
-                okm[j] = ok ? 0xFFFFu : 0;
-            }
-        }
-        bf16x8 bbuf[2][2];
+    // units y*4 + wave in [0,12): (mt, nh)
+    const int u = y * 4 + wave;
+    const int mt = u >> 1, nh = u & 1;
+    f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+    const int m = mt * 16 + l15;
+    const int iy = m / 9, ix = m % 9;
+    const bool mv = m < 81;
+    int ga[8];
+    unsigned short okm[8];
+    {
+        const int n0 = l4 >> 1;
+        const int t16 = (l4 & 1) * 8;
         #pragma unroll
-        for (int nt = 0; nt < 2; ++nt)
-            bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
-                w3tp + ((nh * 2 + nt) << 9) + lane * 8));
-        #pragma unroll 2
-        for (int ks = 0; ks < 32; ++ks) {
-            const int cur = ks & 1;
-            if (ks + 1 < 32) {
-                #pragma unroll
-                for (int nt = 0; nt < 2; ++nt)
-                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                        w3tp + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
-            }
-            u16x8 av;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) av[j] = p3s[ga[j]] & okm[j];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) ga[j] += 2 * 49;
-            bf16x8 a = as_bf16x8(av);
-            acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
-            acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
-        }
-        #pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
-            const int c = (nh * 2 + nt) * 16 + l15;
-            #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                const int mm = mt * 16 + l4 * 4 + j;
-                if (mm < 81) {
-                    float mask = bf(c2s[c * 81 + mm]) > 0.0f ? 1.0f : 0.0f;
-                    d2s[c * 81 + mm] = f32_to_bf16(acc[nt][j] * mask);
-                }
-            }
+        for (int j = 0; j < 8; ++j) {
+            const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+            const int oy = iy - ky, ox = ix - kx;
+            const bool ok = mv && ky < 3 && kx < 3 && oy >= 0 && oy < 7 &&
+                            ox >= 0 && ox < 7;
+            ga[j] = n0 * 49 + (ok ? oy * 7 + ox : 0);
+            okm[j] = ok ? 0xFFFFu : 0;
         }
     }
-    __syncthreads();
-    // db2 + dc2g
-    if (wave < 8) {
-        for (int n = wave; n < 64; n += 8) {
-            float s = 0.0f;
-            for (int px = lane; px < 81; px += 64) s += bf(d2s[n * 81 + px]);
-            s = wave_allreduce_sum(s);
-            if (lane == 0) atomicAdd(&db2s[n], s);
-            for (int px = lane; px < 81; px += 64)
-                dc2g[(int64_t)n * (B * 81) + b * 81 + px] = d2s[n * 81 + px];
+    bf16x8 bbuf[2][2];
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+        bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
+            w3tp + ((nh * 2 + nt) << 9) + lane * 8));
+    #pragma unroll 2
+    for (int ks = 0; ks < 32; ++ks) {
+        const int cur = ks & 1;
+        if (ks + 1 < 32) {
+            #pragma unroll
+            for (int nt = 0; nt < 2; ++nt)
+                bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                    w3tp + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
         }
+        u16x8 av;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) av[j] = p3s[ga[j]] & okm[j];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) ga[j] += 2 * 49;
+        bf16x8 a = as_bf16x8(av);
+        acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
+        acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
     }
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+        const int c = (nh * 2 + nt) * 16 + l15;
+        float bsum = 0.0f;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const int mm = mt * 16 + l4 * 4 + j;
+            if (mm < 81) {
+                float mask =
+                    bf(c2g[(int64_t)b * BW_C2 + c * 81 + mm]) > 0.0f ? 1.0f
+                                                                     : 0.0f;
+                float v = acc[nt][j] * mask;
+                dc2g[(int64_t)c * (B * 81) + b * 81 + mm] = f32_to_bf16(v);
+                bsum += v;
+            }
+        }
+        // db2: sum over pixels; lanes cover distinct (c, mm) pairs
+        bsum = bsum;  // reduce over the 4 l4 groups of the same c
+        #pragma unroll
+        for (int off = 16; off < 64; off <<= 1)
+            bsum += __shfl_xor(bsum, off, 64);
+        if (l4 == 0) atomicAdd(&db2s[c], bsum);
+    }
+}
 
-    // ---- dc1: M=400 (25 mt), N=32c (2 nt), K=1024 (32 ks) ----------------
-    for (int mt = wave; mt < 25; mt += 8) {
-        f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
-        const int m = mt * 16 + l15;
-        const int iy = m / 20, ix = m % 20;
-        const bool mv = m < 400;
-        int ga[8];
-        unsigned short okm[8];
-        {
-            const int n0 = l4 >> 1;
-            const int t16 = (l4 & 1) * 8;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
-                const int ry = iy - ky, rx = ix - kx;
-                const int oy = ry >> 1, ox = rx >> 1;
-                const bool ok = mv && ry >= 0 && rx >= 0 && (ry & 1) == 0 &&
-                                (rx & 1) == 0 && oy < 9 && ox < 9;
-                ga[j] = n0 * 81 + (ok ? oy * 9 + ox : 0);
-                okm[j] = ok ? 0xFFFFu : 0;
-            }
-        }
-        bf16x8 bbuf[2][2];
-        #pragma unroll
-        for (int nt = 0; nt < 2; ++nt)
-            bbuf[0][nt] = as_bf16x8(
-                *(const u16x8*)(w2tp + (nt << 9) + lane * 8));
-        #pragma unroll 2
-        for (int ks = 0; ks < 32; ++ks) {
-            const int cur = ks & 1;
-            if (ks + 1 < 32) {
-                #pragma unroll
-                for (int nt = 0; nt < 2; ++nt)
-                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                        w2tp + (((ks + 1) * 2 + nt) << 9) + lane * 8));
-            }
-            u16x8 av;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) av[j] = d2s[ga[j]] & okm[j];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) ga[j] += 2 * 81;
-            bf16x8 a = as_bf16x8(av);
-            acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
-            acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
-        }
-        #pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
-            const int c = nt * 16 + l15;
-            #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                const int mm = mt * 16 + l4 * 4 + j;
-                if (mm < 400) {
-                    float mask = bf(c1s[c * 400 + mm]) > 0.0f ? 1.0f : 0.0f;
-                    d1s[c * 400 + mm] = f32_to_bf16(acc[nt][j] * mask);
-                }
-            }
-        }
+__global__ __launch_bounds__(256) void conv_bwd_dc1_kernel(
+    const unsigned short* __restrict__ dc2g, // [64][B*81]
+    const unsigned short* __restrict__ c1g,  // [B][32*400] (relu mask)
+    const unsigned short* __restrict__ w2tp, // fragment-major [64*512]
+    unsigned short* __restrict__ dc1g,       // [32][B*400]
+    float* __restrict__ db1s,
+    int B) {
+    __shared__ unsigned short d2s[BW_C2];
+    const int b = blockIdx.x;
+    const int y = blockIdx.y;
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;
+    const int l15 = lane & 15, l4 = lane >> 4;
+
+    // stage this state's dc2 from its channel-major layout
+    #pragma unroll 4
+    for (int i = t; i < BW_C2; i += 256) {
+        int n = i / 81, px = i % 81;
+        d2s[i] = dc2g[(int64_t)n * (B * 81) + b * 81 + px];
     }
     __syncthreads();
-    // db1 + dc1g
-    if (wave < 8) {
-        for (int n = wave; n < 32; n += 8) {
-            float s = 0.0f;
-            for (int px = lane; px < 400; px += 64) s += bf(d1s[n * 400 + px]);
-            s = wave_allreduce_sum(s);
-            if (lane == 0) atomicAdd(&db1s[n], s);
-            for (int px = lane; px < 400; px += 64)
-                dc1g[(int64_t)n * (B * 400) + b * 400 + px] = d1s[n * 400 + px];
+
+    const int mt = y * 4 + wave;  // 25 m-tiles over 28 wave-slots
+    if (mt >= 25) return;
+    f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+    const int m = mt * 16 + l15;
+    const int iy = m / 20, ix = m % 20;
+    const bool mv = m < 400;
+    int ga[8];
+    unsigned short okm[8];
+    {
+        const int n0 = l4 >> 1;
+        const int t16 = (l4 & 1) * 8;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
+            const int ry = iy - ky, rx = ix - kx;
+            const int oy = ry >> 1, ox = rx >> 1;
+            const bool ok = mv && ry >= 0 && rx >= 0 && (ry & 1) == 0 &&
+                            (rx & 1) == 0 && oy < 9 && ox < 9;
+            ga[j] = n0 * 81 + (ok ? oy * 9 + ox : 0);
+            okm[j] = ok ? 0xFFFFu : 0;
         }
+    }
+    bf16x8 bbuf[2][2];
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+        bbuf[0][nt] = as_bf16x8(
+            *(const u16x8*)(w2tp + (nt << 9) + lane * 8));
+    #pragma unroll 2
+    for (int ks = 0; ks < 32; ++ks) {
+        const int cur = ks & 1;
+        if (ks + 1 < 32) {
+            #pragma unroll
+            for (int nt = 0; nt < 2; ++nt)
+                bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
+                    w2tp + (((ks + 1) * 2 + nt) << 9) + lane * 8));
+        }
+        u16x8 av;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) av[j] = d2s[ga[j]] & okm[j];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) ga[j] += 2 * 81;
+        bf16x8 a = as_bf16x8(av);
+        acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
+        acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
+    }
+    #pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+        const int c = nt * 16 + l15;
+        float bsum = 0.0f;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const int mm = mt * 16 + l4 * 4 + j;
+            if (mm < 400) {
+                float mask =
+                    bf(c1g[(int64_t)b * BW_C1 + c * 400 + mm]) > 0.0f ? 1.0f
+                                                                      : 0.0f;
+                float v = acc[nt][j] * mask;
+                dc1g[(int64_t)c * (B * 400) + b * 400 + mm] = f32_to_bf16(v);
+                bsum += v;
+            }
+        }
+        #pragma unroll
+        for (int off = 16; off < 64; off <<= 1)
+            bsum += __shfl_xor(bsum, off, 64);
+        if (l4 == 0) atomicAdd(&db1s[c], bsum);
     }
 }
 
@@ -850,14 +860,17 @@ void riqn_conv_bwd_data(const float* dfeat, const void* feat, const void* c1g,
                         const void* c2g, const void* w3tp, const void* w2tp,
                         void* dpre3g, void* dc2g, void* dc1g, float* db1s,
                         float* db2s, float* db3s, int B, void* stream) {
-    hipLaunchKernelGGL(conv_bwd_data_kernel, dim3(B), dim3(512), 0,
-                       (hipStream_t)stream, dfeat,
-                       (const unsigned short*)feat, (const unsigned short*)c1g,
+    hipStream_t s = (hipStream_t)stream;
+    hipLaunchKernelGGL(conv_bwd_dc2_kernel, dim3(B, 3), dim3(256), 0, s,
+                       dfeat, (const unsigned short*)feat,
                        (const unsigned short*)c2g,
-                       (const unsigned short*)w3tp,
-                       (const unsigned short*)w2tp, (unsigned short*)dpre3g,
-                       (unsigned short*)dc2g, (unsigned short*)dc1g, db1s,
-                       db2s, db3s, B);
+                       (const unsigned short*)w3tp, (unsigned short*)dpre3g,
+                       (unsigned short*)dc2g, db2s, db3s, B);
+    hipLaunchKernelGGL(conv_bwd_dc1_kernel, dim3(B, 7), dim3(256), 0, s,
+                       (const unsigned short*)dc2g,
+                       (const unsigned short*)c1g,
+                       (const unsigned short*)w2tp, (unsigned short*)dc1g,
+                       db1s, B);
 }
 
 void riqn_conv_bwd_dw(int layer, const void* dY, const void* act, float* dWs,
