@@ -63,7 +63,7 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--n-envs", type=int, default=512, help="env streams per actor rank")
+    p.add_argument("--n-envs", type=int, default=1024, help="env streams per actor rank")
     p.add_argument("--chunk-steps", type=int, default=8, help="env steps per round")
     p.add_argument("--learn-per-round", type=int, default=8)
     p.add_argument("--device", type=str, default=None)
