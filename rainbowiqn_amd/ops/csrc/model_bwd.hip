@@ -393,11 +393,14 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
     const unsigned short* __restrict__ cosb,  // [R,64] bf16 (from tau_embed)
     unsigned short* __restrict__ dwiqn,       // [3136,64] bf16 out
     int R) {
+    // one m-tile (16 n-rows) per WG; the 4 waves split the K=R reduction
+    // and combine through LDS (196 WGs instead of 49)
     __shared__ unsigned short cs[256 * 64];  // staged cos table (R <= 256)
+    __shared__ float part[4][16 * 64];       // per-wave D tiles
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int l15 = lane & 15, l4 = lane >> 4;
-    const int mt = blockIdx.x * 4 + wave;  // n-tile per wave
+    const int mt = blockIdx.x;               // 196 m-tiles
     const int n = mt * 16 + l15;
     #pragma unroll 8
     for (int i = t * 4; i < R * 64; i += 256 * 4)
@@ -406,8 +409,9 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
 
     f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
                     f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+    const int kbeg = wave * (R / 4);
     #pragma unroll 2
-    for (int k0 = 0; k0 < R; k0 += 32) {
+    for (int k0 = kbeg; k0 < kbeg + R / 4; k0 += 32) {
         const int kk = k0 + l4 * 8;
         u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
         #pragma unroll
@@ -423,14 +427,18 @@ __global__ __launch_bounds__(256) void tau_dw_kernel(
             acc[it] = mfma16(a, as_bf16x8(bv), acc[it]);
         }
     }
+    // cross-wave combine: D element (nn = l4*4+j within tile, i = it*16+l15)
     #pragma unroll
-    for (int it = 0; it < 4; ++it) {
+    for (int it = 0; it < 4; ++it)
         #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            const int nn = mt * 16 + l4 * 4 + j;  // D row (n)
-            const int i = it * 16 + l15;
-            if (nn < 3136) dwiqn[(int64_t)nn * 64 + i] = f32_to_bf16(acc[it][j]);
-        }
+        for (int j = 0; j < 4; ++j)
+            part[wave][(l4 * 4 + j) * 64 + it * 16 + l15] = acc[it][j];
+    __syncthreads();
+    for (int i = t; i < 16 * 64; i += 256) {
+        const int nn = mt * 16 + i / 64, col = i % 64;
+        if (nn < 3136)
+            dwiqn[(int64_t)nn * 64 + col] = f32_to_bf16(
+                part[0][i] + part[1][i] + part[2][i] + part[3][i]);
     }
 }
 
@@ -851,7 +859,7 @@ void riqn_bias_grad_cols(const void* X, int R, int C, void* db_mu,
 
 void riqn_tau_dw(const void* dphi, const void* cosb, void* dwiqn, int R,
                  void* stream) {
-    hipLaunchKernelGGL(tau_dw_kernel, dim3(3136 / 64), dim3(256), 0,
+    hipLaunchKernelGGL(tau_dw_kernel, dim3(3136 / 16), dim3(256), 0,
                        (hipStream_t)stream, (const unsigned short*)dphi,
                        (const unsigned short*)cosb, (unsigned short*)dwiqn, R);
 }

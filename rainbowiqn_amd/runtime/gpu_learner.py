@@ -75,9 +75,12 @@ class GpuLearner:
             [cfg.seed * 7919 + 13, 0], dtype=torch.int64, device=self.device
         )
         b = self.B
-        self.taus_online = torch.empty(b * self.N, 1, device=self.device)
-        self.taus_target = torch.empty(b * self.Np, 1, device=self.device)
-        self.taus_select = torch.empty(b * self.K, 1, device=self.device)
+        # one contiguous buffer -> ONE rand_uniform launch per step
+        self._taus_all = torch.empty(b * (self.N + self.Np + self.K), 1,
+                                     device=self.device)
+        self.taus_online = self._taus_all[: b * self.N]
+        self.taus_target = self._taus_all[b * self.N : b * (self.N + self.Np)]
+        self.taus_select = self._taus_all[b * (self.N + self.Np):]
         self.loss_static = torch.zeros(1, device=self.device)
         self.use_graph = use_graph
         self.graph: Optional[torch.cuda.CUDAGraph] = None
@@ -162,9 +165,7 @@ class GpuLearner:
         ext = self.ext
         b, N, A = self.B, self.N, self.action_space
         fo, ft = self.fused_online, self.fused_target
-        ext.rand_uniform(self.taus_online, self.rng_state)
-        ext.rand_uniform(self.taus_target, self.rng_state)
-        ext.rand_uniform(self.taus_select, self.rng_state)
+        ext.rand_uniform(self._taus_all, self.rng_state)
         self._reset_noise_device(self.online)
         self._reset_noise_device(self.target)
         fo.pack_weights()
@@ -237,9 +238,7 @@ class GpuLearner:
         cfg = self.cfg
         b = self.B
         ext = self.ext
-        ext.rand_uniform(self.taus_online, self.rng_state)
-        ext.rand_uniform(self.taus_target, self.rng_state)
-        ext.rand_uniform(self.taus_select, self.rng_state)
+        ext.rand_uniform(self._taus_all, self.rng_state)
         self._reset_noise_device(self.online)
         self._reset_noise_device(self.target)
         idx, s, a, r, s2, nt, w = self.replay.sample()
