@@ -100,15 +100,16 @@ __global__ __launch_bounds__(256) void zgrad_kernel(
     unsigned short* __restrict__ db_mu_v, unsigned short* __restrict__ db_sg_v,
     unsigned short* __restrict__ dw_mu_a, unsigned short* __restrict__ dw_sg_a,
     unsigned short* __restrict__ db_mu_a, unsigned short* __restrict__ db_sg_a) {
-    __shared__ float part[4][512];
+    __shared__ float part[4][128];
     __shared__ float bpart[4];
     const int row = blockIdx.x;  // 0 = V, 1..A = advantage rows
+    const int cq = blockIdx.y;   // column quarter [cq*128, +128)
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const float invA = 1.0f / (float)A;
-    const int off = (row == 0) ? 0 : 512;
-    // each wave accumulates rows wave::4, lane covers 8 coalesced columns
-    float wacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    const int off = ((row == 0) ? 0 : 512) + cq * 128;
+    // each wave accumulates rows wave::4, lane covers 2 coalesced columns
+    float wacc[2] = {0, 0};
     float wb = 0.0f;
     #pragma unroll 4
     for (int r = wave; r < R; r += 4) {
@@ -117,49 +118,35 @@ __global__ __launch_bounds__(256) void zgrad_kernel(
             int a = (int)actions[r / nq];
             c *= ((a == row - 1) ? 1.0f : 0.0f) - invA;
         }
-        const unsigned short* hrow = hh + (int64_t)r * 1024 + off + lane * 8;
-        u16x8 hv8 = *(const u16x8*)hrow;
-        #pragma unroll
-        for (int u = 0; u < 8; ++u) wacc[u] += c * bf(hv8[u]);
+        const unsigned short* hrow = hh + (int64_t)r * 1024 + off + lane * 2;
+        wacc[0] += c * bf(hrow[0]);
+        wacc[1] += c * bf(hrow[1]);
         wb += c;
     }
-    #pragma unroll
-    for (int u = 0; u < 8; ++u) part[wave][lane * 8 + u] = wacc[u];
+    part[wave][lane * 2] = wacc[0];
+    part[wave][lane * 2 + 1] = wacc[1];
     if (lane == 0) bpart[wave] = wb;
     __syncthreads();
-    float acc[2];
-    #pragma unroll
-    for (int u = 0; u < 2; ++u) {
-        int j = t * 2 + u;
-        acc[u] = part[0][j] + part[1][j] + part[2][j] + part[3][j];
-    }
+    if (t >= 128) return;
+    const int j0 = cq * 128 + t;  // column within the 512-wide layer
+    float acc1 = part[0][t] + part[1][t] + part[2][t] + part[3][t];
     float bacc = bpart[0] + bpart[1] + bpart[2] + bpart[3];
     if (row == 0) {
-        #pragma unroll
-        for (int u = 0; u < 2; ++u) {
-            int j = t * 2 + u;
-            dw_mu_v[j] = f32_to_bf16(acc[u]);
-            dw_sg_v[j] = f32_to_bf16(acc[u] * bf(weps_v[j]));
-        }
-        if (t == 0) {
+        dw_mu_v[j0] = f32_to_bf16(acc1);
+        dw_sg_v[j0] = f32_to_bf16(acc1 * bf(weps_v[j0]));
+        if (t == 0 && cq == 0) {
             db_mu_v[0] = f32_to_bf16(bacc);
             db_sg_v[0] = f32_to_bf16(bacc * bf(beps_v[0]));
         }
     } else {
         int i = row - 1;
-        #pragma unroll
-        for (int u = 0; u < 2; ++u) {
-            int j = t * 2 + u;
-            dw_mu_a[i * 512 + j] = f32_to_bf16(acc[u]);
-            dw_sg_a[i * 512 + j] = f32_to_bf16(acc[u] * bf(weps_a[i * 512 + j]));
-        }
-        if (t == 0) {
+        dw_mu_a[i * 512 + j0] = f32_to_bf16(acc1);
+        dw_sg_a[i * 512 + j0] = f32_to_bf16(acc1 * bf(weps_a[i * 512 + j0]));
+        if (t == 0 && cq == 0) {
             db_mu_a[i] = f32_to_bf16(bacc);
             db_sg_a[i] = f32_to_bf16(bacc * bf(beps_a[i]));
         }
     }
-    // note: bacc accumulated per thread covers ALL rows -> bias uses t==0's
-    // full sum only if every thread saw every r; it did (r-loop not split).
 }
 
 // ---------------------------------------------------------------------------
@@ -803,7 +790,7 @@ void riqn_zgrad(const float* dtheta, const int64_t* actions, const void* hh,
                 const void* beps_v, const void* beps_a, void* dw_mu_v,
                 void* dw_sg_v, void* db_mu_v, void* db_sg_v, void* dw_mu_a,
                 void* dw_sg_a, void* db_mu_a, void* db_sg_a, void* stream) {
-    hipLaunchKernelGGL(zgrad_kernel, dim3(1 + A), dim3(256), 0,
+    hipLaunchKernelGGL(zgrad_kernel, dim3(1 + A, 4), dim3(256), 0,
                        (hipStream_t)stream, dtheta, actions,
                        (const unsigned short*)hh, R, A, nq,
                        (const unsigned short*)weps_v,

@@ -666,10 +666,10 @@ __global__ __launch_bounds__(256) void heads_finish_kernel(
         *(u16x4*)&wzs[i] = *(const u16x4*)(wz + i);
     __syncthreads();
 
-    const int r0 = blockIdx.x * 16;
-    for (int rr = wave; rr < 16; rr += 4) {
-        const int r = r0 + rr;
-        if (r >= R) break;
+    const int r0 = blockIdx.x * 4;
+    {
+        const int r = r0 + wave;  // one row per wave
+        if (r >= R) return;
         const unsigned short* hrow = hh + (int64_t)r * 1024;
         float hv[8], ha[8];
         #pragma unroll
@@ -832,12 +832,12 @@ void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
                            (const unsigned short*)wz,
                            (const unsigned short*)bz, A, nq, act_io);
     } else if (mode == 2) {
-        hipLaunchKernelGGL(heads_finish_kernel<2>, dim3((R + 15) / 16),
+        hipLaunchKernelGGL(heads_finish_kernel<2>, dim3((R + 3) / 4),
                            dim3(256), 0, s, (const unsigned short*)hh,
                            (const unsigned short*)wz,
                            (const unsigned short*)bz, R, A, nq, q_out, act_io);
     } else {
-        hipLaunchKernelGGL(heads_finish_kernel<0>, dim3((R + 15) / 16),
+        hipLaunchKernelGGL(heads_finish_kernel<0>, dim3((R + 3) / 4),
                            dim3(256), 0, s, (const unsigned short*)hh,
                            (const unsigned short*)wz,
                            (const unsigned short*)bz, R, A, nq, q_out, act_io);
