@@ -172,3 +172,20 @@ class GpuReplay:
 
     def leaf_priorities(self) -> torch.Tensor:
         return self.tree[self.tree_size : self.tree_size + self.capacity]
+
+    def check_invariants(self, atol: float = 1e-2) -> None:
+        """Debug assertion pass over the sum-tree (the on-GPU
+        shared replay needs its own race checks): every internal node must
+        equal the sum of its children and leaves must be non-negative.
+        Host-side; call outside captured graphs."""
+        t = self.tree.cpu()
+        leaves = t[self.tree_size:]
+        assert (leaves >= 0).all(), "negative leaf priority"
+        import torch as _t
+
+        node = t[1 : self.tree_size]
+        kids = t[2 : 2 * self.tree_size].view(-1, 2).sum(dim=1)
+        bad = (node - kids).abs() > atol * (1.0 + node.abs())
+        assert not bool(bad.any()), (
+            f"sum-tree invariant violated at {int(bad.nonzero()[0])}"
+        )

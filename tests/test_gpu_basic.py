@@ -205,3 +205,24 @@ def test_colocated_shares_fused_forward():
     # noise regenerated + Adam steps: the learner's GEMM operand moved
     assert not torch.equal(wcat_before, runner.glearner.fused_online.wcat)
     assert runner.grad_steps > 0
+
+
+@pytest.mark.gpu
+def test_replay_tree_invariants_after_traffic():
+    """Sum-tree node = sum(children) after interleaved appends, samples and
+    priority updates (SURVEY.md 6.2 race-check row)."""
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.replay.gpu_replay import GpuReplay
+
+    cfg = Config(device="cuda", dtype="bf16", batch_size=32)
+    rep = GpuReplay(cfg, n_rings=2, capacity=16384, device="cuda",
+                    out_dtype=torch.bfloat16)
+    cf = torch.randint(0, 255, (1024, 84, 84), dtype=torch.uint8, device="cuda")
+    cm = torch.rand(1024, 3, device="cuda")
+    for it in range(6):
+        rep.append_actor_chunk(it % 2, cf, meta=cm)
+        if len(rep) > 2048:
+            idx, *_ = rep.sample()
+            rep.update_priorities(idx, torch.rand(32, device="cuda") + 0.1)
+    torch.cuda.synchronize()
+    rep.check_invariants()
