@@ -34,7 +34,7 @@ c2g = torch.zeros(2 * B2, 5184, dtype=torch.bfloat16, device="cuda")
 dfeat = torch.rand(B2, 3136, device="cuda")
 w3tp = torch.zeros(64, 1024, dtype=torch.bfloat16, device="cuda")
 w2tp = torch.zeros(32, 1024, dtype=torch.bfloat16, device="cuda")
-ext.pack_bwd_weights(fw.w3, fw.w2, w3tp, w2tp)
+ext.pack_bwd_weights(fw.w3raw, fw.w2raw, w3tp, w2tp)
 dpre3g = torch.zeros(64, B2 * 49, dtype=torch.bfloat16, device="cuda")
 dc2g = torch.zeros(64, B2 * 81, dtype=torch.bfloat16, device="cuda")
 dc1g = torch.zeros(32, B2 * 400, dtype=torch.bfloat16, device="cuda")
