@@ -155,7 +155,9 @@ __global__ __launch_bounds__(256) void zgrad_kernel(
 //   dfeat[r/nq][n] += dh * phi[r][n]          (f32 atomic)
 // B-tile (Wcat slice) is staged transposed through LDS.
 // ---------------------------------------------------------------------------
-#define DX_BK 64
+// glds-staged (2 buffers, raw barrier + counted waits, XOR source swizzle
+// -- the heads-GEMM recipe).  A = dhh rows; B = Wcat k-rows ([k][n] slice is
+// row-major contiguous, no transpose needed).
 __global__ __launch_bounds__(256) void gemm_dx_kernel(
     const unsigned short* __restrict__ dhh,   // [R,1024]
     const unsigned short* __restrict__ wcat,  // [1024,3136]
@@ -164,49 +166,72 @@ __global__ __launch_bounds__(256) void gemm_dx_kernel(
     unsigned short* __restrict__ dphi,        // [R,3136] out
     float* __restrict__ dfeat,                // [B,3136] f32, pre-zeroed
     int R, int nq) {
-    __shared__ unsigned short Bs[DX_BK * 64];  // [k][n] image
+    __shared__ unsigned short smem[2 * 2 * 64 * 64];  // 2 bufs x (A|B)
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;
     const int l15 = lane & 15, l4 = lane >> 4;
     const int bm = blockIdx.x * 64, bn = blockIdx.y * 64;
 
+    auto stage = [&](int buf, int k0) {
+        unsigned short* base = smem + buf * (2 * 64 * 64);
+        #pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            const int inst = wave * 2 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            {   // A: dhh[bm+row][k0 + piece*8]
+                const int gr = min(bm + row, R - 1);
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (dhh + (int64_t)gr * 1024 + k0 + piece * 8);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + inst * 512);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+            }
+            {   // B: Wcat[k0+row][bn + piece*8]  ([k][n] image)
+                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                    (wcat + (int64_t)(k0 + row) * 3136 + bn + piece * 8);
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + 64 * 64 + inst * 512);
+                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+            }
+        }
+    };
+
     f32x4 acc[2][2] = {{{0, 0, 0, 0}, {0, 0, 0, 0}},
                        {{0, 0, 0, 0}, {0, 0, 0, 0}}};
-    for (int k0 = 0; k0 < 1024; k0 += DX_BK) {
-        // stage B = Wcat[k0..+64][bn..+64] as [k][n] (rows coalesced)
-        #pragma unroll 2
-        for (int i = t * 8; i < DX_BK * 64; i += 256 * 8) {
-            int kk = i / 64, nn = i % 64;
-            *(u16x8*)&Bs[kk * 64 + nn] =
-                *(const u16x8*)(wcat + (int64_t)(k0 + kk) * 3136 + bn + nn);
-        }
-        __syncthreads();
+    stage(0, 0);
+    for (int kt = 0; kt < 1024 / 64; ++kt) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + 1 < 1024 / 64) stage((kt + 1) & 1, (kt + 1) * 64);
+        const unsigned short* As = smem + (kt & 1) * (2 * 64 * 64);
+        const unsigned short* Bs = As + 64 * 64;
         #pragma unroll
         for (int ksub = 0; ksub < 2; ++ksub) {
             #pragma unroll
             for (int fm = 0; fm < 2; ++fm) {
-                const int row = bm + wr * 32 + fm * 16 + l15;
-                const int kk = k0 + ksub * 32 + l4 * 8;
-                u16x8 av = {0, 0, 0, 0, 0, 0, 0, 0};
-                if (row < R)
-                    av = *(const u16x8*)(dhh + (int64_t)row * 1024 + kk);
-                bf16x8 a = as_bf16x8(av);
+                const int row = wr * 32 + fm * 16 + l15;
+                const int piece = (ksub * 4 + l4) ^ (row & 7);
+                bf16x8 a = as_bf16x8(*(const u16x8*)&As[row * 64 + piece * 8]);
                 #pragma unroll
                 for (int fn = 0; fn < 2; ++fn) {
                     const int n_l = wc * 32 + fn * 16 + l15;
-                    // B-frag: 8 k-values (stride 64) for fixed n
-                    const int kk_l = ksub * 32 + l4 * 8;
+                    // B fragment: 8 k-values (rows of the [k][n] image) for
+                    // one n; per-element reads honour the source swizzle
                     u16x8 bv;
                     #pragma unroll
-                    for (int u = 0; u < 8; ++u)
-                        bv[u] = Bs[(kk_l + u) * 64 + n_l];
+                    for (int u = 0; u < 8; ++u) {
+                        const int krow = ksub * 32 + l4 * 8 + u;
+                        const int np = ((n_l >> 3) ^ (krow & 7)) * 8 + (n_l & 7);
+                        bv[u] = Bs[krow * 64 + np];
+                    }
                     acc[fm][fn] = mfma16(a, as_bf16x8(bv), acc[fm][fn]);
                 }
             }
         }
-        __syncthreads();
     }
+    __builtin_amdgcn_s_barrier();
     #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
         #pragma unroll
