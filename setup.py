@@ -38,7 +38,9 @@ ext = CUDAExtension(
 setup(
     name="rainbowiqn-amd",
     version="0.1.0",
-    packages=["rainbowiqn_amd"],
+    packages=["rainbowiqn_amd", "rainbowiqn_amd.models", "rainbowiqn_amd.ops",
+              "rainbowiqn_amd.parallel", "rainbowiqn_amd.replay",
+              "rainbowiqn_amd.runtime", "rainbowiqn_amd.utils"],
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=True)},
 )
