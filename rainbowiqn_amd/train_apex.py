@@ -100,7 +100,8 @@ class _EvalProxy:
 def train_apex(cfg: Config, comm: Optional[Communicator] = None,
                n_envs: int = 16, results_dir: Optional[str] = None,
                max_rounds: Optional[int] = None, log_interval: int = 50,
-               metrics_port: Optional[int] = None):
+               metrics_port: Optional[int] = None,
+               chunk_steps: Optional[int] = None):
     """Run Ape-X training to cfg.T_max agent steps.  Returns the runner."""
     device = cfg.resolve_device()
     if device == "cuda":
@@ -108,7 +109,8 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
     comm = comm or Communicator(backend=cfg.backend)
     torch.manual_seed(cfg.seed + comm.rank * 9973)
 
-    runner = ApexRunner(cfg, comm, n_envs=n_envs, device=device)
+    runner = ApexRunner(cfg, comm, n_envs=n_envs, device=device,
+                        chunk_steps=chunk_steps)
     if cfg.model and comm.is_learner:
         _load_model(runner, cfg.model)
     if cfg.memory_save_path and comm.is_learner:
@@ -193,13 +195,16 @@ def main(argv=None):
     p.add_argument("--max-rounds", type=int, default=None)
     p.add_argument("--metrics-port", type=int, default=None,
                    help="Prometheus exporter port (rank offset added)")
+    p.add_argument("--chunk-steps", type=int, default=8,
+                   help="env steps per Ape-X exchange round")
     ns, rest = p.parse_known_args(argv)
     cfg = parse_args(rest)
     comm = Communicator(backend=cfg.backend)
     results_dir = os.path.join("results", cfg.id)
     runner = train_apex(cfg, comm, n_envs=ns.n_envs, results_dir=results_dir,
                         max_rounds=ns.max_rounds,
-                        metrics_port=ns.metrics_port)
+                        metrics_port=ns.metrics_port,
+                        chunk_steps=ns.chunk_steps)
     if comm.is_learner:
         log(f"done: grad_steps={runner.grad_steps}")
     comm.close()
