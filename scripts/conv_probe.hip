@@ -68,7 +68,7 @@ __global__ __launch_bounds__(512, 1) void probe_kernel(
             #pragma unroll
             for (int ks = 0; ks < 8; ++ks)
                 b1f[nt][ks] = as_bf16x8(*(const u16x8*)(
-                    w1 + (nt * 16 + l15) * 256 + ks * 32 + l4 * 8));
+                    w1 + ((ks * 2 + nt) << 9) + lane * 8));
         for (int mt = wave; mt < 25; mt += 8) {
             f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
             const int m = mt * 16 + l15;
@@ -118,7 +118,7 @@ __global__ __launch_bounds__(512, 1) void probe_kernel(
             #pragma unroll
             for (int nt = 0; nt < 2; ++nt)
                 bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
-                    w2 + ((nh * 2 + nt) * 16 + l15) * 512 + l4 * 8));
+                    w2 + ((nh * 2 + nt) << 9) + lane * 8));
             #pragma unroll 2
             for (int ks = 0; ks < 16; ++ks) {
                 const int cur = ks & 1;
@@ -127,7 +127,7 @@ __global__ __launch_bounds__(512, 1) void probe_kernel(
                     #pragma unroll
                     for (int nt = 0; nt < 2; ++nt)
                         bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                            w2 + ((nh * 2 + nt) * 16 + l15) * 512 + k1));
+                            w2 + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
                 }
                 const int k0 = ks * 32 + l4 * 8;
                 const int c = k0 >> 4, t16 = k0 & 15;
@@ -174,7 +174,7 @@ __global__ __launch_bounds__(512, 1) void probe_kernel(
         #pragma unroll
         for (int nt = 0; nt < 2; ++nt)
             bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
-                w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + l4 * 8));
+                w3p + ((nh * 2 + nt) << 9) + lane * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 32; ++ks) {
             const int cur = ks & 1;
@@ -183,7 +183,7 @@ __global__ __launch_bounds__(512, 1) void probe_kernel(
                 #pragma unroll
                 for (int nt = 0; nt < 2; ++nt)
                     bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                        w3p + ((nh * 2 + nt) * 16 + l15) * 1024 + k1));
+                        w3p + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
             }
             const int k0 = ks * 32 + l4 * 8;
             const int c = k0 >> 4, t16 = k0 & 15;
