@@ -150,12 +150,13 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
     __syncthreads();
 
     // ---- conv2: M=81 (6 tiles), N=64 (4 tiles), K=512 (16 steps) ---------
-    // Work unit = (m-tile, n-half): 12 units over 8 waves.  k-outer loop
-    // with a register double-buffer on the B-fragments: the next k-step's
-    // weight loads are in flight while this step's MFMAs run.
-    for (int u = wave; u < 12; u += 8) {
-        const int mt = u >> 1, nh = u & 1;
-        f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+    // Work unit = (m-tile, n-quarter): 24 single-fragment units over 8
+    // waves (3 balanced rounds; the (m, n-half) split left half the waves
+    // idle on its second round).  k-outer loop with a register
+    // double-buffer on the B-fragments.
+    for (int u = wave; u < 24; u += 8) {
+        const int mt = u >> 2, nq2 = u & 3;
+        f32x4 acc1 = f32x4{0, 0, 0, 0};
         const int m = mt * 16 + l15;
         const int oy = m / 9, ox = m % 9;
         const bool mv = m < 81;
@@ -174,38 +175,29 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
                            : LDS_C1 + c0 * 20 * C1_STRIDE;
             }
         }
-        bf16x8 bbuf[2][2];
-        #pragma unroll
-        for (int nt = 0; nt < 2; ++nt)
-            bbuf[0][nt] = as_bf16x8(*(const u16x8*)(
-                w2 + ((nh * 2 + nt) << 9) + lane * 8));
+        bf16x8 bbuf[2];
+        bbuf[0] = as_bf16x8(*(const u16x8*)(w2 + (nq2 << 9) + lane * 8));
         #pragma unroll 2
         for (int ks = 0; ks < 16; ++ks) {
             const int cur = ks & 1;
-            if (ks + 1 < 16) {
-                #pragma unroll
-                for (int nt = 0; nt < 2; ++nt)
-                    bbuf[cur ^ 1][nt] = as_bf16x8(*(const u16x8*)(
-                        w2 + (((ks + 1) * 4 + nh * 2 + nt) << 9) + lane * 8));
-            }
+            if (ks + 1 < 16)
+                bbuf[cur ^ 1] = as_bf16x8(*(const u16x8*)(
+                    w2 + (((ks + 1) * 4 + nq2) << 9) + lane * 8));
             u16x8 av;
             #pragma unroll
             for (int j = 0; j < 8; ++j) av[j] = lds[ga[j]];
             #pragma unroll
             for (int j = 0; j < 8; ++j) ga[j] += 2 * 20 * C1_STRIDE;
-            bf16x8 a = as_bf16x8(av);
-            acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
-            acc[1] = mfma16(a, bbuf[cur][1], acc[1]);
+            acc1 = mfma16(as_bf16x8(av), bbuf[cur], acc1);
         }
-        #pragma unroll
-        for (int nt = 0; nt < 2; ++nt) {
-            const int n = (nh * 2 + nt) * 16 + l15;
+        {
+            const int n = nq2 * 16 + l15;
             const float bias = bf(b2[n]);
             #pragma unroll
             for (int j = 0; j < 4; ++j) {
                 const int mm = mt * 16 + l4 * 4 + j;
                 if (mm < 81) {
-                    unsigned short o = f32_to_bf16(fmaxf(acc[nt][j] + bias, 0.0f));
+                    unsigned short o = f32_to_bf16(fmaxf(acc1[j] + bias, 0.0f));
                     lds[LDS_C2 + (n * 9 + mm / 9) * C2_STRIDE + mm % 9] = o;
                     if (c2g != nullptr)
                         c2g[(int64_t)b * 5184 + n * 81 + mm] = o;
