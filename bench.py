@@ -69,6 +69,11 @@ def main():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--fp8-heads", action="store_true",
                    help="MX-fp8 NoisyLinear heads GEMM (BASELINE config 5)")
+    p.add_argument("--no-hipgraph", dest="hipgraph", action="store_false")
+    p.add_argument("--no-fused-forward", dest="fused_forward",
+                   action="store_false")
+    p.add_argument("--no-fused-backward", dest="fused_backward",
+                   action="store_false")
     args = p.parse_args()
 
     on_gpu = torch.cuda.is_available() and args.device != "cpu"
@@ -85,6 +90,9 @@ def main():
         device="cuda" if on_gpu else "cpu",
         seed=int(os.environ.get("RANK", 0)) + 1,
         fp8_heads=args.fp8_heads,
+        hipgraph=args.hipgraph,
+        fused_forward=args.fused_forward,
+        fused_backward=args.fused_backward,
     )
     comm = Communicator(backend="auto")
     torch.manual_seed(cfg.seed)
