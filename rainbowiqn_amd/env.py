@@ -200,9 +200,12 @@ class VectorSyntheticEnv:
 def make_env(cfg, device: str | torch.device = "cpu"):
     """Return the real ALE wrapper when available, else the synthetic env."""
     try:  # pragma: no cover - no ALE wheel on this box
-        import atari_py  # noqa: F401
+        try:
+            import atari_py  # noqa: F401
+        except ImportError:
+            import ale_py  # noqa: F401
 
-        from .env_ale import AtariEnv  # type: ignore
+        from .env_ale import AtariEnv
 
         return AtariEnv(cfg, device)
     except ImportError:
