@@ -112,3 +112,37 @@ def test_evaluator(cfg, tmp_path):
     assert np.isfinite(avg) and std >= 0
     assert (tmp_path / "rewards.html").exists()
     assert (tmp_path / "best_model.pth").exists()
+
+
+def test_reference_flag_surface_complete():
+    """Every reference argparse flag parses."""
+    from rainbowiqn_amd.config import parse_args
+
+    argv = [
+        "--id", "x", "--game", "breakout", "--seed", "7", "--disable-cuda",
+        "--T-max", "50000000", "--max-episode-length", "108000",
+        "--history-length", "4", "--hidden-size", "512", "--noisy-std", "0.1",
+        "--atoms", "51", "--quantile-embedding-dim", "64",
+        "--num-tau-samples", "8", "--num-tau-prime-samples", "8",
+        "--num-quantile-samples", "32", "--kappa", "1.0",
+        "--memory-capacity", "1000000", "--replay-frequency", "4",
+        "--priority-exponent", "0.5", "--priority-weight", "0.4",
+        "--multi-step", "3", "--learn-start", "80000", "--discount", "0.99",
+        "--target-update", "32000", "--reward-clip", "1", "--lr", "0.0000625",
+        "--adam-eps", "0.00015", "--batch-size", "32",
+        "--evaluation-interval", "100000", "--evaluation-episodes", "10",
+        "--evaluation-size", "500", "--render",
+        "--checkpoint-interval", "1000", "--memory-save-path", "/tmp/m.pt",
+        "--nb-actor", "4", "--id-actor", "1",
+        "--transition-chunk-size", "64", "--weight-sync-period", "25",
+        "--host-redis", "localhost", "--port-redis", "6379",
+    ]
+    cfg = parse_args(argv)
+    assert cfg.game == "breakout" and cfg.nb_actor == 4
+    assert cfg.lr == 0.0000625 and cfg.target_update == 32000
+    assert cfg.memory_save_path == "/tmp/m.pt"
+    # reference defaults preserved when unset
+    d = parse_args([])
+    assert d.T_max == int(50e6) and d.learn_start == int(80e3)
+    assert d.priority_exponent == 0.5 and d.multi_step == 3
+    assert d.adam_eps == 1.5e-4 and d.batch_size == 32
