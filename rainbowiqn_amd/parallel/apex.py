@@ -207,6 +207,11 @@ class ApexRunner:
 
         if not self.fast:
             self.n_local_actors = 1  # multi-local-actor is a fast-path feature
+        self.sharded = bool(getattr(cfg, "sharded_replay", False)) and \
+            comm.world_size > 1
+        if self.sharded:
+            assert self.n_local_actors == 1, "sharded mode uses one shard/rank"
+        self.shard = None
         if self.fast:
             self._init_fast(cfg, comm, dev, n_act, replay_capacity)
         else:
@@ -214,8 +219,23 @@ class ApexRunner:
             self.flat_weights = self.agent.flat_weights()
             if comm.is_actor:
                 self.actor = ActorState(cfg, comm, self.agent, n_envs, self.chunk_steps)
+                if self.sharded:
+                    from .sharded import ActorShardHost, actor_quota
+
+                    self.shard = ActorShardHost(
+                        cfg, comm,
+                        actor_quota(cfg.batch_size, comm.n_actors,
+                                    comm.actor_index()),
+                        cfg.learn_per_round, "cpu", fast=False)
             if comm.is_learner:
-                self.replay = ShardedReplay(cfg, comm.n_actors, replay_capacity)
+                if self.sharded:
+                    from .sharded import RemoteReplay
+
+                    self.replay = RemoteReplay(cfg, comm, cfg.learn_per_round,
+                                               "cpu", dtype=torch.float32)
+                else:
+                    self.replay = ShardedReplay(cfg, comm.n_actors,
+                                                replay_capacity)
 
     def _init_fast(self, cfg, comm, dev, n_act, replay_capacity):
         from ..models.model import RainbowIQN
@@ -231,8 +251,16 @@ class ApexRunner:
             if replay_capacity is not None:
                 cfg_cap = cfg
                 cfg_cap.memory_capacity = replay_capacity
+            ext_replay = None
+            if self.sharded:
+                from .sharded import RemoteReplay
+
+                dtt = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
+                ext_replay = RemoteReplay(cfg, comm, cfg.learn_per_round, dev,
+                                          dtype=dtt)
             self.glearner = GpuLearner(cfg, n_act, n_rings=comm.n_actors * M,
-                                       device=dev, use_graph=cfg.hipgraph)
+                                       device=dev, use_graph=cfg.hipgraph,
+                                       replay=ext_replay)
             self.replay = self.glearner.replay
             self.flat_weights = self.glearner.flat_online
         if comm.is_actor:
@@ -258,6 +286,14 @@ class ApexRunner:
                                  external_weights=shared))
             self.gactor = self.gactors[0]
             self.actor = _ActorSteps(self.gactors)
+            if self.sharded:
+                from .sharded import ActorShardHost, actor_quota
+
+                self.shard = ActorShardHost(
+                    cfg, comm,
+                    actor_quota(cfg.batch_size, comm.n_actors,
+                                comm.actor_index()),
+                    cfg.learn_per_round, dev, fast=True)
         # colocated overlap: actor round runs on its own HIP stream,
         # concurrent with the learner's grad steps (one-round bounded
         # staleness -- the same contract as the multi-GPU split).  The
@@ -277,10 +313,87 @@ class ApexRunner:
 
     # -- one Ape-X round ----------------------------------------------------
     def run_round(self, learn: bool = True) -> None:
-        if self.fast:
+        if self.sharded:
+            self._run_round_sharded(learn)
+        elif self.fast:
             self._run_round_fast(learn)
         else:
             self._run_round_ref(learn)
+
+    def _run_round_sharded(self, learn: bool) -> None:
+        """Config-5 round: replay shards live on the actor ranks; the
+        learner receives assembled batches and returns priorities
+        (parallel/sharded.py).  Collective order is identical on every
+        rank: counts all_gather -> batch P2P -> priorities P2P -> weight
+        broadcast."""
+        import torch.distributed as dist
+
+        cfg, comm = self.cfg, self.comm
+        self._sh_rounds = getattr(self, "_sh_rounds", 0) + 1
+        # 1. actors: generate experience, append to the LOCAL shard
+        if comm.is_actor:
+            if self.fast:
+                loop = self.gactors[0]
+                if self._actor_flats[0] is not None and \
+                        self._actor_flats[0] is not self.flat_weights:
+                    self._actor_flats[0].copy_(self.flat_weights)
+                loop.run_round()
+                self.shard.append_chunk(loop.chunk_frames,
+                                        meta=loop.chunk_meta)
+            else:
+                self.agent.reset_noise()
+                self.actor.run_chunk()
+                mc = self.actor.chunk_meta.cpu().numpy()
+                self.shard.append_chunk(
+                    self.actor.chunk_frames.cpu().numpy().astype("uint8"),
+                    actions=mc[:, 0].astype("int64"), rewards=mc[:, 1],
+                    terminals=mc[:, 2] > 0.5)
+            # local beta anneal (actors own the IS weights)
+            T = self._sh_rounds * self.chunk_steps * self.n_envs * comm.n_actors
+            span = max(1, cfg.T_max - cfg.learn_start)
+            self.shard.anneal_beta(
+                cfg.priority_weight + (1.0 - cfg.priority_weight)
+                * max(0, T - cfg.learn_start) / span)
+        # 2. shard counts
+        cnt = (self.shard.counts() if comm.is_actor
+               else torch.zeros(2)).to(comm.device)
+        gathered = [torch.empty_like(cnt) for _ in range(comm.world_size)]
+        dist.all_gather(gathered, cnt)
+        total = int(sum(float(g[0]) for g in gathered))
+        learning = learn and total >= cfg.learn_start
+        # 3. batches -> learner; grad steps; priorities -> actors
+        if learning:
+            if comm.is_actor:
+                self.shard.sample_and_send()
+                self.shard.recv_priorities()
+            else:
+                self.replay.set_len(total)
+                self.replay.recv_batches()
+                if self.fast:
+                    self.glearner.learn_steps(cfg.learn_per_round)
+                    self.grad_steps = self.glearner.grad_steps
+                    self.losses.append(self.glearner.last_loss())
+                    tgt = max(1, cfg.target_update
+                              // max(1, cfg.replay_frequency))
+                    if self.grad_steps % tgt < cfg.learn_per_round:
+                        self.glearner.update_target()
+                else:
+                    for _ in range(cfg.learn_per_round):
+                        self.losses.append(self.agent.learn(self.replay))
+                        self.grad_steps += 1
+                        if self.grad_steps % max(
+                                1, cfg.target_update
+                                // max(1, cfg.replay_frequency)) == 0:
+                            self.agent.update_target_net()
+                self.replay.send_priorities()
+        elif comm.is_learner:
+            self.replay.set_len(total)
+        # 4. weight flow
+        if not self.fast and comm.is_learner:
+            self.flat_weights = self.agent.flat_weights()
+        comm.broadcast_weights(self.flat_weights, src=0)
+        if not self.fast and comm.is_actor:
+            self.agent.load_flat_weights(self.flat_weights)
 
     def _run_round_fast(self, learn: bool) -> None:
         cfg = self.cfg

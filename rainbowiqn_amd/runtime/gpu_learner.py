@@ -34,7 +34,8 @@ def _noisy_layers(net: RainbowIQN):
 
 class GpuLearner:
     def __init__(self, cfg, action_space: int, n_rings: int = 1,
-                 device: str = "cuda", use_graph: bool = True):
+                 device: str = "cuda", use_graph: bool = True,
+                 replay=None):
         self.cfg = cfg
         self.ext = get_extension()
         self.device = torch.device(device)
@@ -70,7 +71,10 @@ class GpuLearner:
             self.opt = FusedAdam(self.flat_online, self.flat_grad, lr=cfg.lr,
                                  eps=cfg.adam_eps)
 
-        self.replay = GpuReplay(cfg, n_rings=n_rings, device=device, out_dtype=dt)
+        # replay may be external (sharded mode: a RemoteReplay facade over
+        # the actor ranks' shards -- parallel/sharded.py)
+        self.replay = replay if replay is not None else GpuReplay(
+            cfg, n_rings=n_rings, device=device, out_dtype=dt)
         self.rng_state = torch.tensor(
             [cfg.seed * 7919 + 13, 0], dtype=torch.int64, device=self.device
         )

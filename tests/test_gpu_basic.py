@@ -226,3 +226,35 @@ def test_replay_tree_invariants_after_traffic():
             rep.update_priorities(idx, torch.rand(32, device="cuda") + 0.1)
     torch.cuda.synchronize()
     rep.check_invariants()
+
+
+@pytest.mark.gpu
+def test_actor_shard_host_gpu():
+    """Config-5 actor-side shard on GPU: HBM-resident PER shard samples its
+    learner-batch quota with local frame-stack/n-step assembly."""
+    from types import SimpleNamespace
+
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.sharded import ActorShardHost
+
+    cfg = Config(device="cuda", dtype="bf16", memory_capacity=65536,
+                 batch_size=32, learn_per_round=4)
+    comm = SimpleNamespace(n_actors=7, rank=3)
+    host = ActorShardHost(cfg, comm, quota=5, steps=4, device="cuda",
+                          fast=True)
+    cf = torch.randint(0, 255, (2048, 84, 84), dtype=torch.uint8,
+                       device="cuda")
+    cm = torch.rand(2048, 3, device="cuda")
+    host.append_chunk(cf, meta=cm)
+    host.append_chunk(cf, meta=cm)
+    host.fill_buffers()
+    torch.cuda.synchronize()
+    assert host.st.shape == (4, 5, 4, 84, 84)
+    assert torch.isfinite(host.ww).all() and (host.ww > 0).all()
+    assert float(host.counts()[0]) == 4096
+    # priority write-back path
+    host.pri.uniform_(0.1, 1.0)
+    for s in range(4):
+        host.replay.update_priorities(host._idxs[s], host.pri[s])
+    torch.cuda.synchronize()
+    host.replay.check_invariants()
