@@ -113,8 +113,16 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
                         chunk_steps=chunk_steps)
     if cfg.model and comm.is_learner:
         _load_model(runner, cfg.model)
-    if cfg.memory_save_path and comm.is_learner:
-        _load_memory(runner, cfg.memory_save_path)
+    if cfg.memory_save_path:
+        if runner.sharded and comm.is_actor:
+            # sharded mode: each actor owns (and persists) its shard
+            p = f"{cfg.memory_save_path}.rank{comm.rank}"
+            if os.path.exists(p):
+                runner.shard.replay.load_state_dict(
+                    torch.load(p, map_location="cpu", weights_only=False))
+                log(f"replay shard restored from {p}")
+        elif comm.is_learner:
+            _load_memory(runner, cfg.memory_save_path)
 
     from .utils.metrics import Metrics
 
@@ -168,8 +176,11 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
 
     if comm.is_learner and results_dir:
         _save_model(runner, results_dir)
-        if cfg.memory_save_path:
+        if cfg.memory_save_path and not runner.sharded:
             _save_memory(runner, cfg.memory_save_path)
+    if runner.sharded and comm.is_actor and cfg.memory_save_path:
+        torch.save(runner.shard.replay.state_dict(),
+                   f"{cfg.memory_save_path}.rank{comm.rank}")
     return runner
 
 
