@@ -186,6 +186,7 @@ class ApexRunner:
         self.n_local_actors = n_local_actors
         self.grad_steps = 0
         self.losses = []
+        self.tracer = None  # RoundTracer (utils/trace.py) via set_tracer()
         self.actor = None
         self.replay = None
         self.glearner = None
@@ -312,6 +313,16 @@ class ApexRunner:
             self._round_idx = 0
 
     # -- one Ape-X round ----------------------------------------------------
+    def set_tracer(self, tracer) -> None:
+        """Attach a utils.trace.RoundTracer; round phases get named spans."""
+        self.tracer = tracer
+
+    def _sp(self, name: str):
+        from contextlib import nullcontext
+
+        return self.tracer.span(name) if self.tracer is not None \
+            else nullcontext()
+
     def run_round(self, learn: bool = True) -> None:
         if self.sharded:
             self._run_round_sharded(learn)
@@ -403,16 +414,18 @@ class ApexRunner:
         M = self.n_local_actors
         local_f, local_m = [], []
         if self.gactors:
-            for j, loop in enumerate(self.gactors):
-                if self._actor_flats[j] is not None:
-                    # refresh this actor's net from the latest weights
-                    src = (self.glearner.flat_online if comm.world_size == 1
-                           else self.flat_weights)
-                    if self._actor_flats[j] is not src:
-                        self._actor_flats[j].copy_(src)
-                loop.run_round()
-                local_f.append(loop.chunk_frames)
-                local_m.append(loop.chunk_meta)
+            with self._sp("actor"):
+                for j, loop in enumerate(self.gactors):
+                    if self._actor_flats[j] is not None:
+                        # refresh this actor's net from the latest weights
+                        src = (self.glearner.flat_online
+                               if comm.world_size == 1
+                               else self.flat_weights)
+                        if self._actor_flats[j] is not src:
+                            self._actor_flats[j].copy_(src)
+                    loop.run_round()
+                    local_f.append(loop.chunk_frames)
+                    local_m.append(loop.chunk_meta)
         else:
             # learner-only rank: shape templates for the P2P receives
             if not hasattr(self, "_dummy_chunks"):
@@ -425,26 +438,29 @@ class ApexRunner:
 
         frames_list, meta_list = [], []
         if comm.world_size > 1:
-            for j in range(M):
-                fl = comm.gather_chunks(local_f[j])
-                ml = comm.gather_chunks(local_m[j])
-                if comm.is_learner:
-                    frames_list.extend(fl)
-                    meta_list.extend(ml)
+            with self._sp("gather"):
+                for j in range(M):
+                    fl = comm.gather_chunks(local_f[j])
+                    ml = comm.gather_chunks(local_m[j])
+                    if comm.is_learner:
+                        frames_list.extend(fl)
+                        meta_list.extend(ml)
         else:
             frames_list, meta_list = local_f, local_m
         if comm.is_learner:
             # ring id = actor_rank * M + local actor (interleaved per j-gather:
             # frames_list order is [j0: ranks 1..W-1][j1: ...] when W > 1)
-            for i, (f, m) in enumerate(zip(frames_list, meta_list)):
-                if comm.world_size > 1:
-                    j, a = divmod(i, comm.n_actors)
-                    ring = a * M + j
-                else:
-                    ring = i
-                self.glearner.append_actor_chunk(ring, f, meta=m)
+            with self._sp("append"):
+                for i, (f, m) in enumerate(zip(frames_list, meta_list)):
+                    if comm.world_size > 1:
+                        j, a = divmod(i, comm.n_actors)
+                        ring = a * M + j
+                    else:
+                        ring = i
+                    self.glearner.append_actor_chunk(ring, f, meta=m)
             if learn and len(self.glearner.replay) >= cfg.learn_start:
-                self.glearner.learn_steps(cfg.learn_per_round)
+                with self._sp("learn"):
+                    self.glearner.learn_steps(cfg.learn_per_round)
                 self.grad_steps = self.glearner.grad_steps
                 self.losses.append(self.glearner.last_loss())
                 tgt = max(1, cfg.target_update // max(1, cfg.replay_frequency))
@@ -453,7 +469,8 @@ class ApexRunner:
                     self.glearner.maintenance()
 
         if comm.world_size > 1:
-            comm.broadcast_weights(self.flat_weights, src=0)
+            with self._sp("broadcast"):
+                comm.broadcast_weights(self.flat_weights, src=0)
 
     def _run_round_overlap(self, learn: bool) -> None:
         """Colocated round with actor/learner stream concurrency."""
@@ -488,8 +505,9 @@ class ApexRunner:
         cfg = self.cfg
         # 1. actors generate a chunk of experience
         if self.actor is not None:
-            self.agent.reset_noise()
-            self.actor.run_chunk()
+            with self._sp("actor"):
+                self.agent.reset_noise()
+                self.actor.run_chunk()
             chunk_f, chunk_m = self.actor.chunk_frames, self.actor.chunk_meta
         else:
             c = self.n_envs * self.chunk_steps
@@ -497,31 +515,35 @@ class ApexRunner:
             chunk_m = torch.zeros(c, 3, device=self.comm.device)
 
         # 2. transition flow: actors -> learner replay (reference: redis push)
-        frames_list = self.comm.gather_chunks(chunk_f)
-        meta_list = self.comm.gather_chunks(chunk_m)
+        with self._sp("gather"):
+            frames_list = self.comm.gather_chunks(chunk_f)
+            meta_list = self.comm.gather_chunks(chunk_m)
         if self.comm.is_learner:
             if self.comm.world_size == 1:
                 frames_list, meta_list = [chunk_f], [chunk_m]
-            for a, (f, m) in enumerate(zip(frames_list, meta_list)):
-                mc = m.cpu().numpy()
-                self.replay.append_actor_chunk(
-                    a, f.cpu().numpy(), mc[:, 0].astype(np.int64), mc[:, 1],
-                    mc[:, 2] > 0.5,
-                )
+            with self._sp("append"):
+                for a, (f, m) in enumerate(zip(frames_list, meta_list)):
+                    mc = m.cpu().numpy()
+                    self.replay.append_actor_chunk(
+                        a, f.cpu().numpy(), mc[:, 0].astype(np.int64),
+                        mc[:, 1], mc[:, 2] > 0.5,
+                    )
 
         # 3. learner trains
         if self.comm.is_learner and learn and len(self.replay) >= cfg.learn_start:
-            for _ in range(cfg.learn_per_round):
-                loss = self.agent.learn(self.replay)
-                self.losses.append(loss)
-                self.grad_steps += 1
-                if self.grad_steps % max(1, cfg.target_update // max(1, cfg.replay_frequency)) == 0:
-                    self.agent.update_target_net()
+            with self._sp("learn"):
+                for _ in range(cfg.learn_per_round):
+                    loss = self.agent.learn(self.replay)
+                    self.losses.append(loss)
+                    self.grad_steps += 1
+                    if self.grad_steps % max(1, cfg.target_update // max(1, cfg.replay_frequency)) == 0:
+                        self.agent.update_target_net()
 
         # 4. weight flow: learner -> actors (reference: redis weight blob)
         if self.comm.world_size > 1:
-            if self.comm.is_learner:
-                self.flat_weights = self.agent.flat_weights()
-            self.comm.broadcast_weights(self.flat_weights, src=0)
-            if not self.comm.is_learner:
-                self.agent.load_flat_weights(self.flat_weights)
+            with self._sp("broadcast"):
+                if self.comm.is_learner:
+                    self.flat_weights = self.agent.flat_weights()
+                self.comm.broadcast_weights(self.flat_weights, src=0)
+                if not self.comm.is_learner:
+                    self.agent.load_flat_weights(self.flat_weights)

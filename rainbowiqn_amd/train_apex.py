@@ -101,7 +101,8 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
                n_envs: int = 16, results_dir: Optional[str] = None,
                max_rounds: Optional[int] = None, log_interval: int = 50,
                metrics_port: Optional[int] = None,
-               chunk_steps: Optional[int] = None):
+               chunk_steps: Optional[int] = None,
+               trace_file: Optional[str] = None):
     """Run Ape-X training to cfg.T_max agent steps.  Returns the runner."""
     device = cfg.resolve_device()
     if device == "cuda":
@@ -111,6 +112,15 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
 
     runner = ApexRunner(cfg, comm, n_envs=n_envs, device=device,
                         chunk_steps=chunk_steps)
+    tracer = None
+    if trace_file:
+        from .utils.trace import RoundTracer
+
+        path = (trace_file if comm.world_size == 1
+                else f"{trace_file}.rank{comm.rank}")
+        tracer = RoundTracer(path, use_cuda=(device == "cuda"),
+                             pid=comm.rank)
+        runner.set_tracer(tracer)
     if cfg.model and comm.is_learner:
         _load_model(runner, cfg.model)
     if cfg.memory_save_path:
@@ -133,6 +143,13 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
         evaluator = Evaluator(cfg, device="cpu")
         proxy = _EvalProxy(cfg, runner, device)
 
+    # failure handling: SIGTERM/SIGINT -> checkpoint at the
+    # round boundary; non-finite loss -> roll back to last healthy snapshot
+    from .utils.failsafe import GracefulShutdown, NanGuard
+
+    shutdown = GracefulShutdown()
+    guard = NanGuard(runner) if comm.is_learner else None
+
     steps_per_round = runner.chunk_steps * runner.n_envs * comm.n_actors
     T, rounds = 0, 0
     next_eval = cfg.evaluation_interval
@@ -146,7 +163,15 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
         runner.run_round()
         rounds += 1
         T += steps_per_round
+        if shutdown.requested:
+            if comm.is_learner and results_dir:
+                _save_model(runner, results_dir)
+                if cfg.memory_save_path and not runner.sharded:
+                    _save_memory(runner, cfg.memory_save_path)
+            break
         if comm.is_learner:
+            if guard.check(runner.losses[-1] if runner.losses else None):
+                metrics.update(recoveries=guard.recoveries)
             _anneal_beta(runner, T, cfg)
             if evaluator is not None and T >= next_eval:
                 next_eval += cfg.evaluation_interval
@@ -181,6 +206,9 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
     if runner.sharded and comm.is_actor and cfg.memory_save_path:
         torch.save(runner.shard.replay.state_dict(),
                    f"{cfg.memory_save_path}.rank{comm.rank}")
+    if tracer is not None:
+        log(f"round-phase trace written to {tracer.dump()}")
+    shutdown.restore()
     return runner
 
 
@@ -208,6 +236,8 @@ def main(argv=None):
                    help="Prometheus exporter port (rank offset added)")
     p.add_argument("--chunk-steps", type=int, default=8,
                    help="env steps per Ape-X exchange round")
+    p.add_argument("--trace-file", type=str, default=None,
+                   help="write a Chrome-trace round-phase timeline here")
     ns, rest = p.parse_known_args(argv)
     cfg = parse_args(rest)
     comm = Communicator(backend=cfg.backend)
@@ -215,7 +245,8 @@ def main(argv=None):
     runner = train_apex(cfg, comm, n_envs=ns.n_envs, results_dir=results_dir,
                         max_rounds=ns.max_rounds,
                         metrics_port=ns.metrics_port,
-                        chunk_steps=ns.chunk_steps)
+                        chunk_steps=ns.chunk_steps,
+                        trace_file=ns.trace_file)
     if comm.is_learner:
         log(f"done: grad_steps={runner.grad_steps}")
     comm.close()

@@ -28,14 +28,19 @@ class Metrics:
                                  "transitions resident in replay")
         self.steps_per_sec = Gauge("riqn_env_steps_per_sec",
                                    "recent env steps/sec (this rank)")
+        self.recoveries = Gauge("riqn_nan_recoveries_total",
+                                "divergence rollbacks (utils/failsafe.py)")
         self.enabled = True
 
     def update(self, d_env_steps: int = 0, d_grad_steps: int = 0,
                loss: Optional[float] = None,
                replay_size: Optional[int] = None,
-               steps_per_sec: Optional[float] = None) -> None:
+               steps_per_sec: Optional[float] = None,
+               recoveries: Optional[int] = None) -> None:
         if not self.enabled:
             return
+        if recoveries is not None:
+            self.recoveries.set(recoveries)
         if d_env_steps:
             self.env_steps.inc(d_env_steps)
         if d_grad_steps:
