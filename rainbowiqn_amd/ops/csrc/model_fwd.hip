@@ -319,6 +319,10 @@ __global__ void pack_fwd_frag_kernel(
 // WG owns leaves as coalesced 8-byte row stores instead of strided 2-byte
 // scatters; the <=4 feat rows the tile touches are staged the same way.
 // Optionally also writes phi (pre-multiply embedding) for the backward pass.
+// DIRECT=1 skips the LDS output staging and stores straight from the MFMA
+// accumulators (2-byte lanes, 32 B segments per quarter-wave): fewer LDS
+// ops + one barrier less, and LDS drops 49 KB -> occupancy is VGPR-bound.
+template <int DIRECT>
 __global__ __launch_bounds__(256) void tau_embed_kernel(
     const float* __restrict__ taus,         // [R]
     const unsigned short* __restrict__ w,   // [3136,64] bf16 (iqn_fc.weight)
@@ -328,7 +332,7 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
     unsigned short* __restrict__ phi_out,   // [R,3136] bf16 or nullptr
     unsigned short* __restrict__ cosb_out,  // [R,64] bf16 or nullptr (bwd)
     int R, int nq) {
-    __shared__ unsigned short tile[2 * 16 * 784];  // two m-tiles per WG
+    __shared__ unsigned short tile[DIRECT ? 2 : 2 * 16 * 784];
     __shared__ unsigned short ftile[4 * 784];
     const int t = threadIdx.x;
     const int lane = t & 63;
@@ -411,22 +415,30 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
                 const int r = (mt0 + mi) * 16 + rr;
                 unsigned short ph = f32_to_bf16(fmaxf(acc[mi][j] + bb, 0.0f));
                 float fv = bf(ftile[min(r / nq - s0, span - 1) * 784 + col]);
-                tile[(mi * 16 + rr) * 784 + col] =
-                    f32_to_bf16(bf16_to_f32(ph) * fv);
+                if (DIRECT) {
+                    if (r < R)
+                        h[(int64_t)r * 3136 + nt * 16 + l15] =
+                            f32_to_bf16(bf16_to_f32(ph) * fv);
+                } else {
+                    tile[(mi * 16 + rr) * 784 + col] =
+                        f32_to_bf16(bf16_to_f32(ph) * fv);
+                }
                 if (phi_out != nullptr && r < R)
                     phi_out[(int64_t)r * 3136 + nt * 16 + l15] = ph;
             }
         }
     }
-    __syncthreads();
-    // coalesced row stores: 32 rows x 784 cols
-    #pragma unroll 8
-    for (int i = t * 4; i < 32 * 784; i += 256 * 4) {
-        int rr = i / 784, cc = i % 784;
-        int r = mt0 * 16 + rr;
-        if (r < R)
-            *(u16x4*)(h + (int64_t)r * 3136 + nt0 * 16 + cc) =
-                *(const u16x4*)&tile[rr * 784 + cc];
+    if (!DIRECT) {
+        __syncthreads();
+        // coalesced row stores: 32 rows x 784 cols
+        #pragma unroll 8
+        for (int i = t * 4; i < 32 * 784; i += 256 * 4) {
+            int rr = i / 784, cc = i % 784;
+            int r = mt0 * 16 + rr;
+            if (r < R)
+                *(u16x4*)(h + (int64_t)r * 3136 + nt0 * 16 + cc) =
+                    *(const u16x4*)&tile[rr * 784 + cc];
+        }
     }
 }
 
@@ -778,9 +790,10 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
 
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     const void* feat, void* h, void* phi_out, void* cosb_out,
-                    int R, int nq, void* stream) {
+                    int R, int nq, int direct, void* stream) {
     dim3 grid((R + 31) / 32, 4);
-    hipLaunchKernelGGL(tau_embed_kernel, grid, dim3(256), 0,
+    auto kern = direct ? tau_embed_kernel<1> : tau_embed_kernel<0>;
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0,
                        (hipStream_t)stream, taus, (const unsigned short*)w,
                        (const unsigned short*)bias,
                        (const unsigned short*)feat, (unsigned short*)h,

@@ -93,7 +93,7 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       void* stream);
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     const void* feat, void* h, void* phi_out, void* cosb_out,
-                    int R, int nq, void* stream);
+                    int R, int nq, int direct, void* stream);
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
                      void* out, int R, void* stream);
 void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,

@@ -21,12 +21,17 @@ All methods are hipGraph-capturable: fixed shapes, static scratch buffers
 
 from __future__ import annotations
 
+import os
 from typing import Dict, Tuple
 
 import torch
 
 from ..models.model import RainbowIQN
 from ..ops import get_extension
+
+# tau_embed variant: 1 = direct accumulator stores (no LDS output staging).
+# Opt-in until the kernel microbench decides a default (scripts/kbench.py).
+TAU_DIRECT = os.environ.get("RIQN_TAU_V2", "0") == "1"
 
 
 class FusedForward:
@@ -138,7 +143,8 @@ class FusedForward:
         feat, h, hh = bufs[:3]
         self.ext.conv_encode(x, self.w1, self.b1, self.w2, self.b2, self.w3p,
                              self.b3, feat)
-        self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h, nq)
+        self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h, nq,
+                           direct=TAU_DIRECT)
         if self.fp8_heads:
             hq, hsc = bufs[3], bufs[4]
             self.ext.quant_fp8(h, hq, hsc)
@@ -170,7 +176,7 @@ class FusedForward:
         B = featv.shape[0]
         _, h, hh = self._bufs(B, nq)[:3]
         self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn,
-                           featv.contiguous(), h, nq)
+                           featv.contiguous(), h, nq, direct=TAU_DIRECT)
         self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
         dummy = torch.zeros(1, dtype=torch.float32, device=self.device)
         self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 1, dummy,

@@ -182,8 +182,10 @@ class GpuLearner:
         ext.conv_encode(self.replay.states_cat, fo.w1, fo.b1, fo.w2, fo.b2,
                         fo.w3p, fo.b3, self.feat_cat, self.c1g, self.c2g)
         feat_g = self.feat_cat[:b]
+        from .fused_forward import TAU_DIRECT
         ext.tau_embed(self.taus_online.view(-1), fo.wiqn, fo.biqn, feat_g,
-                      h_g, N, phi_out=self.phi, cosb_out=self.cosb)
+                      h_g, N, phi_out=self.phi, cosb_out=self.cosb,
+                      direct=TAU_DIRECT)
         ext.heads_gemm(h_g, fo.wcat, fo.bcat, hh_g)
         ext.heads_finish(hh_g, fo.wz, fo.bz, A, N, 2,
                          self.theta_buf.view(-1), a)

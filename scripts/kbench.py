@@ -53,6 +53,9 @@ def main():
         us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq),
                     iters)
         print(f"{'tau_embed':<28} {f'R={R}':<18} {us:9.2f}")
+        us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq,
+                                          direct=True), iters)
+        print(f"{'tau_embed direct':<28} {f'R={R}':<18} {us:9.2f}")
         us = timeit(lambda: ext.heads_gemm(h, fw.wcat, fw.bcat, hh), iters)
         gf = R * 1024 * 3136 * 2 / 1e9
         print(f"{'heads_gemm':<28} {f'R={R}':<18} {us:9.2f}  "

@@ -60,6 +60,34 @@ def test_conv_encode_parity():
     assert err < max(3 * err16, 0.02 * max(scale, 1.0)), (err, err16)
 
 
+def test_tau_embed_direct_matches_staged():
+    """tau_embed DIRECT=1 (accumulator stores, no LDS staging) is bit-exact
+    vs the staged variant on h, phi and cosb -- same math, different path."""
+    from rainbowiqn_amd.ops import get_extension
+
+    ext = get_extension()
+    _, _, net16 = make_nets()
+    fw = fused_of(net16)
+    for B, nq in [(32, 8), (1024, 8), (255, 32)]:  # 255: ragged tail rows
+        R = B * nq
+        x = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
+        feat = fw.encode(x).clone()
+        taus = torch.rand(R, device="cuda")
+        h1 = torch.empty(R, 3136, dtype=torch.bfloat16, device="cuda")
+        h2 = torch.empty_like(h1)
+        p1 = torch.empty_like(h1)
+        p2 = torch.empty_like(h1)
+        c1 = torch.empty(R, 64, dtype=torch.bfloat16, device="cuda")
+        c2 = torch.empty_like(c1)
+        ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h1, nq,
+                      phi_out=p1, cosb_out=c1)
+        ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h2, nq,
+                      phi_out=p2, cosb_out=c2, direct=True)
+        torch.cuda.synchronize()
+        assert torch.equal(h1, h2), (B, nq)
+        assert torch.equal(p1, p2) and torch.equal(c1, c2), (B, nq)
+
+
 def test_full_forward_parity():
     A = 6
     _, net32, net16 = make_nets(A)
