@@ -345,12 +345,13 @@ class ApexRunner:
         if comm.is_actor:
             if self.fast:
                 loop = self.gactors[0]
-                if self._actor_flats[0] is not None and \
-                        self._actor_flats[0] is not self.flat_weights:
-                    self._actor_flats[0].copy_(self.flat_weights)
-                loop.run_round()
-                self.shard.append_chunk(loop.chunk_frames,
-                                        meta=loop.chunk_meta)
+                with self._sp("actor"):
+                    if self._actor_flats[0] is not None and \
+                            self._actor_flats[0] is not self.flat_weights:
+                        self._actor_flats[0].copy_(self.flat_weights)
+                    loop.run_round()
+                    self.shard.append_chunk(loop.chunk_frames,
+                                            meta=loop.chunk_meta)
             else:
                 self.agent.reset_noise()
                 self.actor.run_chunk()
@@ -375,13 +376,16 @@ class ApexRunner:
         # 3. batches -> learner; grad steps; priorities -> actors
         if learning:
             if comm.is_actor:
-                self.shard.sample_and_send()
-                self.shard.recv_priorities()
+                with self._sp("batch-service"):
+                    self.shard.sample_and_send()
+                    self.shard.recv_priorities()
             else:
                 self.replay.set_len(total)
-                self.replay.recv_batches()
+                with self._sp("gather"):
+                    self.replay.recv_batches()
                 if self.fast:
-                    self.glearner.learn_steps(cfg.learn_per_round)
+                    with self._sp("learn"):
+                        self.glearner.learn_steps(cfg.learn_per_round)
                     self.grad_steps = self.glearner.grad_steps
                     self.losses.append(self.glearner.last_loss())
                     tgt = max(1, cfg.target_update
@@ -482,16 +486,19 @@ class ApexRunner:
         # learner stream: append LAST round's chunk, then grad steps; these
         # small-grid kernels fill the CUs the actor's big grids leave idle
         if self._round_idx > 0:
-            gl.append_actor_chunk(0, self._staging_f[prev],
-                                  meta=self._staging_m[prev])
+            with self._sp("append"):
+                gl.append_actor_chunk(0, self._staging_f[prev],
+                                      meta=self._staging_m[prev])
         # actor stream: round + chunk stage-out, concurrent with the learner
         self._astream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(self._astream):
-            loop.run_round()
-            self._staging_f[j].copy_(loop.chunk_frames, non_blocking=True)
-            self._staging_m[j].copy_(loop.chunk_meta, non_blocking=True)
+            with self._sp("actor"):
+                loop.run_round()
+                self._staging_f[j].copy_(loop.chunk_frames, non_blocking=True)
+                self._staging_m[j].copy_(loop.chunk_meta, non_blocking=True)
         if learn and len(gl.replay) >= cfg.learn_start:
-            gl.learn_steps(cfg.learn_per_round)
+            with self._sp("learn"):
+                gl.learn_steps(cfg.learn_per_round)
             self.grad_steps = gl.grad_steps
             self.losses.append(gl.last_loss())
             tgt = max(1, cfg.target_update // max(1, cfg.replay_frequency))
