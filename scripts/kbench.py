@@ -4,7 +4,11 @@ uses (act: B=E envs x K taus; learn: B=32 x {K, Np}).
 
 Usage (GPU box): python scripts/kbench.py [iters]
 """
+import os
 import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import time
 
 import torch

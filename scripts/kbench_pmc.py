@@ -1,5 +1,10 @@
 #!/usr/bin/env python3
 """Minimal kernel loop for PMC counter collection (rocprofv3 --pmc)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 from rainbowiqn_amd.config import Config
 from rainbowiqn_amd.models.model import RainbowIQN

@@ -229,6 +229,43 @@ def test_replay_tree_invariants_after_traffic():
 
 
 @pytest.mark.gpu
+def test_nan_guard_fast_rollback():
+    """Failure recovery on the fast path: corrupt the fp32 master + feed a
+    non-finite loss -> NanGuard restores master, bf16 weights and Adam
+    moments, and the runner keeps training."""
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+    from rainbowiqn_amd.utils.failsafe import NanGuard
+
+    cfg = Config(device="cuda", dtype="bf16", learn_start=256,
+                 memory_capacity=8192, learn_per_round=2)
+    runner = ApexRunner(cfg, Communicator(backend="auto"), n_envs=16,
+                        chunk_steps=8)
+    for _ in range(4):
+        runner.run_round()
+    torch.cuda.synchronize()
+    guard = NanGuard(runner, interval=1)
+    assert not guard.check(runner.losses[-1])  # snapshot
+    gl = runner.glearner
+    healthy = gl.opt.master.detach().clone()
+    with torch.no_grad():
+        gl.opt.master.add_(float("inf"))
+        gl.opt.m.fill_(float("nan"))
+    assert guard.check(float("nan"))
+    torch.cuda.synchronize()
+    assert torch.equal(gl.opt.master, healthy)
+    assert torch.isfinite(gl.opt.m).all()
+    assert torch.equal(gl.flat_online.float(),
+                       healthy.to(torch.bfloat16).float())
+    # training continues with finite losses
+    for _ in range(2):
+        runner.run_round()
+    torch.cuda.synchronize()
+    assert np.isfinite(runner.losses[-1])
+
+
+@pytest.mark.gpu
 def test_actor_shard_host_gpu():
     """Config-5 actor-side shard on GPU: HBM-resident PER shard samples its
     learner-batch quota with local frame-stack/n-step assembly."""
