@@ -304,9 +304,18 @@ class ApexRunner:
                         and self.glearner is not None
                         and getattr(cfg, "overlap_colocated", True))
         if self.overlap:
+            import os as _os
+
             d = self.gactors[0].device
             c = self.n_envs * self.chunk_steps
             self._astream = torch.cuda.Stream(device=d)
+            # RIQN_LPRIO=1: learner work on a high-priority HSA queue so its
+            # small grids (batch 32) dispatch ahead of the actor's queued
+            # chip-filling WGs instead of waiting behind them (opt-in until
+            # the bench decides a default).
+            self._lstream = (torch.cuda.Stream(device=d, priority=-1)
+                             if _os.environ.get("RIQN_LPRIO", "0") == "1"
+                             else None)
             self._staging_f = [torch.zeros(c, 84, 84, dtype=torch.uint8,
                                            device=d) for _ in range(2)]
             self._staging_m = [torch.zeros(c, 3, device=d) for _ in range(2)]
@@ -483,29 +492,43 @@ class ApexRunner:
         loop = self.gactors[0]
         j = self._round_idx % 2
         prev = (self._round_idx - 1) % 2
+        outer = torch.cuda.current_stream()
+        # learner work optionally rides a high-priority queue (RIQN_LPRIO):
+        # its small grids then dispatch ahead of the actor's queued WGs
+        from contextlib import nullcontext
+
+        lctx = (lambda: torch.cuda.stream(self._lstream)) \
+            if self._lstream is not None else nullcontext
+        if self._lstream is not None:
+            self._lstream.wait_stream(outer)
         # learner stream: append LAST round's chunk, then grad steps; these
         # small-grid kernels fill the CUs the actor's big grids leave idle
-        if self._round_idx > 0:
-            with self._sp("append"):
-                gl.append_actor_chunk(0, self._staging_f[prev],
-                                      meta=self._staging_m[prev])
+        with lctx():
+            if self._round_idx > 0:
+                with self._sp("append"):
+                    gl.append_actor_chunk(0, self._staging_f[prev],
+                                          meta=self._staging_m[prev])
         # actor stream: round + chunk stage-out, concurrent with the learner
-        self._astream.wait_stream(torch.cuda.current_stream())
+        self._astream.wait_stream(outer)
         with torch.cuda.stream(self._astream):
             with self._sp("actor"):
                 loop.run_round()
                 self._staging_f[j].copy_(loop.chunk_frames, non_blocking=True)
                 self._staging_m[j].copy_(loop.chunk_meta, non_blocking=True)
-        if learn and len(gl.replay) >= cfg.learn_start:
-            with self._sp("learn"):
-                gl.learn_steps(cfg.learn_per_round)
-            self.grad_steps = gl.grad_steps
-            self.losses.append(gl.last_loss())
-            tgt = max(1, cfg.target_update // max(1, cfg.replay_frequency))
-            if self.grad_steps % tgt < cfg.learn_per_round:
-                gl.update_target()
-                gl.maintenance()
-        torch.cuda.current_stream().wait_stream(self._astream)
+        with lctx():
+            if learn and len(gl.replay) >= cfg.learn_start:
+                with self._sp("learn"):
+                    gl.learn_steps(cfg.learn_per_round)
+                self.grad_steps = gl.grad_steps
+                self.losses.append(gl.last_loss())
+                tgt = max(1, cfg.target_update
+                          // max(1, cfg.replay_frequency))
+                if self.grad_steps % tgt < cfg.learn_per_round:
+                    gl.update_target()
+                    gl.maintenance()
+        if self._lstream is not None:
+            outer.wait_stream(self._lstream)
+        outer.wait_stream(self._astream)
         self._round_idx += 1
 
     def _run_round_ref(self, learn: bool) -> None:
