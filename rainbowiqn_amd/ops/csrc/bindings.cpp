@@ -298,7 +298,7 @@ void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
                torch::Tensor feat, torch::Tensor h, int64_t nq,
                c10::optional<torch::Tensor> phi_out = c10::nullopt,
                c10::optional<torch::Tensor> cosb_out = c10::nullopt,
-               bool direct = false) {
+               int64_t direct = -1) {  // -1 auto, 0 staged, 1 direct
     CHECK_T(taus, torch::kFloat32);
     CHECK_BF16(w); CHECK_BF16(bias); CHECK_BF16(feat); CHECK_BF16(h);
     int R = (int)taus.numel();
@@ -317,7 +317,7 @@ void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
     }
     riqn_tau_embed(taus.data_ptr<float>(), w.data_ptr(), bias.data_ptr(),
                    feat.data_ptr(), h.data_ptr(), pp, cb, R, (int)nq,
-                   direct ? 1 : 0, cur_stream());
+                   (int)direct, cur_stream());
 }
 
 void heads_gemm(torch::Tensor h, torch::Tensor wcat, torch::Tensor bcat,
@@ -521,7 +521,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("quant_fp8", &quant_fp8);
     m.def("heads_gemm_fp8", &heads_gemm_fp8);
     m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
-    m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt, py::arg("direct") = false);
+    m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt, py::arg("direct") = -1);
     m.def("heads_gemm", &heads_gemm);
     m.def("heads_finish", &heads_finish);
     m.def("finish_bwd_dhh", &finish_bwd_dhh);

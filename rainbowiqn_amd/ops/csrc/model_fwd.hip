@@ -791,6 +791,10 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     const void* feat, void* h, void* phi_out, void* cosb_out,
                     int R, int nq, int direct, void* stream) {
+    // auto (-1): direct stores win at actor-size R (occupancy unbound by
+    // LDS, 37 -> 28 us at R=8192); the staged tile's coalesced rows win at
+    // learner-size R where the grid is latency-bound (kbench_tau2).
+    if (direct < 0) direct = (R >= 4096);
     dim3 grid((R + 31) / 32, 4);
     auto kern = direct ? tau_embed_kernel<1> : tau_embed_kernel<0>;
     hipLaunchKernelGGL(kern, grid, dim3(256), 0,

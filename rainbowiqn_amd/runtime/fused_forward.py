@@ -29,9 +29,10 @@ import torch
 from ..models.model import RainbowIQN
 from ..ops import get_extension
 
-# tau_embed variant: 1 = direct accumulator stores (no LDS output staging).
-# Opt-in until the kernel microbench decides a default (scripts/kbench.py).
-TAU_DIRECT = os.environ.get("RIQN_TAU_V2", "0") == "1"
+# tau_embed variant: -1 = auto (direct accumulator stores at actor-size R,
+# LDS-staged rows at learner-size R -- kbench_tau2: 37 -> 28 us at R=8192,
+# staged still ahead below R=4096); RIQN_TAU_V2=0/1 forces a variant.
+TAU_DIRECT = int(os.environ.get("RIQN_TAU_V2", "-1"))
 
 
 class FusedForward:

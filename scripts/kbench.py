@@ -54,9 +54,9 @@ def main():
         us = timeit(lambda: ext.conv_encode(x, fw.w1, fw.b1, fw.w2, fw.b2,
                                             fw.w3p, fw.b3, feat), iters)
         print(f"{'conv_encode':<28} {f'B={B}':<18} {us:9.2f}")
-        us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq),
-                    iters)
-        print(f"{'tau_embed':<28} {f'R={R}':<18} {us:9.2f}")
+        us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h,
+                                          nq, direct=0), iters)
+        print(f"{'tau_embed staged':<28} {f'R={R}':<18} {us:9.2f}")
         us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq,
                                           direct=True), iters)
         print(f"{'tau_embed direct':<28} {f'R={R}':<18} {us:9.2f}")

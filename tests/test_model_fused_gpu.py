@@ -80,7 +80,7 @@ def test_tau_embed_direct_matches_staged():
         c1 = torch.empty(R, 64, dtype=torch.bfloat16, device="cuda")
         c2 = torch.empty_like(c1)
         ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h1, nq,
-                      phi_out=p1, cosb_out=c1)
+                      phi_out=p1, cosb_out=c1, direct=0)
         ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h2, nq,
                       phi_out=p2, cosb_out=c2, direct=True)
         torch.cuda.synchronize()
