@@ -42,21 +42,22 @@ DEV bf16x8 as_bf16x8(u16x8 v) { return __builtin_bit_cast(bf16x8, v); }
 DEV float bf(const unsigned short u) { return bf16_to_f32(u); }
 
 // ---------------------------------------------------------------------------
-// 1. Fused conv encoder.  One WG (256 threads, 4 waves) per state.
+// 1. Fused conv encoder.  One WG (512 threads, 8 waves) per state.
 // LDS layout (ushorts):
-//   in   : [4][84][88]   -> 29568 ush (59136 B)
 //   c1out: [32][20][24]  -> 15360 ush (30720 B)
 //   c2out: [64][9][12]   ->  6912 ush (13824 B)
+// conv1 reads its im2col A-operand straight from global memory: the input
+// state is 56 KB (L2-resident; quarter-wave-coalesced 16 B segments), and
+// NOT staging it drops LDS 103 -> 44.5 KB, doubling occupancy to 2 WGs/CU
+// -- the B=1024 actor launch goes from 4 serial occupancy rounds to 2.
 // ---------------------------------------------------------------------------
-#define IN_STRIDE 88
 #define C1_STRIDE 24
 #define C2_STRIDE 12
-#define LDS_IN 0
-#define LDS_C1 29568
-#define LDS_C2 (29568 + 15360)
-#define LDS_TOTAL (29568 + 15360 + 6912)
+#define LDS_C1 0
+#define LDS_C2 15360
+#define LDS_TOTAL (15360 + 6912)
 
-__global__ __launch_bounds__(512, 1) void conv_encode_kernel(
+__global__ __launch_bounds__(512) void conv_encode_kernel(
     const unsigned short* __restrict__ x,    // [B,4,84,84] bf16
     const unsigned short* __restrict__ w1,   // [32,256]  bf16
     const unsigned short* __restrict__ b1,   // [32]      bf16
@@ -75,33 +76,11 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
     const int l15 = lane & 15;
     const int l4 = lane >> 4;  // 0..3
 
-    // ---- stage input state into LDS (row stride 88) ----------------------
-    {
-        // two-phase: issue ALL loads, then all LDS stores -- a fused
-        // load->store loop serializes at one HBM latency per iteration
-        const unsigned short* src = x + (int64_t)b * (4 * 84 * 84);
-        u16x4 tmp[14];
-        #pragma unroll
-        for (int i = 0; i < 14; ++i) {
-            int e = t * 4 + i * 2048;
-            tmp[i] = (e < 4 * 84 * 84) ? *(const u16x4*)(src + e)
-                                       : u16x4{0, 0, 0, 0};
-        }
-        #pragma unroll
-        for (int i = 0; i < 14; ++i) {
-            int e = t * 4 + i * 2048;
-            if (e < 4 * 84 * 84) {
-                int c = e / 7056, r = (e % 7056) / 84, cc = e % 84;
-                *(u16x4*)&lds[LDS_IN + (c * 84 + r) * IN_STRIDE + cc] = tmp[i];
-            }
-        }
-    }
-    __syncthreads();
+    const unsigned short* src = x + (int64_t)b * (4 * 84 * 84);
 
     // ---- conv1: M=400 (25 tiles), N=32 (2 tiles), K=256 (8 steps) --------
-    // All 16 B-fragments live in registers, loaded before the input barrier
-    // so their latency hides under the staging; the m-loop then runs on LDS
-    // reads + MFMA only.
+    // All 16 B-fragments live in registers; the m-loop runs on L2-cached
+    // global A reads + MFMA.
     bf16x8 b1f[2][8];
     #pragma unroll
     for (int nt = 0; nt < 2; ++nt)
@@ -121,9 +100,9 @@ __global__ __launch_bounds__(512, 1) void conv_encode_kernel(
             // A: 8 taps = input row segment [c][oy*4+ky][ox*4 .. +7]
             u16x8 av;
             {
-                const int idx = LDS_IN + (c * 84 + oy * 4 + ky) * IN_STRIDE + ox * 4;
-                u16x4 lo = *(const u16x4*)&lds[idx];
-                u16x4 hi = *(const u16x4*)&lds[idx + 4];
+                const int idx = (c * 84 + oy * 4 + ky) * 84 + ox * 4;
+                u16x4 lo = *(const u16x4*)(src + idx);
+                u16x4 hi = *(const u16x4*)(src + idx + 4);
                 av = u16x8{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
             }
             bf16x8 a = as_bf16x8(av);
