@@ -78,7 +78,17 @@ __global__ __launch_bounds__(256) void heads_gemm_fp8_kernel(
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;
     const int l15 = lane & 15, l4 = lane >> 4;
-    const int bm = blockIdx.x * 128, bn = blockIdx.y * 128;
+    // XCD-aware tile map (see heads_gemm_big_kernel): each XCD takes a
+    // contiguous row-block band x all n-blocks so co-resident WGs share
+    // operands through their XCD's L2.
+    int bxi = blockIdx.x, byi = blockIdx.y;
+    if ((gridDim.x & 7) == 0) {
+        const int id = blockIdx.x + blockIdx.y * gridDim.x;
+        const int xcd = id & 7, s = id >> 3;
+        byi = s % gridDim.y;
+        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
+    }
+    const int bm = bxi * 128, bn = byi * 128;
 
     // 16 glds insts per 128x128-byte operand; wave w issues [4w, 4w+4).
     // Source XOR swizzle on the 16-byte piece index (8 pieces per row).

@@ -553,7 +553,20 @@ __global__ __launch_bounds__(256) void heads_gemm_big_kernel(
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;
     const int l15 = lane & 15, l4 = lane >> 4;
-    const int bm = blockIdx.x * 128, bn = blockIdx.y * 128;
+    // XCD-aware tile map: the dispatcher assigns WGs to XCDs round-robin on
+    // the linear id, so id%8 IS the XCD.  Give each XCD a contiguous band of
+    // row-blocks x all n-blocks: the 64 co-resident WGs of one XCD then
+    // share their A rows and B cols through that XCD's L2 (per k-window the
+    // joint working set is ~256 KB) instead of re-pulling A from HBM once
+    // per n-block (MI355X_MICROARCH.md XCD/L2 locality rule).
+    int bxi = blockIdx.x, byi = blockIdx.y;
+    if ((gridDim.x & 7) == 0) {
+        const int id = blockIdx.x + blockIdx.y * gridDim.x;
+        const int xcd = id & 7, s = id >> 3;
+        byi = s % gridDim.y;
+        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
+    }
+    const int bm = bxi * 128, bn = byi * 128;
 
     // 16 glds insts cover a 128x64 operand; wave w issues insts [4w, 4w+4)
     // for A and for B (8 per wave per buffer).
