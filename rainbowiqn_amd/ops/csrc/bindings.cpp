@@ -294,6 +294,28 @@ void conv_encode(torch::Tensor x, torch::Tensor w1, torch::Tensor b1,
                      feat.data_ptr(), p1, p2, B, cur_stream());
 }
 
+void conv_encode_gather(torch::Tensor ring, torch::Tensor env_state,
+                        int64_t history, int64_t ring_head, torch::Tensor w1,
+                        torch::Tensor b1, torch::Tensor w2, torch::Tensor b2,
+                        torch::Tensor w3p, torch::Tensor b3,
+                        torch::Tensor feat) {
+    CHECK_T(ring, torch::kUInt8);
+    CHECK_T(env_state, torch::kInt32);
+    CHECK_BF16(w1); CHECK_BF16(w2); CHECK_BF16(w3p);
+    CHECK_BF16(b1); CHECK_BF16(b2); CHECK_BF16(b3);
+    CHECK_BF16(feat);
+    TORCH_CHECK(history == 4, "fused frame-stack gather assumes history 4");
+    int64_t E = env_state.numel() / 4;
+    TORCH_CHECK(ring.numel() == history * E * 7056, "ring shape");
+    TORCH_CHECK(feat.numel() == E * 3136);
+    TORCH_CHECK(ring_head >= 0 && ring_head < history);
+    riqn_conv_encode_gather(ring.data_ptr(), env_state.data_ptr(),
+                            (int)ring_head, w1.data_ptr(), b1.data_ptr(),
+                            w2.data_ptr(), b2.data_ptr(), w3p.data_ptr(),
+                            b3.data_ptr(), feat.data_ptr(), (int)E,
+                            cur_stream());
+}
+
 void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
                torch::Tensor feat, torch::Tensor h, int64_t nq,
                c10::optional<torch::Tensor> phi_out = c10::nullopt,
@@ -521,6 +543,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("quant_fp8", &quant_fp8);
     m.def("heads_gemm_fp8", &heads_gemm_fp8);
     m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
+    m.def("conv_encode_gather", &conv_encode_gather, py::arg("ring"), py::arg("env_state"), py::arg("history"), py::arg("ring_head"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"));
     m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt, py::arg("direct") = -1);
     m.def("heads_gemm", &heads_gemm);
     m.def("heads_finish", &heads_finish);

@@ -57,8 +57,17 @@ DEV float bf(const unsigned short u) { return bf16_to_f32(u); }
 #define LDS_C2 15360
 #define LDS_TOTAL (15360 + 6912)
 
+// GATHER=1 additionally fuses the actor's frame-stack assembly (K15,
+// env.hip stack_gather): conv1 gathers its taps straight from the u8 frame
+// RING (per-channel slot pointers, pre-episode frames zeroed via a 0 scale)
+// -- the [E,4,84,84] bf16 stack tensor never exists, saving its ~58 MB
+// write + re-read per act at E=1024.
+template <int GATHER>
 __global__ __launch_bounds__(512) void conv_encode_kernel(
-    const unsigned short* __restrict__ x,    // [B,4,84,84] bf16
+    const unsigned short* __restrict__ x,    // [B,4,84,84] bf16 (GATHER=0)
+    const uint8_t* __restrict__ ring,        // [h][E][7056] u8 (GATHER=1)
+    const int32_t* __restrict__ env_state,   // [E,4] (GATHER=1)
+    int ring_head,                           // newest-frame slot (GATHER=1)
     const unsigned short* __restrict__ w1,   // [32,256]  bf16
     const unsigned short* __restrict__ b1,   // [32]      bf16
     const unsigned short* __restrict__ w2,   // [64,512]  bf16
@@ -77,6 +86,20 @@ __global__ __launch_bounds__(512) void conv_encode_kernel(
     const int l4 = lane >> 4;  // 0..3
 
     const unsigned short* src = x + (int64_t)b * (4 * 84 * 84);
+    // per-channel ring frame base + scale (WG-uniform; history fixed at 4)
+    const uint8_t* rb[4];
+    float rs[4];
+    if (GATHER) {
+        const int steps = env_state[4 * b + 1];  // frames since episode start
+        const int E = gridDim.x;
+        #pragma unroll
+        for (int ch = 0; ch < 4; ++ch) {
+            const int back = 3 - ch;
+            const int slot = (ring_head - back + 8) & 3;
+            rb[ch] = ring + ((int64_t)slot * E + b) * 7056;
+            rs[ch] = (steps >= back) ? (1.0f / 255.0f) : 0.0f;
+        }
+    }
 
     // ---- conv1: M=400 (25 tiles), N=32 (2 tiles), K=256 (8 steps) --------
     // All 16 B-fragments live in registers; the m-loop runs on L2-cached
@@ -99,7 +122,19 @@ __global__ __launch_bounds__(512) void conv_encode_kernel(
             const int c = k0 >> 6, ky = (k0 >> 3) & 7;
             // A: 8 taps = input row segment [c][oy*4+ky][ox*4 .. +7]
             u16x8 av;
-            {
+            if (GATHER) {
+                // c == ks>>1 is wave-uniform, so rb[c]/rs[c] stay scalar
+                const uint8_t* p = rb[c] + (oy * 4 + ky) * 84 + ox * 4;
+                const uint32_t lo = *(const uint32_t*)p;
+                const uint32_t hi = *(const uint32_t*)(p + 4);
+                const float sc = rs[c];
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    av[j] = f32_to_bf16((float)((lo >> (8 * j)) & 0xFF) * sc);
+                    av[4 + j] =
+                        f32_to_bf16((float)((hi >> (8 * j)) & 0xFF) * sc);
+                }
+            } else {
                 const int idx = (c * 84 + oy * 4 + ky) * 84 + ox * 4;
                 u16x4 lo = *(const u16x4*)(src + idx);
                 u16x4 hi = *(const u16x4*)(src + idx + 4);
@@ -771,13 +806,30 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* w2, const void* b2, const void* w3p,
                       const void* b3, void* feat, void* c1g, void* c2g, int B,
                       void* stream) {
-    hipLaunchKernelGGL(conv_encode_kernel, dim3(B), dim3(512), 0,
+    hipLaunchKernelGGL(conv_encode_kernel<0>, dim3(B), dim3(512), 0,
                        (hipStream_t)stream, (const unsigned short*)x,
+                       (const uint8_t*)nullptr, (const int32_t*)nullptr, 0,
                        (const unsigned short*)w1, (const unsigned short*)b1,
                        (const unsigned short*)w2, (const unsigned short*)b2,
                        (const unsigned short*)w3p, (const unsigned short*)b3,
                        (unsigned short*)feat, (unsigned short*)c1g,
                        (unsigned short*)c2g);
+}
+
+void riqn_conv_encode_gather(const void* ring, const void* env_state,
+                             int ring_head, const void* w1, const void* b1,
+                             const void* w2, const void* b2, const void* w3p,
+                             const void* b3, void* feat, int E,
+                             void* stream) {
+    hipLaunchKernelGGL(conv_encode_kernel<1>, dim3(E), dim3(512), 0,
+                       (hipStream_t)stream, (const unsigned short*)nullptr,
+                       (const uint8_t*)ring, (const int32_t*)env_state,
+                       ring_head,
+                       (const unsigned short*)w1, (const unsigned short*)b1,
+                       (const unsigned short*)w2, (const unsigned short*)b2,
+                       (const unsigned short*)w3p, (const unsigned short*)b3,
+                       (unsigned short*)feat, (unsigned short*)nullptr,
+                       (unsigned short*)nullptr);
 }
 
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,

@@ -91,6 +91,10 @@ void riqn_conv_encode(const void* x, const void* w1, const void* b1,
                       const void* w2, const void* b2, const void* w3p,
                       const void* b3, void* feat, void* c1g, void* c2g, int B,
                       void* stream);
+void riqn_conv_encode_gather(const void* ring, const void* env_state,
+                             int ring_head, const void* w1, const void* b1,
+                             const void* w2, const void* b2, const void* w3p,
+                             const void* b3, void* feat, int E, void* stream);
 void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     const void* feat, void* h, void* phi_out, void* cosb_out,
                     int R, int nq, int direct, void* stream);

@@ -170,6 +170,31 @@ class FusedForward:
                               actions)
         return actions
 
+    def act_gather(self, ring: torch.Tensor, env_state: torch.Tensor,
+                   history: int, ring_head: int, taus, nq: int,
+                   actions: torch.Tensor) -> torch.Tensor:
+        """Greedy action straight from the u8 frame ring: conv1 gathers and
+        normalizes the frame stack itself (fused K15), so the bf16 stack
+        tensor never materializes (actor hot path)."""
+        E = env_state.numel() // 4
+        bufs = self._bufs(E, nq)
+        feat, h, hh = bufs[:3]
+        self.ext.conv_encode_gather(ring, env_state, history, ring_head,
+                                    self.w1, self.b1, self.w2, self.b2,
+                                    self.w3p, self.b3, feat)
+        self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h, nq,
+                           direct=TAU_DIRECT)
+        if self.fp8_heads:
+            hq, hsc = bufs[3], bufs[4]
+            self.ext.quant_fp8(h, hq, hsc)
+            self.ext.heads_gemm_fp8(hq, hsc, self.wq, self.wsc, self.bcat, hh)
+        else:
+            self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
+        dummy = torch.zeros(1, dtype=torch.float32, device=self.device)
+        self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 1, dummy,
+                              actions)
+        return actions
+
     def act_from_feat(self, featv: torch.Tensor, taus, nq: int,
                       actions: torch.Tensor) -> torch.Tensor:
         """Greedy action from precomputed conv features (lets the learner

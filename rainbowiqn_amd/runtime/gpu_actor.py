@@ -80,15 +80,19 @@ class GpuActorLoop:
                 self.fused.pack_weights()
         newest = self._newest
         for s in range(self.chunk_steps):
-            self.ext.stack_gather(
-                self.ring.view(self.h, -1), self.env_state.view(-1), self.h,
-                newest, self.E, self.stack_f32,
-            )
             self.ext.rand_uniform(self.taus, self.rng_state)
             if self.fused is not None:
-                self.fused.act(self.stack_f32, self.taus.view(-1), self.K,
-                               self.actions)
+                # conv1 gathers the frame stack from the u8 ring itself
+                # (fused K15); the bf16 stack tensor never materializes
+                self.fused.act_gather(self.ring.view(self.h, -1, 7056),
+                                      self.env_state.view(-1), self.h,
+                                      newest, self.taus.view(-1), self.K,
+                                      self.actions)
             else:
+                self.ext.stack_gather(
+                    self.ring.view(self.h, -1), self.env_state.view(-1),
+                    self.h, newest, self.E, self.stack_f32,
+                )
                 with torch.no_grad():
                     z, _ = self.net(self.stack_f32, self.K, taus=self.taus)
                     self.actions.copy_(z.float().mean(dim=1).argmax(dim=1))

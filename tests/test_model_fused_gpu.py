@@ -60,6 +60,37 @@ def test_conv_encode_parity():
     assert err < max(3 * err16, 0.02 * max(scale, 1.0)), (err, err16)
 
 
+def test_conv_encode_gather_matches_staged_stack():
+    """Fused frame-stack gather (conv1 reads the u8 ring directly) produces
+    bit-identical features to stack_gather -> conv_encode, including the
+    pre-episode zeroing of short episodes."""
+    from rainbowiqn_amd.ops import get_extension
+
+    ext = get_extension()
+    _, _, net16 = make_nets(seed=5)
+    fw = fused_of(net16)
+    E, h = 64, 4
+    torch.manual_seed(7)
+    ring = torch.randint(0, 256, (h, E, 7056), dtype=torch.uint8,
+                         device="cuda")
+    env_state = torch.zeros(E, 4, dtype=torch.int32, device="cuda")
+    # mix of fresh episodes (steps 0..2 -> zeroed history) and running ones
+    env_state[:, 1] = torch.tensor([i % 6 for i in range(E)],
+                                   dtype=torch.int32, device="cuda")
+    for ring_head in range(h):
+        stack = torch.empty(E, h, 84, 84, dtype=torch.bfloat16,
+                            device="cuda")
+        ext.stack_gather(ring.view(h, -1), env_state.view(-1), h, ring_head,
+                         E, stack)
+        ref = fw.encode(stack).clone()
+        feat = torch.empty(E, 3136, dtype=torch.bfloat16, device="cuda")
+        ext.conv_encode_gather(ring, env_state.view(-1), h, ring_head,
+                               fw.w1, fw.b1, fw.w2, fw.b2, fw.w3p, fw.b3,
+                               feat)
+        torch.cuda.synchronize()
+        assert torch.equal(feat, ref), ring_head
+
+
 def test_tau_embed_direct_matches_staged():
     """tau_embed DIRECT=1 (accumulator stores, no LDS staging) is bit-exact
     vs the staged variant on h, phi and cosb -- same math, different path."""
