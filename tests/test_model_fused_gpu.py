@@ -119,6 +119,26 @@ def test_tau_embed_direct_matches_staged():
         assert torch.equal(p1, p2) and torch.equal(c1, c2), (B, nq)
 
 
+def test_heads_gemm_large_r_parity():
+    """R=8192 routes to the 256x128 big-tile kernel: check vs torch matmul
+    (fp32 accumulate both sides), including a ragged-R launch."""
+    from rainbowiqn_amd.ops import get_extension
+
+    ext = get_extension()
+    _, _, net16 = make_nets(seed=9)
+    fw = fused_of(net16)
+    for R in (8192, 8192 + 128):  # ragged: last 256-row block half full
+        h = (torch.randn(R, 3136, device="cuda") * 0.1).to(torch.bfloat16)
+        hh = torch.empty(R, 1024, dtype=torch.bfloat16, device="cuda")
+        ext.heads_gemm(h, fw.wcat, fw.bcat, hh)
+        torch.cuda.synchronize()
+        ref = torch.relu(h.float() @ fw.wcat.float().t()
+                         + fw.bcat.float())
+        err = (hh.float() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err < 0.02 * max(scale, 1.0) + 0.05, (R, err, scale)
+
+
 def test_full_forward_parity():
     A = 6
     _, net32, net16 = make_nets(A)
