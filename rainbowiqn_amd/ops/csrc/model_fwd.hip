@@ -681,13 +681,14 @@ __global__ __launch_bounds__(256) void heads_gemm_big_kernel(
 // 2-buffer glds (96 KB LDS, one WG/CU).  vs the 128x128 tile this moves
 // 28% fewer operand bytes -- PMC showed the 128x128 version saturating L2
 // at ~15 TB/s (878 MB/dispatch at R=8192), so tile growth is the lever.
+template <int NBUF>  // 2 or 3 buffers (3 = 144 KB LDS, deeper pipe)
 __global__ __launch_bounds__(512) void heads_gemm_big2_kernel(
     const unsigned short* __restrict__ hin,  // [R,3136]
     const unsigned short* __restrict__ wcat, // [1024,3136]
     const unsigned short* __restrict__ bcat, // [1024]
     unsigned short* __restrict__ out,        // [R,1024]
     int R) {
-    __shared__ unsigned short smem[2 * (256 + 128) * 64];
+    __shared__ unsigned short smem[NBUF * (256 + 128) * 64];
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;   // 8 waves
     const int wr = wave >> 1, wc = wave & 1;  // 4x2 grid of 64x64 quadrants
@@ -737,11 +738,18 @@ __global__ __launch_bounds__(512) void heads_gemm_big2_kernel(
         #pragma unroll
         for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
     stage(0, 0);
+    if (NBUF == 3) stage(1, 64);
     for (int kt = 0; kt < 3136 / 64; ++kt) {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        // wait own glds for buf[kt]; with NBUF=3 the newest stage's 6 stay
+        // in flight across the barrier (counted vmcnt, not a full drain)
+        if (NBUF == 3 && kt + 1 < 3136 / 64)
+            asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
-        if (kt + 1 < 3136 / 64) stage((kt + 1) & 1, (kt + 1) * 64);
-        const unsigned short* As = smem + (kt & 1) * ((256 + 128) * 64);
+        if (kt + NBUF - 1 < 3136 / 64)
+            stage((kt + NBUF - 1) % NBUF, (kt + NBUF - 1) * 64);
+        const unsigned short* As = smem + (kt % NBUF) * ((256 + 128) * 64);
         const unsigned short* Bs = As + 256 * 64;
         #pragma unroll
         for (int ksub = 0; ksub < 2; ++ksub) {
@@ -954,7 +962,7 @@ void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
                      void* out, int R, void* stream) {
     if (R >= 8192) {
         dim3 gridb((R + 255) / 256, 1024 / 128);
-        hipLaunchKernelGGL(heads_gemm_big2_kernel, gridb, dim3(512), 0,
+        hipLaunchKernelGGL(heads_gemm_big2_kernel<3>, gridb, dim3(512), 0,
                            (hipStream_t)stream, (const unsigned short*)h,
                            (const unsigned short*)wcat,
                            (const unsigned short*)bcat, (unsigned short*)out,
