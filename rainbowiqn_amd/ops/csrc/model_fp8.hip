@@ -78,7 +78,7 @@ __global__ __launch_bounds__(256) void heads_gemm_fp8_kernel(
     const int lane = t & 63, wave = t >> 6;
     const int wr = wave >> 1, wc = wave & 1;
     const int l15 = lane & 15, l4 = lane >> 4;
-    // XCD-aware tile map (see heads_gemm_big_kernel): each XCD takes a
+    // XCD-aware tile map (see heads_gemm_big2_kernel): each XCD takes a
     // contiguous row-block band x all n-blocks so co-resident WGs share
     // operands through their XCD's L2.
     int bxi = blockIdx.x, byi = blockIdx.y;

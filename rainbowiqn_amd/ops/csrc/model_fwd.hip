@@ -576,107 +576,6 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
 // 128x128-tile variant for large R (act path): 4x the MFMA per barrier and
 // half the on-chip traffic per FLOP vs the 64x64 tile.  Each of 4 waves owns
 // a 64x64 quadrant (4x4 fragments); 2 glds buffers, raw barrier + counted
-// waits, same XOR source swizzle as the small-tile kernel.
-__global__ __launch_bounds__(256) void heads_gemm_big_kernel(
-    const unsigned short* __restrict__ hin,  // [R,3136]
-    const unsigned short* __restrict__ wcat, // [1024,3136]
-    const unsigned short* __restrict__ bcat, // [1024]
-    unsigned short* __restrict__ out,        // [R,1024]
-    int R) {
-    __shared__ unsigned short smem[2 * 2 * 128 * 64];  // 2 bufs x (A|B 16KB)
-    const int t = threadIdx.x;
-    const int lane = t & 63, wave = t >> 6;
-    const int wr = wave >> 1, wc = wave & 1;
-    const int l15 = lane & 15, l4 = lane >> 4;
-    // XCD-aware tile map: the dispatcher assigns WGs to XCDs round-robin on
-    // the linear id, so id%8 IS the XCD.  Give each XCD a contiguous band of
-    // row-blocks x all n-blocks: the 64 co-resident WGs of one XCD then
-    // share their A rows and B cols through that XCD's L2 (per k-window the
-    // joint working set is ~256 KB) instead of re-pulling A from HBM once
-    // per n-block (MI355X_MICROARCH.md XCD/L2 locality rule).
-    int bxi = blockIdx.x, byi = blockIdx.y;
-    if ((gridDim.x & 7) == 0) {
-        const int id = blockIdx.x + blockIdx.y * gridDim.x;
-        const int xcd = id & 7, s = id >> 3;
-        byi = s % gridDim.y;
-        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
-    }
-    const int bm = bxi * 128, bn = byi * 128;
-
-    // 16 glds insts cover a 128x64 operand; wave w issues insts [4w, 4w+4)
-    // for A and for B (8 per wave per buffer).
-    auto stage = [&](int buf, int k0) {
-        unsigned short* base = smem + buf * (2 * 128 * 64);
-        #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-            const int inst = wave * 4 + i;
-            const int row = inst * 8 + (lane >> 3);
-            const int piece = (lane & 7) ^ (row & 7);
-            {
-                const int gr = min(bm + row, R - 1);
-                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
-                    (hin + (int64_t)gr * 3136 + k0 + piece * 8);
-                auto* lp = (__attribute__((address_space(3))) uint32_t*)
-                    (base + inst * 512);
-                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
-            }
-            {
-                auto* gp = (const __attribute__((address_space(1))) uint32_t*)
-                    (wcat + (int64_t)(bn + row) * 3136 + k0 + piece * 8);
-                auto* lp = (__attribute__((address_space(3))) uint32_t*)
-                    (base + 128 * 64 + inst * 512);
-                __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
-            }
-        }
-    };
-
-    f32x4 acc[4][4];
-    #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
-        #pragma unroll
-        for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
-    stage(0, 0);
-    for (int kt = 0; kt < 3136 / 64; ++kt) {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
-        if (kt + 1 < 3136 / 64) stage((kt + 1) & 1, (kt + 1) * 64);
-        const unsigned short* As = smem + (kt & 1) * (2 * 128 * 64);
-        const unsigned short* Bs = As + 128 * 64;
-        #pragma unroll
-        for (int ksub = 0; ksub < 2; ++ksub) {
-            #pragma unroll
-            for (int fm = 0; fm < 4; ++fm) {
-                const int row = wr * 64 + fm * 16 + l15;
-                const int piece = (ksub * 4 + l4) ^ (row & 7);
-                bf16x8 a = as_bf16x8(*(const u16x8*)&As[row * 64 + piece * 8]);
-                #pragma unroll
-                for (int fn = 0; fn < 4; ++fn) {
-                    const int nrow = wc * 64 + fn * 16 + l15;
-                    const int bp = (ksub * 4 + l4) ^ (nrow & 7);
-                    bf16x8 b = as_bf16x8(*(const u16x8*)&Bs[nrow * 64 + bp * 8]);
-                    acc[fm][fn] = mfma16(a, b, acc[fm][fn]);
-                }
-            }
-        }
-    }
-    __builtin_amdgcn_s_barrier();
-    #pragma unroll
-    for (int fm = 0; fm < 4; ++fm) {
-        #pragma unroll
-        for (int fn = 0; fn < 4; ++fn) {
-            const int n = bn + wc * 64 + fn * 16 + l15;
-            const float bias = bf(bcat[n]);
-            #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                const int r = bm + wr * 64 + fm * 16 + l4 * 4 + j;
-                if (r < R)
-                    out[(int64_t)r * 1024 + n] =
-                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
-            }
-        }
-    }
-}
-
 // 256x128 block tile, 512 threads (8 waves on 64x64 quadrants), BK=64,
 // 2-buffer glds (96 KB LDS, one WG/CU).  vs the 128x128 tile this moves
 // 28% fewer operand bytes -- PMC showed the 128x128 version saturating L2
@@ -693,7 +592,10 @@ __global__ __launch_bounds__(512) void heads_gemm_big2_kernel(
     const int lane = t & 63, wave = t >> 6;   // 8 waves
     const int wr = wave >> 1, wc = wave & 1;  // 4x2 grid of 64x64 quadrants
     const int l15 = lane & 15, l4 = lane >> 4;
-    // XCD-aware tile map (see heads_gemm_big_kernel)
+    // XCD-aware tile map: the dispatcher assigns WGs to XCDs round-robin
+    // on the linear id, so id%8 IS the XCD; give each XCD a contiguous
+    // row-block band x all n-blocks so its co-resident WGs share operands
+    // through that XCD's L2 (MI355X_MICROARCH.md XCD/L2 locality rule).
     int bxi = blockIdx.x, byi = blockIdx.y;
     if ((gridDim.x & 7) == 0) {
         const int id = blockIdx.x + blockIdx.y * gridDim.x;
