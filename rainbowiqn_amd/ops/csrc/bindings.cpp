@@ -316,6 +316,24 @@ void conv_encode_gather(torch::Tensor ring, torch::Tensor env_state,
                             cur_stream());
 }
 
+void heads_gemm_tau(torch::Tensor taus, torch::Tensor wiqnf,
+                    torch::Tensor biqn, torch::Tensor feat,
+                    torch::Tensor wcat, torch::Tensor bcat,
+                    torch::Tensor out, int64_t nq) {
+    CHECK_T(taus, torch::kFloat32);
+    CHECK_BF16(wiqnf); CHECK_BF16(biqn); CHECK_BF16(feat);
+    CHECK_BF16(wcat); CHECK_BF16(bcat); CHECK_BF16(out);
+    int R = (int)taus.numel();
+    TORCH_CHECK(wiqnf.numel() == 392 * 512 && biqn.numel() == 3136);
+    TORCH_CHECK(feat.numel() * nq == (int64_t)R * 3136, "feat rows * nq != R");
+    TORCH_CHECK(wcat.numel() == 1024 * 3136);
+    TORCH_CHECK(out.numel() == (int64_t)R * 1024);
+    riqn_heads_gemm_tau(taus.data_ptr<float>(), wiqnf.data_ptr(),
+                        biqn.data_ptr(), feat.data_ptr(), wcat.data_ptr(),
+                        bcat.data_ptr(), out.data_ptr(), R, (int)nq,
+                        cur_stream());
+}
+
 void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
                torch::Tensor feat, torch::Tensor h, int64_t nq,
                c10::optional<torch::Tensor> phi_out = c10::nullopt,
@@ -544,6 +562,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("heads_gemm_fp8", &heads_gemm_fp8);
     m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
     m.def("conv_encode_gather", &conv_encode_gather, py::arg("ring"), py::arg("env_state"), py::arg("history"), py::arg("ring_head"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"));
+    m.def("heads_gemm_tau", &heads_gemm_tau, py::arg("taus"), py::arg("wiqnf"), py::arg("biqn"), py::arg("feat"), py::arg("wcat"), py::arg("bcat"), py::arg("out"), py::arg("nq"));
     m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt, py::arg("direct") = -1);
     m.def("heads_gemm", &heads_gemm);
     m.def("heads_finish", &heads_finish);

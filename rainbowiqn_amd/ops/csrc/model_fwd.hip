@@ -688,6 +688,178 @@ __global__ __launch_bounds__(512) void heads_gemm_big2_kernel(
     }
 }
 
+// Actor-path fusion of tau_embed INTO the heads GEMM (K4-K6 as one kernel):
+// h[r,k] = relu(cos_r . Wiqn[k,:] + b_iqn[k]) * feat[r/nq, k] is produced
+// per 256x64 k-tile straight into the LDS A-image the main MFMA loop reads
+// -- h [R,3136] never exists in HBM (saves its 51 MB write + 8x re-read at
+// R=8192 and the whole tau_embed dispatch).  The per-wave cos fragments
+// don't depend on k, so they are built once (Chebyshev recurrence, 3 cosf)
+// and reused for all 49 k-tiles.  bf16 rounding matches tau_embed exactly
+// (round after relu, then after the feat multiply), so the output is
+// bit-identical to tau_embed + heads_gemm.  The learner keeps the unfused
+// pair (it needs h/phi for the backward).
+__global__ __launch_bounds__(512) void heads_gemm_tau_kernel(
+    const float* __restrict__ taus,          // [R]
+    const unsigned short* __restrict__ wiqnf,// [392][512] fragment-major
+    const unsigned short* __restrict__ biqn, // [3136] bf16
+    const unsigned short* __restrict__ feat, // [R/nq, 3136] bf16
+    const unsigned short* __restrict__ wcat, // [1024,3136] bf16
+    const unsigned short* __restrict__ bcat, // [1024] bf16
+    unsigned short* __restrict__ out,        // [R,1024] bf16
+    int R, int nq) {
+    __shared__ unsigned short aimg[256 * 64];      // computed A-tile (32 KB)
+    __shared__ unsigned short bsm[2 * 128 * 64];   // staged wcat (2 x 16 KB)
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;   // 8 waves
+    const int wr = wave >> 1, wc = wave & 1;  // main-GEMM quadrant map
+    const int l15 = lane & 15, l4 = lane >> 4;
+    int bxi = blockIdx.x, byi = blockIdx.y;
+    if ((gridDim.x & 7) == 0) {
+        const int id = blockIdx.x + blockIdx.y * gridDim.x;
+        const int xcd = id & 7, s = id >> 3;
+        byi = s % gridDim.y;
+        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
+    }
+    const int bm = bxi * 256, bn = byi * 128;
+
+    // cos fragments for this wave's 32 A'-rows (k-invariant; tau_embed's
+    // recurrence verbatim so the bf16 basis values match bit-for-bit)
+    bf16x8 afrag[2][2];
+    #pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+        const int r_a = bm + wave * 32 + mi * 16 + l15;
+        const float tau = (r_a < R) ? taus[r_a] : 0.0f;
+        const float th = 3.14159265358979f * tau;
+        const int kbase = l4 * 8;
+        float cm1 = cosf((float)kbase * th);
+        float c0 = cosf((float)(kbase + 1) * th);
+        const float two_ct = 2.0f * cosf(th);
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+            u16x8 av;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                if (j == 0) {
+                    av[j] = f32_to_bf16(cm1);
+                } else if (j == 1) {
+                    av[j] = f32_to_bf16(c0);
+                } else {
+                    float c1 = two_ct * c0 - cm1;
+                    cm1 = c0;
+                    c0 = c1;
+                    av[j] = f32_to_bf16(c1);
+                }
+            }
+            if (ks == 0) {
+                cm1 = cosf((float)(kbase + 32) * th);
+                c0 = two_ct * cm1 - cosf((float)(kbase + 31) * th);
+            }
+            afrag[mi][ks] = as_bf16x8(av);
+        }
+    }
+
+    // wcat staging: 16 glds insts per buffer, wave w issues [2w, 2w+2)
+    auto stageb = [&](int buf, int k0) {
+        unsigned short* base = bsm + buf * (128 * 64);
+        #pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            const int inst = wave * 2 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (wcat + (int64_t)(bn + row) * 3136 + k0 + piece * 8);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + inst * 512);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][4];
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stageb(0, 0);
+    const int smax = min((R - 1) / nq, 0x7fffffff);
+    for (int kt = 0; kt < 3136 / 64; ++kt) {
+        // A-image from last iteration fully consumed before overwrite
+        __builtin_amdgcn_s_barrier();
+        if (kt + 1 < 3136 / 64) stageb((kt + 1) & 1, (kt + 1) * 64);
+        // ---- produce A' = bf16(relu(cos.Wiqn + b)) * feat for this k-tile
+        // (B DMA for kt+1 rides under this); wave owns rows [32w, 32w+32)
+        #pragma unroll
+        for (int ntl = 0; ntl < 4; ++ntl) {
+            const int ntg = kt * 4 + ntl;          // global 16-col tile
+            const int kcol = ntg * 16 + l15;
+            f32x4 acc2[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+            #pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                bf16x8 bfr = as_bf16x8(
+                    *(const u16x8*)(wiqnf + ((ntg * 2 + ks) << 9) + lane * 8));
+                acc2[0] = mfma16(afrag[0][ks], bfr, acc2[0]);
+                acc2[1] = mfma16(afrag[1][ks], bfr, acc2[1]);
+            }
+            const float bb = bf(biqn[kcol]);
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi) {
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int rl = wave * 32 + mi * 16 + l4 * 4 + j;
+                    const unsigned short ph =
+                        f32_to_bf16(fmaxf(acc2[mi][j] + bb, 0.0f));
+                    const int s = min((bm + rl) / nq, smax);
+                    const float fv = bf(feat[(int64_t)s * 3136 + kcol]);
+                    const int cl = ntl * 16 + l15;   // col within k-tile
+                    aimg[rl * 64 + (((cl >> 3) ^ (rl & 7)) << 3) + (cl & 7)] =
+                        f32_to_bf16(bf16_to_f32(ph) * fv);
+                }
+            }
+        }
+        // buf[kt] done; at the tail nothing newer is in flight, so drain
+        if (kt + 1 < 3136 / 64)
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        // ---- main MFMA on the computed A-image + staged wcat tile -------
+        const unsigned short* Bs = bsm + (kt & 1) * (128 * 64);
+        #pragma unroll
+        for (int ksub = 0; ksub < 2; ++ksub) {
+            #pragma unroll
+            for (int fm = 0; fm < 4; ++fm) {
+                const int row = wr * 64 + fm * 16 + l15;
+                const int piece = (ksub * 4 + l4) ^ (row & 7);
+                bf16x8 a =
+                    as_bf16x8(*(const u16x8*)&aimg[row * 64 + piece * 8]);
+                #pragma unroll
+                for (int fn = 0; fn < 4; ++fn) {
+                    const int nrow = wc * 64 + fn * 16 + l15;
+                    const int bp = (ksub * 4 + l4) ^ (nrow & 7);
+                    bf16x8 b =
+                        as_bf16x8(*(const u16x8*)&Bs[nrow * 64 + bp * 8]);
+                    acc[fm][fn] = mfma16(a, b, acc[fm][fn]);
+                }
+            }
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+            const int n = bn + wc * 64 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * 64 + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
 // MODE 0: q [R,A] f32 out.  MODE 2: theta[r] = q[r][act[r/nq]] out.
 // Two-pass per row: advantage sums stashed in LDS (sv) -- no dynamically
 // indexed register array (which spills to scratch).
@@ -858,6 +1030,21 @@ void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                        (const unsigned short*)feat, (unsigned short*)h,
                        (unsigned short*)phi_out, (unsigned short*)cosb_out,
                        R, nq);
+}
+
+void riqn_heads_gemm_tau(const float* taus, const void* wiqnf,
+                         const void* biqn, const void* feat, const void* wcat,
+                         const void* bcat, void* out, int R, int nq,
+                         void* stream) {
+    dim3 grid((R + 255) / 256, 1024 / 128);
+    hipLaunchKernelGGL(heads_gemm_tau_kernel, grid, dim3(512), 0,
+                       (hipStream_t)stream, taus,
+                       (const unsigned short*)wiqnf,
+                       (const unsigned short*)biqn,
+                       (const unsigned short*)feat,
+                       (const unsigned short*)wcat,
+                       (const unsigned short*)bcat, (unsigned short*)out, R,
+                       nq);
 }
 
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,

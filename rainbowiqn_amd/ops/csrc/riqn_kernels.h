@@ -100,6 +100,10 @@ void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     int R, int nq, int direct, void* stream);
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
                      void* out, int R, void* stream);
+void riqn_heads_gemm_tau(const float* taus, const void* wiqnf,
+                         const void* biqn, const void* feat, const void* wcat,
+                         const void* bcat, void* out, int R, int nq,
+                         void* stream);
 void riqn_heads_finish(const void* hh, const void* wz, const void* bz, int R,
                        int A, int nq, int mode, float* q_out, int64_t* act_io,
                        void* stream);

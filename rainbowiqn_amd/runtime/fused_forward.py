@@ -182,14 +182,17 @@ class FusedForward:
         self.ext.conv_encode_gather(ring, env_state, history, ring_head,
                                     self.w1, self.b1, self.w2, self.b2,
                                     self.w3p, self.b3, feat)
-        self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h, nq,
-                           direct=TAU_DIRECT)
         if self.fp8_heads:
+            self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h,
+                               nq, direct=TAU_DIRECT)
             hq, hsc = bufs[3], bufs[4]
             self.ext.quant_fp8(h, hq, hsc)
             self.ext.heads_gemm_fp8(hq, hsc, self.wq, self.wsc, self.bcat, hh)
         else:
-            self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
+            # tau-embed fused into the GEMM's A-operand production: the
+            # [R,3136] h tensor never exists (bit-identical to the pair)
+            self.ext.heads_gemm_tau(taus.view(-1), self.wiqn, self.biqn,
+                                    feat, self.wcat, self.bcat, hh, nq)
         dummy = torch.zeros(1, dtype=torch.float32, device=self.device)
         self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 1, dummy,
                               actions)

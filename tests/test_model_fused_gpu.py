@@ -119,6 +119,30 @@ def test_tau_embed_direct_matches_staged():
         assert torch.equal(p1, p2) and torch.equal(c1, c2), (B, nq)
 
 
+def test_heads_gemm_tau_fused_matches_pair():
+    """tau-embed fused into the heads GEMM (actor path) is bit-identical to
+    tau_embed -> heads_gemm, including clamped tail rows."""
+    from rainbowiqn_amd.ops import get_extension
+
+    ext = get_extension()
+    _, _, net16 = make_nets(seed=13)
+    fw = fused_of(net16)
+    for B, nq in [(1024, 8), (256, 32), (100, 8)]:  # 100*8: ragged tail
+        R = B * nq
+        x = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
+        feat = fw.encode(x).clone()
+        taus = torch.rand(R, device="cuda")
+        h = torch.empty(R, 3136, dtype=torch.bfloat16, device="cuda")
+        hh_ref = torch.empty(R, 1024, dtype=torch.bfloat16, device="cuda")
+        hh = torch.empty_like(hh_ref)
+        ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq)
+        ext.heads_gemm(h, fw.wcat, fw.bcat, hh_ref)
+        ext.heads_gemm_tau(taus, fw.wiqn, fw.biqn, feat, fw.wcat, fw.bcat,
+                           hh, nq)
+        torch.cuda.synchronize()
+        assert torch.equal(hh, hh_ref), (B, nq)
+
+
 def test_heads_gemm_large_r_parity():
     """R=8192 routes to the 256x128 big-tile kernel: check vs torch matmul
     (fp32 accumulate both sides), including a ragged-R launch."""
