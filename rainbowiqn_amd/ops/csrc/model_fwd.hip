@@ -707,7 +707,7 @@ __global__ __launch_bounds__(512) void heads_gemm_tau_kernel(
     const unsigned short* __restrict__ bcat, // [1024] bf16
     unsigned short* __restrict__ out,        // [R,1024] bf16
     int R, int nq) {
-    __shared__ unsigned short aimg[256 * 64];      // computed A-tile (32 KB)
+    __shared__ unsigned short aimg[2][256 * 64];   // computed A (2 x 32 KB)
     __shared__ unsigned short bsm[2 * 128 * 64];   // staged wcat (2 x 16 KB)
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;   // 8 waves
@@ -774,19 +774,11 @@ __global__ __launch_bounds__(512) void heads_gemm_tau_kernel(
         }
     };
 
-    f32x4 acc[4][4];
-    #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
-        #pragma unroll
-        for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
-    stageb(0, 0);
-    const int smax = min((R - 1) / nq, 0x7fffffff);
-    for (int kt = 0; kt < 3136 / 64; ++kt) {
-        // A-image from last iteration fully consumed before overwrite
-        __builtin_amdgcn_s_barrier();
-        if (kt + 1 < 3136 / 64) stageb((kt + 1) & 1, (kt + 1) * 64);
-        // ---- produce A' = bf16(relu(cos.Wiqn + b)) * feat for this k-tile
-        // (B DMA for kt+1 rides under this); wave owns rows [32w, 32w+32)
+    // A'-producer for one 256x64 k-tile (wave owns rows [32w, 32w+32));
+    // bf16 rounding matches tau_embed bit-for-bit
+    const int smax = (R - 1) / nq;
+    auto produceA = [&](int kt) {
+        unsigned short* dst = aimg[kt & 1];
         #pragma unroll
         for (int ntl = 0; ntl < 4; ++ntl) {
             const int ntg = kt * 4 + ntl;          // global 16-col tile
@@ -810,18 +802,32 @@ __global__ __launch_bounds__(512) void heads_gemm_tau_kernel(
                     const int s = min((bm + rl) / nq, smax);
                     const float fv = bf(feat[(int64_t)s * 3136 + kcol]);
                     const int cl = ntl * 16 + l15;   // col within k-tile
-                    aimg[rl * 64 + (((cl >> 3) ^ (rl & 7)) << 3) + (cl & 7)] =
+                    dst[rl * 64 + (((cl >> 3) ^ (rl & 7)) << 3) + (cl & 7)] =
                         f32_to_bf16(bf16_to_f32(ph) * fv);
                 }
             }
         }
-        // buf[kt] done; at the tail nothing newer is in flight, so drain
-        if (kt + 1 < 3136 / 64)
-            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-        else
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
-        // ---- main MFMA on the computed A-image + staged wcat tile -------
+    };
+
+    f32x4 acc[4][4];
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stageb(0, 0);
+    produceA(0);
+    // software pipeline, ONE barrier per iteration: produce A(kt+1) and
+    // run the main MFMA on A(kt) back-to-back (independent LDS buffers, so
+    // the scheduler interleaves the two instruction chains); the barrier
+    // publishes A(kt+1) + guards B-buffer reuse for the next iteration.
+    for (int kt = 0; kt < 3136 / 64; ++kt) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // B buf[kt] landed
+        __builtin_amdgcn_s_barrier();                     // + A(kt) visible
+        if (kt + 1 < 3136 / 64) {
+            stageb((kt + 1) & 1, (kt + 1) * 64);
+            produceA(kt + 1);
+        }
+        const unsigned short* As = aimg[kt & 1];
         const unsigned short* Bs = bsm + (kt & 1) * (128 * 64);
         #pragma unroll
         for (int ksub = 0; ksub < 2; ++ksub) {
@@ -830,7 +836,7 @@ __global__ __launch_bounds__(512) void heads_gemm_tau_kernel(
                 const int row = wr * 64 + fm * 16 + l15;
                 const int piece = (ksub * 4 + l4) ^ (row & 7);
                 bf16x8 a =
-                    as_bf16x8(*(const u16x8*)&aimg[row * 64 + piece * 8]);
+                    as_bf16x8(*(const u16x8*)&As[row * 64 + piece * 8]);
                 #pragma unroll
                 for (int fn = 0; fn < 4; ++fn) {
                     const int nrow = wc * 64 + fn * 16 + l15;
