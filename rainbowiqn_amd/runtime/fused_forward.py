@@ -34,6 +34,11 @@ from ..ops import get_extension
 # staged still ahead below R=4096); RIQN_TAU_V2=0/1 forces a variant.
 TAU_DIRECT = int(os.environ.get("RIQN_TAU_V2", "-1"))
 
+# RIQN_FUSE_TAU_GEMM=1: route the actor through the single fused
+# tau-embed+heads-GEMM kernel (parity-tested, but measured slower -- see
+# act_gather; kept as a reachable, gated experiment).
+FUSE_TAU_GEMM = os.environ.get("RIQN_FUSE_TAU_GEMM", "0") == "1"
+
 
 class FusedForward:
     @classmethod
@@ -188,11 +193,18 @@ class FusedForward:
             hq, hsc = bufs[3], bufs[4]
             self.ext.quant_fp8(h, hq, hsc)
             self.ext.heads_gemm_fp8(hq, hsc, self.wq, self.wsc, self.bcat, hh)
-        else:
-            # tau-embed fused into the GEMM's A-operand production: the
-            # [R,3136] h tensor never exists (bit-identical to the pair)
+        elif FUSE_TAU_GEMM:
+            # experiment (off by default): tau-embed fused into the GEMM's
+            # A-operand production.  Bit-identical to the pair but measured
+            # SLOWER end-to-end (10.6 vs 6.3 ms/round at the bench config:
+            # the scattered feat loads + A-image LDS writes stall the main
+            # MFMA chain harder than the h round-trip costs).
             self.ext.heads_gemm_tau(taus.view(-1), self.wiqn, self.biqn,
                                     feat, self.wcat, self.bcat, hh, nq)
+        else:
+            self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h,
+                               nq, direct=TAU_DIRECT)
+            self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
         dummy = torch.zeros(1, dtype=torch.float32, device=self.device)
         self.ext.heads_finish(hh, self.wz, self.bz, self.A, nq, 1, dummy,
                               actions)
