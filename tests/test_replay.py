@@ -138,3 +138,59 @@ def test_chunk_append_matches_single(cfg):
     assert (m1.frames[:40] == m2.frames[:40]).all()
     assert (m1.timesteps[:40] == m2.timesteps[:40]).all()
     assert m1.tree.total() == pytest.approx(m2.tree.total())
+
+
+def test_randomized_differential_vs_bruteforce(cfg):
+    """Randomized episodes: every sampled transition's frame stack, n-step
+    return, nonterminal flag and next-state stack must match a brute-force
+    replayer of the raw stream (reference memory.py semantics)."""
+    rng = np.random.default_rng(42)
+    mem = ReplayMemory(cfg, capacity=512)
+    N = 400
+    frames = rng.integers(0, 256, size=(N, 84, 84)).astype(np.uint8)
+    actions = rng.integers(0, 6, size=N)
+    rewards = rng.normal(size=N).astype(np.float64)
+    # random episode ends, ~1/12 of steps
+    terminals = rng.random(N) < 1.0 / 12
+    t_in_ep = np.zeros(N, dtype=int)
+    t = 0
+    for i in range(N):
+        t_in_ep[i] = t
+        t = 0 if terminals[i] else t + 1
+    h, n, g = cfg.history_length, cfg.multi_step, cfg.discount
+    for i in range(N):
+        state = np.repeat(frames[i][None], h, 0).astype(np.float32) / 255.0
+        mem.append(state, int(actions[i]), float(rewards[i]),
+                   bool(terminals[i]))
+
+    def brute_stack(i):
+        """Frame stack ending at slot i, zeroed before the episode start."""
+        out = np.zeros((h, 84, 84), dtype=np.float32)
+        for j in range(h):
+            back = h - 1 - j
+            if t_in_ep[i] >= back and i - back >= 0:
+                out[j] = frames[i - back].astype(np.float32) / 255.0
+        return out
+
+    for _ in range(30):
+        idxs, s, a, r, s2, nt, w = mem.sample(16)
+        for k, i in enumerate(np.asarray(idxs)):
+            i = int(i)
+            assert a[k].item() == actions[i]
+            # n-step return with terminal cutoff
+            expect_r, done = 0.0, False
+            steps = 0
+            for sct in range(n):
+                expect_r += (g ** sct) * rewards[i + sct]
+                steps = sct + 1
+                if terminals[i + sct]:
+                    done = True
+                    break
+            assert r[k].item() == pytest.approx(expect_r, rel=1e-4, abs=1e-5)
+            assert nt[k].item() == (0.0 if done else 1.0)
+            np.testing.assert_allclose(s[k].numpy(), brute_stack(i),
+                                       rtol=0, atol=1e-6)
+            if not done:
+                np.testing.assert_allclose(s2[k].numpy(),
+                                           brute_stack(i + n),
+                                           rtol=0, atol=1e-6)
