@@ -3,6 +3,8 @@
 // (riqn_kernels.h).
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime_api.h>
+#include <vector>
 
 #include "riqn_kernels.h"
 
@@ -187,6 +189,19 @@ void adam_step_mixed(torch::Tensor master, torch::Tensor grad_bf16,
                          exp_avg_sq.data_ptr<float>(), master.numel(),
                          (float)lr, (float)beta1, (float)beta2, (float)eps,
                          step.data_ptr<int32_t>(), cur_stream());
+}
+
+// CU-masked HIP streams (colocated overlap): partition the 256 CUs between
+// the actor and learner queues so the actor's chip-filling grids cannot
+// starve the learner's small ones (AMD per-queue CU mask; wrapped on the
+// Python side with torch.cuda.ExternalStream).
+int64_t create_cu_masked_stream(std::vector<int64_t> mask_words) {
+    std::vector<uint32_t> m(mask_words.begin(), mask_words.end());
+    hipStream_t s = nullptr;
+    auto err = hipExtStreamCreateWithCUMask(&s, (uint32_t)m.size(), m.data());
+    TORCH_CHECK(err == hipSuccess, "hipExtStreamCreateWithCUMask failed: ",
+                hipGetErrorString(err));
+    return (int64_t)(uintptr_t)s;
 }
 
 void env_step(torch::Tensor actions, torch::Tensor env_state,
@@ -553,6 +568,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rand_uniform", &rand_uniform);
     m.def("noisy_reset", &noisy_reset);
     m.def("env_step", &env_step);
+    m.def("create_cu_masked_stream", &create_cu_masked_stream);
     m.def("adam_step_mixed", &adam_step_mixed);
     m.def("noisy_material", &noisy_material);
     m.def("noisy_material4", &noisy_material4);
