@@ -85,6 +85,9 @@ class Config:
     fused_backward: bool = True  # hand-written learn-step backward kernels
     fp8_heads: bool = False  # MX-fp8 NoisyLinear heads GEMM (config 5 path)
     overlap_colocated: bool = True  # actor/learner stream concurrency on 1 GPU
+    cu_split: int = 96  # colocated overlap: CUs [0,N) -> learner queue,
+    # rest -> actor queue (per-queue CU mask; 96 = 3 XCDs, measured best;
+    # 0 disables the partition)
     sharded_replay: bool = False  # replay shards live on the actor GPUs (config 5)
     backend: str = "auto"  # dist backend: auto|nccl|gloo
     learn_per_round: int = 8  # learner grad-steps per exchange round (Ape-X)
@@ -155,6 +158,7 @@ def _add_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--no-fused-backward", dest="fused_backward", action="store_false")
     p.add_argument("--fp8-heads", action="store_true")
     p.add_argument("--no-overlap", dest="overlap_colocated", action="store_false")
+    p.add_argument("--cu-split", dest="cu_split", type=int, default=96)
     p.add_argument("--sharded-replay", action="store_true")
     p.add_argument("--backend", type=str, default=d.backend, choices=["auto", "nccl", "gloo"])
     p.add_argument("--learn-per-round", type=int, default=d.learn_per_round)
