@@ -7,7 +7,6 @@ from __future__ import annotations
 import os
 from typing import List, Tuple
 
-import torch
 
 from .env import make_env
 from .utils.logging import log, plot_rewards

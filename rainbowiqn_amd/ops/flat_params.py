@@ -11,7 +11,7 @@ all of it hipGraph-capturable (device-side Adam step counter).
 
 from __future__ import annotations
 
-from typing import Iterable, List
+from typing import List
 
 import torch
 from torch import nn

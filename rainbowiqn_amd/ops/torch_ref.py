@@ -10,7 +10,6 @@ from __future__ import annotations
 from typing import Tuple
 
 import torch
-from torch.nn import functional as F
 
 
 def quantile_huber_loss(

@@ -19,7 +19,6 @@ contract; tests compare the two.
 
 from __future__ import annotations
 
-from typing import Tuple
 
 import numpy as np
 import torch
