@@ -204,6 +204,10 @@ int64_t create_cu_masked_stream(std::vector<int64_t> mask_words) {
     return (int64_t)(uintptr_t)s;
 }
 
+void destroy_stream(int64_t ptr) {
+    if (ptr) hipStreamDestroy((hipStream_t)(uintptr_t)ptr);
+}
+
 void env_step(torch::Tensor actions, torch::Tensor env_state,
               torch::Tensor rng_state, torch::Tensor ring, int64_t history,
               int64_t ring_head_new, double reward_clip, double p_done,
@@ -569,6 +573,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("noisy_reset", &noisy_reset);
     m.def("env_step", &env_step);
     m.def("create_cu_masked_stream", &create_cu_masked_stream);
+    m.def("destroy_stream", &destroy_stream);
     m.def("adam_step_mixed", &adam_step_mixed);
     m.def("noisy_material", &noisy_material);
     m.def("noisy_material4", &noisy_material4);

@@ -339,11 +339,16 @@ class ApexRunner:
                             w[cu // 32] |= 1 << (cu % 32)
                         return w
 
-                    self._lstream = torch.cuda.ExternalStream(
-                        ext.create_cu_masked_stream(_mask(0, cus)), device=d)
-                    self._astream = torch.cuda.ExternalStream(
-                        ext.create_cu_masked_stream(_mask(cus, total)),
-                        device=d)
+                    lp = ext.create_cu_masked_stream(_mask(0, cus))
+                    ap = ext.create_cu_masked_stream(_mask(cus, total))
+                    self._lstream = torch.cuda.ExternalStream(lp, device=d)
+                    self._astream = torch.cuda.ExternalStream(ap, device=d)
+                    # destroy at exit: a leaked HSA queue crashes profiler
+                    # teardown (__cxa_finalize segfault under rocprofv3)
+                    import atexit
+
+                    atexit.register(ext.destroy_stream, lp)
+                    atexit.register(ext.destroy_stream, ap)
                 except Exception:  # no CU-mask support: plain streams
                     pass
             self._staging_f = [torch.zeros(c, 84, 84, dtype=torch.uint8,
