@@ -295,3 +295,47 @@ def test_actor_shard_host_gpu():
         host.replay.update_priorities(host._idxs[s], host.pri[s])
     torch.cuda.synchronize()
     host.replay.check_invariants()
+
+
+@pytest.mark.gpu
+def test_actor_loop_fp8_heads():
+    """--fp8-heads actor routing (act_gather keeps the unfused
+    tau_embed -> quant -> MX GEMM chain): valid actions, finite state."""
+    import torch as t
+
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.models import RainbowIQN
+    from rainbowiqn_amd.runtime.gpu_actor import GpuActorLoop
+
+    cfg = Config(device="cuda", dtype="bf16", fp8_heads=True)
+    net = RainbowIQN(cfg, 6).cuda().to(t.bfloat16).train()
+    for layer in (net.fc_h_v, net.fc_h_a, net.fc_z_v, net.fc_z_a):
+        layer.enable_fused_noise()
+    loop = GpuActorLoop(cfg, net, n_envs=64, chunk_steps=8, use_graph=True)
+    assert loop.fused is not None and loop.fused.fp8_heads
+    for _ in range(2):
+        loop.run_round()
+    t.cuda.synchronize()
+    meta = loop.chunk_meta.cpu().numpy()
+    assert ((meta[:, 0] >= 0) & (meta[:, 0] < 6)).all()
+    assert loop.chunk_frames.cpu().numpy().max() > 0
+
+
+@pytest.mark.gpu
+def test_round_tracer_cuda_events(tmp_path):
+    """GPU tracer path: hipEvent spans resolve to a valid Chrome trace."""
+    import json
+
+    from rainbowiqn_amd.utils.trace import RoundTracer
+
+    tr = RoundTracer(str(tmp_path / "t.json"), use_cuda=True)
+    x = torch.rand(512, 512, device="cuda")
+    for _ in range(3):
+        with tr.span("matmul"):
+            x = x @ x * 1e-3
+    torch.cuda.synchronize()
+    ev = json.load(open(tr.dump()))["traceEvents"]
+    assert len(ev) == 3 and all(e["dur"] > 0 for e in ev)
+    # spans are ordered and non-overlapping on one stream
+    starts = [e["ts"] for e in ev]
+    assert starts == sorted(starts)
