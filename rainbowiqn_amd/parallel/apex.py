@@ -349,6 +349,8 @@ class ApexRunner:
 
                     atexit.register(ext.destroy_stream, lp)
                     atexit.register(ext.destroy_stream, ap)
+                    # unmasked stream for the pre-learn_start warmup phase
+                    self._astream_full = torch.cuda.Stream(device=d)
                 except Exception:  # no CU-mask support: plain streams
                     pass
             self._staging_f = [torch.zeros(c, 84, 84, dtype=torch.uint8,
@@ -543,9 +545,16 @@ class ApexRunner:
                 with self._sp("append"):
                     gl.append_actor_chunk(0, self._staging_f[prev],
                                           meta=self._staging_m[prev])
-        # actor stream: round + chunk stage-out, concurrent with the learner
-        self._astream.wait_stream(outer)
-        with torch.cuda.stream(self._astream):
+        # actor stream: round + chunk stage-out, concurrent with the
+        # learner.  Phase-aware partition: before learn_start the learner
+        # queue is idle, so the actor runs on the UNMASKED stream (all 256
+        # CUs) instead of its 160-CU share.
+        astream = self._astream
+        if (getattr(self, "_astream_full", None) is not None
+                and len(gl.replay) < cfg.learn_start):
+            astream = self._astream_full
+        astream.wait_stream(outer)
+        with torch.cuda.stream(astream):
             with self._sp("actor"):
                 loop.run_round()
                 self._staging_f[j].copy_(loop.chunk_frames, non_blocking=True)
@@ -563,7 +572,7 @@ class ApexRunner:
                     gl.maintenance()
         if self._lstream is not None:
             outer.wait_stream(self._lstream)
-        outer.wait_stream(self._astream)
+        outer.wait_stream(astream)
         self._round_idx += 1
 
     def _run_round_ref(self, learn: bool) -> None:
