@@ -182,6 +182,10 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
                 if cfg.memory_save_path:
                     _save_memory(runner, cfg.memory_save_path)
             if rounds % log_interval == 0:
+                if device == "cuda":
+                    # drain queued rounds so steps/s is an execution rate
+                    # (pre-learn_start rounds have no host sync point)
+                    torch.cuda.synchronize()
                 dt = time.perf_counter() - t_log
                 sps = (T - T_log) / max(dt, 1e-9)
                 loss = runner.losses[-1] if runner.losses else float("nan")
