@@ -3,11 +3,12 @@
 //
 //   1. conv_encode : Nature-CNN (4->32 8x8s4, 32->64 4x4s2, 64->64 3x3s1,
 //      ReLU after each) fused into ONE kernel.  One workgroup per state;
-//      the input frame-stack and both intermediate activation planes stay
-//      in LDS (59+30+14 KB of the 160 KB/CU); each conv layer is an
-//      implicit GEMM on v_mfma_f32_16x16x32_bf16 tiles with im2col gathers
-//      straight from LDS.  Weights ride in from L2 (re-read by all B
-//      workgroups).  Replaces MIOpen igemm + NHWC transposes + eager ReLU.
+//      conv1 gathers its im2col operand straight from L2 (optionally from
+//      the u8 frame ring, fusing the actor's frame-stack assembly) and the
+//      two intermediate activation planes stay in LDS (30+14 KB -> 2
+//      WGs/CU); each conv layer is an implicit GEMM on
+//      v_mfma_f32_16x16x32_bf16 tiles.  Weights ride in from L2 (re-read by
+//      all B workgroups).  Replaces MIOpen igemm + NHWC transposes + ReLU.
 //   2. tau_embed   : cos(pi*i*tau) basis computed in-register (no [R,64]
 //      cosine tensor in HBM), GEMM against iqn_fc.weight via MFMA, bias +
 //      ReLU + broadcast-multiply with the conv features fused in the
@@ -573,13 +574,11 @@ __global__ __launch_bounds__(256) void heads_gemm_kernel(
 // mode 1: actions[e] = argmax_i mean_{k<nq} q[e*nq+k][i]   (one WG per env)
 // mode 2: theta[r] = q[r][act[r/nq]]                        (f32 [R])
 // ---------------------------------------------------------------------------
-// 128x128-tile variant for large R (act path): 4x the MFMA per barrier and
-// half the on-chip traffic per FLOP vs the 64x64 tile.  Each of 4 waves owns
-// a 64x64 quadrant (4x4 fragments); 2 glds buffers, raw barrier + counted
-// 256x128 block tile, 512 threads (8 waves on 64x64 quadrants), BK=64,
-// 2-buffer glds (96 KB LDS, one WG/CU).  vs the 128x128 tile this moves
-// 28% fewer operand bytes -- PMC showed the 128x128 version saturating L2
-// at ~15 TB/s (878 MB/dispatch at R=8192), so tile growth is the lever.
+// Large-R (act path) tile: 256x128 block, 512 threads, each of 8 waves on
+// a 64x64 quadrant (4x4 fragments), BK=64 glds pipeline.  Tile growth is
+// the lever here: PMC showed a 128x128 version saturating L2 at ~15 TB/s
+// (878 MB/dispatch at R=8192, profiles/r01_gemm_pmc_summary.txt); this
+// shape moves 28% fewer operand bytes (917 -> 1076 TF).
 template <int NBUF>  // 2 or 3 buffers (3 = 144 KB LDS, deeper pipe)
 __global__ __launch_bounds__(512) void heads_gemm_big2_kernel(
     const unsigned short* __restrict__ hin,  // [R,3136]
