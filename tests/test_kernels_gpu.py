@@ -133,7 +133,7 @@ def test_flat_params_backward_into_flat_grad(ext, gcfg):
     # parameter mutation through the flat buffer reaches the module
     with torch.no_grad():
         flat.add_(1.0)
-    assert float(m.fc_h_v.bias_mu[0]) != 0.0
+    assert float(m.fc_h_v.bias_mu[0].detach()) != 0.0
 
 
 # ---------------------------------------------------------------------------
