@@ -7,7 +7,9 @@
 // in HBM3E and runs as three kernel families so the learner step can be
 // hipGraph-captured end to end:
 //
-//   append  : meta scan (1 thread) -> frame blit (C blocks) -> leaf updates
+//   append  : meta blit + ballot-scan episode timesteps (1 WG) -> frame
+//             blit (C blocks) -> leaf stores + level-wise ancestor range
+//             recompute (1 WG, handles ring wrap)
 //   sample  : S1 stratified tree descent + IS weights (1 block)
 //             S2 gather/assemble + fused u8->f32 preprocess (B blocks)
 //   update  : priority delta propagation with atomicExch/atomicAdd
