@@ -1,4 +1,7 @@
-"""Autograd wrappers over the HIP kernels (GPU) with torch_ref semantics."""
+"""Autograd wrappers over the HIP kernels (GPU) with torch_ref semantics
+(reference: rainbowiqn/agent.py learn() loss math, SURVEY.md section 4.1;
+used by the eager-autograd fallback path -- the fast path calls the raw
+kernels from runtime/gpu_learner.py instead)."""
 
 from __future__ import annotations
 

@@ -1,5 +1,6 @@
 """Sharded replay across GPUs (BASELINE config 5: 100M-transition PER across
-8 x 288 GB HBM3E).
+8 x 288 GB HBM3E; the reference's closest analog is Redis cluster sharding
+of the transition DB, SURVEY.md sections 3.5/6.8).
 
 The single-learner replay tops out at ~35M transitions per GPU (frames are
 ~7 KB each).  In sharded mode each ACTOR rank keeps its own HBM-resident

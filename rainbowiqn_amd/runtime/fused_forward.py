@@ -1,4 +1,6 @@
-"""Fused inference forward for RainbowIQN (MI355X fast path).
+"""Fused inference forward for RainbowIQN (MI355X fast path; the model it
+reproduces is the reference rainbowiqn/model.py RainbowIQN, SURVEY.md
+section 3.1 -- semantics parity-tested against models/model.py).
 
 Binds a bf16 RainbowIQN module to the four hand-written gfx950 kernels in
 ops/csrc/model_fwd.hip (fused 3-conv encoder, cos-tau-embed GEMM, concat
