@@ -186,3 +186,49 @@ def test_apex_sharded_replay_gloo(world):
         assert np.isfinite(a["tree_total"]) and a["tree_total"] > 0
         # weights broadcast reached everyone identically
         assert a["bcast_sum"] == pytest.approx(learner["bcast_sum"])
+
+
+def _worker_entry(rank, world, port, results, tmpdir):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.comm import Communicator
+    from rainbowiqn_amd.train_apex import train_apex
+
+    cfg = Config(
+        memory_capacity=4096, learn_start=64, batch_size=8,
+        num_tau_samples=3, num_tau_prime_samples=3, num_quantile_samples=4,
+        hidden_size=32, learn_per_round=2, device="cpu",
+        evaluation_interval=10**9, checkpoint_interval=0, T_max=10**9,
+    )
+    comm = Communicator(backend="gloo", device=torch.device("cpu"))
+    runner = train_apex(cfg, comm, n_envs=4, results_dir=tmpdir,
+                        max_rounds=5,
+                        trace_file=os.path.join(tmpdir, "trace.json"))
+    results[rank] = {
+        "is_learner": comm.is_learner,
+        "grad_steps": runner.grad_steps,
+        "model_saved": os.path.exists(os.path.join(tmpdir, "model.pth")),
+        "trace": os.path.exists(os.path.join(
+            tmpdir, f"trace.json.rank{rank}")),
+    }
+    comm.close()
+
+
+def test_train_apex_entry_gloo(tmp_path):
+    """Full distributed entry path (train_apex: runner + failsafe + tracer +
+    final checkpoint) at world 2 over gloo."""
+    world = 2
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_entry, args=(world, port, results, str(tmp_path)),
+             nprocs=world, join=True)
+    res = dict(results)
+    assert len(res) == world
+    learner = next(r for r in res.values() if r["is_learner"])
+    assert learner["grad_steps"] > 0
+    assert learner["model_saved"]
+    assert all(r["trace"] for r in res.values())
