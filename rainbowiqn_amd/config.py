@@ -158,7 +158,7 @@ def _add_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--no-fused-backward", dest="fused_backward", action="store_false")
     p.add_argument("--fp8-heads", action="store_true")
     p.add_argument("--no-overlap", dest="overlap_colocated", action="store_false")
-    p.add_argument("--cu-split", dest="cu_split", type=int, default=96)
+    p.add_argument("--cu-split", dest="cu_split", type=int, default=d.cu_split)
     p.add_argument("--sharded-replay", action="store_true")
     p.add_argument("--backend", type=str, default=d.backend, choices=["auto", "nccl", "gloo"])
     p.add_argument("--learn-per-round", type=int, default=d.learn_per_round)
