@@ -124,3 +124,30 @@ def test_nan_guard_before_first_snapshot(cfg):
     guard = NanGuard(runner)
     assert not guard.check(float("inf"))  # nothing to restore: no crash
     assert guard.recoveries == 0
+
+
+def test_metrics_exporter(tmp_path):
+    """Prometheus exporter serves the gauges when enabled and
+    no-ops cleanly when disabled."""
+    import socket
+    import urllib.request
+
+    from rainbowiqn_amd.utils.metrics import Metrics
+
+    off = Metrics(None)
+    off.update(d_env_steps=10, loss=1.0)  # disabled: no-op, no crash
+    assert not off.enabled
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    m = Metrics(port)
+    if not m.enabled:  # prometheus_client missing on this box
+        return
+    m.update(d_env_steps=123, d_grad_steps=7, loss=0.5, replay_size=42,
+             steps_per_sec=1000.0, recoveries=1)
+    body = urllib.request.urlopen(
+        f"http://127.0.0.1:{port}/metrics", timeout=10).read().decode()
+    assert "riqn_env_steps_total 123" in body
+    assert "riqn_loss 0.5" in body
+    assert "riqn_nan_recoveries_total 1" in body
