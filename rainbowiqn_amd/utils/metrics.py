@@ -15,21 +15,29 @@ class Metrics:
         if not port:
             return
         try:
-            from prometheus_client import Counter, Gauge, start_http_server
+            from prometheus_client import (CollectorRegistry, Counter, Gauge,
+                                           start_http_server)
         except ImportError:  # pragma: no cover
             return
-        start_http_server(port)
+        # private registry: repeated Metrics construction in one process
+        # (tests, notebooks) must not collide in the global REGISTRY
+        reg = CollectorRegistry()
+        start_http_server(port, registry=reg)
         self.env_steps = Counter("riqn_env_steps_total",
-                                 "agent env steps on this rank")
+                                 "agent env steps on this rank", registry=reg)
         self.grad_steps = Counter("riqn_grad_steps_total",
-                                  "learner gradient steps")
-        self.loss = Gauge("riqn_loss", "latest quantile-Huber loss")
+                                  "learner gradient steps", registry=reg)
+        self.loss = Gauge("riqn_loss", "latest quantile-Huber loss",
+                          registry=reg)
         self.replay_size = Gauge("riqn_replay_transitions",
-                                 "transitions resident in replay")
+                                 "transitions resident in replay",
+                                 registry=reg)
         self.steps_per_sec = Gauge("riqn_env_steps_per_sec",
-                                   "recent env steps/sec (this rank)")
+                                   "recent env steps/sec (this rank)",
+                                   registry=reg)
         self.recoveries = Gauge("riqn_nan_recoveries_total",
-                                "divergence rollbacks (utils/failsafe.py)")
+                                "divergence rollbacks (utils/failsafe.py)",
+                                registry=reg)
         self.enabled = True
 
     def update(self, d_env_steps: int = 0, d_grad_steps: int = 0,
