@@ -116,3 +116,33 @@ def test_param_count_matches_survey(cfg):
     m = RainbowIQN(Config(), action_space=18)
     n = sum(p.numel() for p in m.parameters())
     assert 6e6 < n < 8e6, n
+
+
+def test_fused_noise_path_matches_standard_forward(cfg):
+    """enable_fused_noise + manual weight materialization must reproduce the
+    standard NoisyLinear forward bit-for-bit (same eps), and grads must route
+    to mu/sigma identically (CPU check of the GPU fast-path contract)."""
+    import torch
+
+    from rainbowiqn_amd.models.model import NoisyLinear
+
+    torch.manual_seed(0)
+    a = NoisyLinear(16, 8)
+    b = NoisyLinear(16, 8)
+    b.load_state_dict(a.state_dict())
+    b.enable_fused_noise()
+    b.weight_epsilon.copy_(a.weight_epsilon)
+    b.bias_epsilon.copy_(a.bias_epsilon)
+    with torch.no_grad():
+        b.weight_noisy.copy_(b.weight_mu + b.weight_sigma * b.weight_epsilon)
+        b.bias_noisy.copy_(b.bias_mu + b.bias_sigma * b.bias_epsilon)
+    x = torch.randn(4, 16)
+    ya = a(x)
+    yb = b(x)
+    assert torch.allclose(ya, yb, atol=1e-6), (ya - yb).abs().max()
+    ga = torch.autograd.grad(ya.square().sum(), [a.weight_mu, a.weight_sigma,
+                                                 a.bias_mu, a.bias_sigma])
+    gb = torch.autograd.grad(yb.square().sum(), [b.weight_mu, b.weight_sigma,
+                                                 b.bias_mu, b.bias_sigma])
+    for u, v in zip(ga, gb):
+        assert torch.allclose(u, v, atol=1e-6)
