@@ -170,6 +170,12 @@ class RemoteReplay:
         # assembled step-indexed batch (static: the learn graph reads these)
         self.bst = torch.zeros(steps, self.B, h, 84, 84, dtype=dtype, device=d)
         self.bs2 = torch.zeros(steps, self.B, h, 84, 84, dtype=dtype, device=d)
+        # GpuReplay contract for the manual fused backward (gpu_learner
+        # reads replay.states_cat after sample()): one contiguous [s; s']
+        # buffer refreshed per sample() call.  Static, so graph-capturable:
+        # each unrolled learn step records its own slice copy into it.
+        self.states_cat = torch.zeros(2 * self.B, h, 84, 84, dtype=dtype,
+                                      device=d)
         self.baa = torch.zeros(steps, self.B, dtype=torch.int64, device=d)
         self.brr = torch.zeros(steps, self.B, device=d)
         self.bnt = torch.zeros(steps, self.B, device=d)
@@ -216,8 +222,11 @@ class RemoteReplay:
         s = self._step % self.S
         self._cur = s
         self._step += 1
-        return (None, self.bst[s], self.baa[s], self.brr[s], self.bs2[s],
-                self.bnt[s], self.bww[s])
+        # refresh the contiguous [s; s'] view (device copy, capture-safe)
+        self.states_cat[: self.B].copy_(self.bst[s])
+        self.states_cat[self.B:].copy_(self.bs2[s])
+        return (None, self.states_cat[: self.B], self.baa[s], self.brr[s],
+                self.states_cat[self.B:], self.bnt[s], self.bww[s])
 
     def update_priorities(self, idx, priorities) -> None:
         self.bpri[self._cur].copy_(priorities.view(-1))

@@ -194,3 +194,31 @@ def test_randomized_differential_vs_bruteforce(cfg):
                 np.testing.assert_allclose(s2[k].numpy(),
                                            brute_stack(i + n),
                                            rtol=0, atol=1e-6)
+
+
+def test_remote_replay_states_cat_contract(cfg):
+    """RemoteReplay (sharded mode learner facade) must expose the GpuReplay
+    states_cat contract the manual fused backward reads: sample() fills one
+    contiguous [s; s'] buffer and returns views into it."""
+    import torch
+
+    from rainbowiqn_amd.parallel.sharded import RemoteReplay
+
+    class _Comm:
+        n_actors = 3
+
+    cfg.batch_size = 8
+    rep = RemoteReplay(cfg, _Comm(), steps=4, device="cpu",
+                       dtype=torch.float32)
+    B = cfg.batch_size
+    rep.bst.normal_()
+    rep.bs2.normal_()
+    for s in range(6):  # wraps past S to check the cycling counter
+        _, st, a, r, s2, nt, w = rep.sample()
+        assert st.data_ptr() == rep.states_cat.data_ptr()
+        assert torch.equal(st, rep.bst[s % 4])
+        assert torch.equal(s2, rep.bs2[s % 4])
+        assert torch.equal(rep.states_cat[:B], rep.bst[s % 4])
+        assert torch.equal(rep.states_cat[B:], rep.bs2[s % 4])
+        rep.update_priorities(None, torch.full((B,), float(s)))
+        assert torch.equal(rep.bpri[s % 4], torch.full((B,), float(s)))
