@@ -29,6 +29,8 @@ DEV uint8_t synth_pixel(int x, int y, int t, int cy, int cx, uint8_t noise) {
     return v | noise;
 }
 
+__global__ void env_bump_kernel(uint64_t* rng_state) { rng_state[1] += 1; }
+
 __global__ void env_step_kernel(
     const int64_t* __restrict__ actions,  // [E]
     int32_t* __restrict__ env_state,      // [E,4]
@@ -100,7 +102,9 @@ __global__ void env_step_kernel(
         uint8_t noise = (uint8_t)(h & 31);
         dst[i] = synth_pixel(x, y, t, cy, cx, noise);
     }
-    if (threadIdx.x == 0 && e == 0) rng_state[1] += 1;
+    // NOTE: offset bump rides a trailing 1-thread kernel (launcher):
+    // bumping from block 0 here would race late blocks' offset reads when
+    // the grid exceeds resident occupancy.
 }
 
 template <int OUT_BF16>
@@ -160,6 +164,8 @@ void riqn_env_step(const int64_t* actions, int32_t* env_state,
                        ring, history, ring_head_new, n_envs, reward_clip,
                        p_done, max_episode_frames, chunk_frames, chunk_meta,
                        chunk_pos);
+    hipLaunchKernelGGL(env_bump_kernel, dim3(1), dim3(1), 0,
+                       (hipStream_t)stream, rng_state);
 }
 
 void riqn_stack_gather(const uint8_t* ring, const int32_t* env_state,

@@ -254,6 +254,9 @@ __global__ void sample_indices_kernel(
     if (threadIdx.x == 0) lds[0] = wmax;
     __syncthreads();
     if (b < batch) out_weights[b] = w / lds[0];
+    // single-block kernel (grid=1): every thread's offset read happened
+    // before the block_reduce barriers above, so this bump cannot race a
+    // late block's read (contrast rng.hip's multi-block discipline)
     if (threadIdx.x == 0) rng_state[1] += 1;  // advance philox offset
 }
 

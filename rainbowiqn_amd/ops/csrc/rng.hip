@@ -6,6 +6,14 @@
 
 namespace {
 
+// Philox-offset advance discipline: a kernel that READS rng_state[1] must
+// never bump it from one of its own blocks -- with more blocks than resident
+// occupancy a late block can observe the bumped value and duplicate the next
+// launch's counter stream.  The bump therefore always rides
+// a LATER kernel in stream order: the second kernel of a fill pair, or this
+// dedicated 1-thread kernel appended by the launcher.
+__global__ void rng_bump_kernel(uint64_t* rng_state) { rng_state[1] += 1; }
+
 __global__ void rand_uniform_kernel(float* __restrict__ out, int64_t n,
                                     uint64_t* rng_state) {
     uint64_t seed = rng_state[0], off = rng_state[1];
@@ -17,7 +25,6 @@ __global__ void rand_uniform_kernel(float* __restrict__ out, int64_t n,
     #pragma unroll
     for (int i = 0; i < 4; ++i)
         if (base + i < n) out[base + i] = u[i];
-    if (tid == 0) rng_state[1] = off + 1;
 }
 
 // f(x) = sign(x) * sqrt(|x|) of a standard normal
@@ -43,7 +50,6 @@ __global__ void noisy_eps_vec_kernel(float* __restrict__ eps_in, int64_t n_in,
         if (j + 1 < n_out) eps_out[j + 1] = scale_noise(z1);
     }
     (void)buf;
-    if (tid == 0) rng_state[1] = off + 1;
 }
 
 template <int OUT_BF16>
@@ -52,7 +58,11 @@ __global__ void noisy_eps_outer_kernel(const float* __restrict__ eps_in,
                                        const float* __restrict__ eps_out,
                                        int64_t n_out,
                                        float* __restrict__ weight_eps,
-                                       float* __restrict__ bias_eps) {
+                                       float* __restrict__ bias_eps,
+                                       uint64_t* rng_state) {
+    // advances the offset CONSUMED by the preceding eps-vec kernel (stream
+    // order makes this race-free; see rng_bump_kernel comment)
+    if (threadIdx.x == 0 && blockIdx.x == 0) rng_state[1] += 1;
     for (int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
          i < n_in * n_out; i += (int64_t)gridDim.x * blockDim.x) {
         float v = eps_out[i / n_in] * eps_in[i % n_in];
@@ -80,7 +90,9 @@ __global__ void noisy_material_kernel(
     const void* __restrict__ w_mu, const void* __restrict__ w_sigma,
     const void* __restrict__ b_mu, const void* __restrict__ b_sigma,
     void* __restrict__ weight_eps, void* __restrict__ bias_eps,
-    void* __restrict__ w_noisy, void* __restrict__ b_noisy) {
+    void* __restrict__ w_noisy, void* __restrict__ b_noisy,
+    uint64_t* rng_state) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) rng_state[1] += 1;
     for (int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
          i < n_in * n_out; i += (int64_t)gridDim.x * blockDim.x) {
         float e = eps_out[i / n_in] * eps_in[i % n_in];
@@ -144,11 +156,11 @@ __global__ void noisy_eps_vec4_kernel(NoisyPack4 p, uint64_t* rng_state) {
         }
         i -= nout;
     }
-    if (tid == 0) rng_state[1] = off + 1;
 }
 
 template <int BF16>
-__global__ void noisy_material4_kernel(NoisyPack4 p) {
+__global__ void noisy_material4_kernel(NoisyPack4 p, uint64_t* rng_state) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) rng_state[1] += 1;
     int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     // weight segments
@@ -203,10 +215,10 @@ void riqn_noisy_material4(NoisyPack4 pack, int bf16, uint64_t* rng_state,
                        pack, rng_state);
     if (bf16)
         hipLaunchKernelGGL(noisy_material4_kernel<1>, dim3(2048), dim3(256),
-                           0, s, pack);
+                           0, s, pack, rng_state);
     else
         hipLaunchKernelGGL(noisy_material4_kernel<0>, dim3(2048), dim3(256),
-                           0, s, pack);
+                           0, s, pack, rng_state);
 }
 
 void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state,
@@ -216,6 +228,7 @@ void riqn_rand_uniform(float* out, int64_t n, uint64_t* rng_state,
     int blocks = (int)min((threads + 255) / 256, (int64_t)2048);
     hipLaunchKernelGGL(rand_uniform_kernel, dim3(max(blocks, 1)), dim3(256), 0,
                        s, out, n, rng_state);
+    hipLaunchKernelGGL(rng_bump_kernel, dim3(1), dim3(1), 0, s, rng_state);
 }
 
 void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
@@ -231,11 +244,11 @@ void riqn_noisy_reset(float* eps_in, int64_t n_in, float* eps_out,
     if (eps_bf16)
         hipLaunchKernelGGL(noisy_eps_outer_kernel<1>, dim3(blocks2), dim3(256),
                            0, s, eps_in, n_in, eps_out, n_out, weight_eps,
-                           bias_eps);
+                           bias_eps, rng_state);
     else
         hipLaunchKernelGGL(noisy_eps_outer_kernel<0>, dim3(blocks2), dim3(256),
                            0, s, eps_in, n_in, eps_out, n_out, weight_eps,
-                           bias_eps);
+                           bias_eps, rng_state);
 }
 
 void riqn_noisy_material(float* eps_in, int64_t n_in, float* eps_out,
@@ -255,12 +268,12 @@ void riqn_noisy_material(float* eps_in, int64_t n_in, float* eps_out,
         hipLaunchKernelGGL(noisy_material_kernel<1>, dim3(blocks2), dim3(256),
                            0, s, eps_in, n_in, eps_out, n_out, w_mu, w_sigma,
                            b_mu, b_sigma, weight_eps, bias_eps, w_noisy,
-                           b_noisy);
+                           b_noisy, rng_state);
     else
         hipLaunchKernelGGL(noisy_material_kernel<0>, dim3(blocks2), dim3(256),
                            0, s, eps_in, n_in, eps_out, n_out, w_mu, w_sigma,
                            b_mu, b_sigma, weight_eps, bias_eps, w_noisy,
-                           b_noisy);
+                           b_noisy, rng_state);
 }
 
 }  // extern "C"

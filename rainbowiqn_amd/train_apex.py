@@ -150,7 +150,12 @@ def train_apex(cfg: Config, comm: Optional[Communicator] = None,
     shutdown = GracefulShutdown()
     guard = NanGuard(runner) if comm.is_learner else None
 
-    steps_per_round = runner.chunk_steps * runner.n_envs * comm.n_actors
+    # env steps generated per round across ALL actor loops: n_actors ranks
+    # x n_local_actors loops per rank (--nb-actor, BASELINE config 3).
+    # Undercounting here would stretch beta annealing and the eval/target/
+    # checkpoint cadences by that factor.
+    steps_per_round = (runner.chunk_steps * runner.n_envs * comm.n_actors
+                       * runner.n_local_actors)
     T, rounds = 0, 0
     next_eval = cfg.evaluation_interval
     next_ckpt = cfg.checkpoint_interval or float("inf")
