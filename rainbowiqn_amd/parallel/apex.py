@@ -267,24 +267,37 @@ class ApexRunner:
         if comm.is_actor:
             dt = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
             self._actor_flats = []
+            # Colocated overlap (world 1, M 1, overlap_colocated on): actor
+            # and learner rounds run CONCURRENTLY on separate streams, so the
+            # actor must NOT read the learner's live net -- Adam/pack/noise
+            # writes on the learner stream would tear mid-forward.  The actor
+            # gets its own net (own noise, own packs); _run_round_overlap
+            # copies flat weights at the round boundary on the outer stream,
+            # making the actor's view a consistent one-round-stale snapshot
+            # (the same bounded-staleness contract as the multi-GPU split).
+            # Only the strictly sequential non-overlap colocated mode shares
+            # the online net directly.
+            seq_colocated = (comm.world_size == 1 and M == 1
+                             and not getattr(cfg, "overlap_colocated", True))
             for j in range(M):
-                if comm.world_size == 1 and M == 1:
-                    net = self.glearner.online  # colocated: share the online net
+                if seq_colocated:
+                    net = self.glearner.online  # sequential: safe to share
                     flat = None
                 else:
                     # independent actor: own net copy (own NoisyNet noise),
                     # refreshed from the learner's flat weights each round
                     net = RainbowIQN(cfg, n_act).to(dev).to(dt)
                     flat = flatten_module(net)
+                    if comm.world_size == 1:
+                        flat.copy_(self.glearner.flat_online)
                     if comm.world_size > 1 and j == 0:
                         self.flat_weights = flat
                 self._actor_flats.append(flat)
-                shared = comm.world_size == 1 and M == 1
                 self.gactors.append(
                     GpuActorLoop(cfg, net, self.n_envs, self.chunk_steps,
                                  device=dev, use_graph=cfg.hipgraph,
                                  seed_offset=comm.rank * 7717 + j * 131071,
-                                 external_weights=shared))
+                                 external_weights=seq_colocated))
             self.gactor = self.gactors[0]
             self.actor = _ActorSteps(self.gactors)
             if self.sharded:
@@ -530,6 +543,11 @@ class ApexRunner:
         j = self._round_idx % 2
         prev = (self._round_idx - 1) % 2
         outer = torch.cuda.current_stream()
+        # round-boundary weight snapshot: the previous round's learner work
+        # is complete w.r.t. the outer stream here, so the actor's flat copy
+        # sees consistent (one-round-stale) weights -- never torn mid-step
+        if self._actor_flats[0] is not None:
+            self._actor_flats[0].copy_(gl.flat_online)
         # learner work optionally rides a high-priority queue (RIQN_LPRIO):
         # its small grids then dispatch ahead of the actor's queued WGs
         from contextlib import nullcontext
