@@ -34,7 +34,18 @@ class Communicator:
             if backend == "auto":
                 backend = "nccl" if torch.cuda.is_available() else "gloo"
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-            os.environ.setdefault("MASTER_PORT", "29571")
+            # torchrun sets MASTER_PORT; this fallback covers hand-rolled
+            # RANK/WORLD_SIZE launches.  Derive the port from job-stable
+            # state (cwd + optional RIQN_JOB_ID) so two concurrent jobs on
+            # one node don't collide on a hardcoded port -- every rank of
+            # one job computes the same value.
+            if "MASTER_PORT" not in os.environ:
+                import zlib
+
+                tag = os.environ.get("RIQN_JOB_ID", "") + os.getcwd() + \
+                    os.environ["WORLD_SIZE"]
+                port = 29500 + zlib.crc32(tag.encode()) % 2000
+                os.environ["MASTER_PORT"] = str(port)
             dist.init_process_group(
                 backend=backend, timeout=datetime.timedelta(seconds=300)
             )

@@ -54,8 +54,9 @@ def _worker_apex(rank, world, port, results):
     comm.close()
 
 
-@pytest.mark.parametrize("world", [3])
+@pytest.mark.parametrize("world", [3, 8])
 def test_apex_round_gloo(world):
+    """world 8 = the target 7+1 topology (BASELINE config 4) on gloo."""
     port = _find_free_port()
     mgr = mp.Manager()
     results = mgr.dict()
@@ -64,9 +65,10 @@ def test_apex_round_gloo(world):
     assert len(res) == world
     learner = next(r for r in res.values() if r["is_learner"])
     actors = [r for r in res.values() if not r["is_learner"]]
-    # 2 actors x 6 rounds x 4 envs x 8 steps = 384 transitions
-    assert learner["replay_len"] == 384
+    # (world-1) actors x 6 rounds x 4 envs x 8 steps
+    assert learner["replay_len"] == (world - 1) * 6 * 4 * 8
     assert all(l == 192 for l in learner["ring_lens"])
+    assert len(learner["ring_lens"]) == world - 1
     assert learner["grad_steps"] > 0
     assert learner["losses_finite"]
     # weight flow: every rank ends with identical broadcast weights
@@ -166,7 +168,7 @@ def _worker_sharded(rank, world, port, results):
     comm.close()
 
 
-@pytest.mark.parametrize("world", [3])
+@pytest.mark.parametrize("world", [3, 8])
 def test_apex_sharded_replay_gloo(world):
     """Config-5 flow on CPU/gloo: actor-resident replay shards, P2P batch
     service to the learner, priorities returned to the owning shards."""
@@ -232,3 +234,101 @@ def test_train_apex_entry_gloo(tmp_path):
     assert learner["grad_steps"] > 0
     assert learner["model_saved"]
     assert all(r["trace"] for r in res.values())
+
+
+def _worker_tags(rank, world, port, results):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    comm = Communicator(backend="gloo", device=torch.device("cpu"))
+    out = {}
+    # two back-to-back batched-P2P gathers of DIFFERENT shapes per round,
+    # several rounds: exactly the frames+meta pattern of the Ape-X round.
+    # Payloads are (round, rank)-stamped so any tag/order mix-up between the
+    # two gathers or between rounds is detected by content, deterministically.
+    for rnd in range(4):
+        f = torch.full((16, 7), float(rank * 1000 + rnd))
+        m = torch.full((5,), float(-(rank * 1000 + rnd)))
+        fl = comm.gather_chunks(f)
+        ml = comm.gather_chunks(m)
+        if comm.is_learner:
+            for a in range(comm.n_actors):
+                src_rank = a + 1
+                assert fl[a].shape == (16, 7)
+                assert (fl[a] == src_rank * 1000 + rnd).all(), (rnd, a)
+                assert (ml[a] == -(src_rank * 1000 + rnd)).all(), (rnd, a)
+    out["ok"] = True
+    results[rank] = out
+    comm.close()
+
+
+def test_gather_tag_ordering_world8():
+    """P2P tag ordering of the chunk gather at the target world size: frames
+    and meta land in the right per-actor slots across consecutive rounds
+    (the batched_isend_irecv ordering risk flagged in round 1)."""
+    world = 8
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_tags, args=(world, port, results), nprocs=world,
+             join=True)
+    assert all(r["ok"] for r in dict(results).values())
+
+
+def _worker_ring_interleave(rank, world, port, results):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    torch.manual_seed(rank)
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    cfg = Config(
+        memory_capacity=8192, learn_start=10**9, batch_size=8,
+        num_tau_samples=3, num_tau_prime_samples=3, num_quantile_samples=4,
+        hidden_size=32, learn_per_round=2, device="cpu",
+    )
+    comm = Communicator(backend="gloo", device=torch.device("cpu"))
+    runner = ApexRunner(cfg, comm, n_envs=2, chunk_steps=8)
+    # stamp each actor's generated rewards with its rank so the learner can
+    # verify which ring each actor's chunk landed in
+    if comm.is_actor:
+        orig = runner.actor.run_chunk
+
+        def stamped():
+            orig()
+            runner.actor.chunk_meta[:, 1] = float(comm.rank)
+
+        runner.actor.run_chunk = stamped
+    for _ in range(3):
+        runner.run_round(learn=False)
+    out = {"rank": rank}
+    if comm.is_learner:
+        ring_rewards = []
+        for ring in runner.replay.rings:
+            vals = set(float(v) for v in ring.rewards[: len(ring)])
+            ring_rewards.append(vals)
+        out["ring_rewards"] = ring_rewards
+    results[rank] = out
+    comm.close()
+
+
+def test_ring_id_assignment_world8():
+    """Each actor rank's chunks must land in ITS OWN learner replay ring
+    (ring id = actor index), never mixed -- the frame-stack/n-step
+    reconstruction correctness depends on per-actor stream contiguity."""
+    world = 8
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_ring_interleave, args=(world, port, results),
+             nprocs=world, join=True)
+    res = dict(results)
+    learner = res[0]
+    for a, vals in enumerate(learner["ring_rewards"]):
+        assert vals == {float(a + 1)}, (a, vals)
