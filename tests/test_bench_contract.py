@@ -28,3 +28,17 @@ def test_bench_json_contract():
     cfgd = d["config"]
     for k in ("model", "global_batch", "seq_len", "parallelism"):
         assert k in cfgd, k
+
+
+def test_bench_learner_only_contract():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "0",
+         "--learn-per-round", "1", "--device", "cpu", "--learner-only"],
+        capture_output=True, text=True, timeout=600, check=True,
+    )
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["unit"] == "grad-steps/s" and d["value"] > 0
+    assert d["env_steps_per_sec"] == 0.0
+    assert "learner-only" in d["config"]["parallelism"]
