@@ -31,7 +31,12 @@ _GAME_ACTIONS = {
     "space_invaders": 6,
     "seaquest": 18,
     "qbert": 6,
+    # learnable synthetic task (reward a deterministic function of the
+    # frame): 4 cue slots -> 4 actions.  See CueEnv below.
+    "cue": 4,
 }
+
+CUE_SLOTS = 4  # cue task: frame shows a bright block in one of these slots
 
 
 def n_actions_for(game: str) -> int:
@@ -133,6 +138,43 @@ class SyntheticAtariEnv:
         reward = 0.0
         if u < 0.02:
             reward = 1.0 if (action + self._t) % 3 else -1.0
+        done = self.rng.random() < (4.0 / self.mean_episode_len)
+        if self._t >= self.max_episode_length:
+            done = True
+        return state, reward, done
+
+
+class CueEnv(SyntheticAtariEnv):
+    """Learnable synthetic task: the newest frame
+    shows a bright 20x20 block in one of CUE_SLOTS horizontal slots on a dim
+    background; acting with the matching slot index yields +1, anything else
+    -1.  Reward is a deterministic function of the frame, so a correct
+    learner's average reward climbs from ~(2/A - 1) (random, -0.5 at A=4)
+    toward +1 -- the end-to-end learning signal the random-reward synthetic
+    env cannot provide.  Same Env interface as the reference wrapper.
+    """
+
+    def __init__(self, cfg, device="cpu", mean_episode_len: int = 512):
+        super().__init__(cfg, device, mean_episode_len)
+        self.n_actions = CUE_SLOTS
+        self._cue = int(self.rng.integers(0, CUE_SLOTS))
+
+    def _raw_frame(self) -> np.ndarray:
+        t = int(self._t + self._phase) & 0xFFFF
+        y = np.arange(FRAME_H, dtype=np.int64)[:, None]
+        x = np.arange(FRAME_W, dtype=np.int64)[None, :]
+        img = ((x * 3 + y * 2 + t * 5) & 0x3F).astype(np.uint8)  # dim bg
+        x0 = self._cue * (FRAME_W // CUE_SLOTS) + 1
+        img[32:52, x0 : x0 + 19] = 255
+        noise = self.rng.integers(0, 32, size=(FRAME_H, FRAME_W), dtype=np.uint8)
+        return img | noise
+
+    def step(self, action: int) -> Tuple[torch.Tensor, float, bool]:
+        assert 0 <= action < self.n_actions
+        reward = 1.0 if action == self._cue else -1.0
+        self._cue = int(self.rng.integers(0, CUE_SLOTS))
+        self._t += 4
+        state = self._observe()  # renders the NEW cue
         done = self.rng.random() < (4.0 / self.mean_episode_len)
         if self._t >= self.max_episode_length:
             done = True

@@ -185,10 +185,11 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
 
 
 @pytest.mark.gpu
-def test_colocated_shares_fused_forward():
-    """Colocated actor+learner share ONE FusedForward on the online net --
-    two instances would leave the learner's GEMM weight buffers stale after
-    noise resets (regression test)."""
+def test_colocated_overlap_double_buffered_weights():
+    """Colocated OVERLAP mode: actor and learner run concurrently on
+    separate streams, so the actor must have its OWN net/FusedForward packs
+    (a one-round-stale snapshot copied at round boundaries) -- sharing the
+    learner's live buffers would tear mid-forward."""
     from rainbowiqn_amd.config import Config
     from rainbowiqn_amd.parallel.apex import ApexRunner
     from rainbowiqn_amd.parallel.comm import Communicator
@@ -197,13 +198,42 @@ def test_colocated_shares_fused_forward():
                  memory_capacity=8192, learn_per_round=2)
     runner = ApexRunner(cfg, Communicator(backend="auto"), n_envs=16,
                         chunk_steps=8)
-    assert runner.gactor.fused is runner.glearner.fused_online
+    assert runner.overlap
+    assert runner.gactor.fused is not runner.glearner.fused_online
+    assert runner.gactor.net is not runner.glearner.online
+    assert runner._actor_flats[0] is not None
     wcat_before = runner.glearner.fused_online.wcat.clone()
     for _ in range(6):
         runner.run_round()
     torch.cuda.synchronize()
     # noise regenerated + Adam steps: the learner's GEMM operand moved
     assert not torch.equal(wcat_before, runner.glearner.fused_online.wcat)
+    assert runner.grad_steps > 0
+    # bounded staleness: after a final boundary copy the actor's flat
+    # weights equal the learner's current flat weights
+    runner.run_round(learn=False)
+    torch.cuda.synchronize()
+    assert torch.equal(runner._actor_flats[0], runner.glearner.flat_online)
+
+
+@pytest.mark.gpu
+def test_colocated_sequential_shares_fused_forward():
+    """With overlap OFF the colocated round is strictly sequential and the
+    actor shares the learner's online net + FusedForward (no copy cost)."""
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    cfg = Config(device="cuda", dtype="bf16", learn_start=512,
+                 memory_capacity=8192, learn_per_round=2,
+                 overlap_colocated=False)
+    runner = ApexRunner(cfg, Communicator(backend="auto"), n_envs=16,
+                        chunk_steps=8)
+    assert not runner.overlap
+    assert runner.gactor.fused is runner.glearner.fused_online
+    for _ in range(6):
+        runner.run_round()
+    torch.cuda.synchronize()
     assert runner.grad_steps > 0
 
 
