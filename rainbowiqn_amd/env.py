@@ -239,8 +239,49 @@ class VectorSyntheticEnv:
         return frames, rewards, dones
 
 
+class VectorCueEnv(VectorSyntheticEnv):
+    """Batched CueEnv: E independent streams of the learnable cue task."""
+
+    def __init__(self, cfg, n_envs: int, seed_offset: int = 0):
+        super().__init__(cfg, n_envs, seed_offset)
+        self.n_actions = CUE_SLOTS
+        self._cue = self.rng.integers(0, CUE_SLOTS, size=n_envs)
+
+    def frames(self) -> np.ndarray:
+        t = ((self._t + self._phase) & 0xFFFF)[:, None, None]
+        img = ((self._x * 3 + self._y * 2 + t * 5) & 0x3F).astype(np.uint8)
+        w = FRAME_W // CUE_SLOTS
+        for e in range(self.n_envs):
+            x0 = int(self._cue[e]) * w + 1
+            img[e, 32:52, x0 : x0 + 19] = 255
+        noise = self.rng.integers(0, 32, size=img.shape, dtype=np.uint8)
+        return img | noise
+
+    def step(self, actions: np.ndarray):
+        rewards = np.where(actions == self._cue, 1.0, -1.0).astype(np.float32)
+        self._cue = self.rng.integers(0, CUE_SLOTS, size=self.n_envs)
+        self._t += 4
+        dones = self.rng.random(self.n_envs) < self.p_done
+        dones |= self._t >= self.max_episode_length
+        frames = self.frames()
+        if dones.any():
+            self._t[dones] = 0
+            self._phase[dones] = self.rng.integers(0, 1 << 16, size=int(dones.sum()))
+        return frames, rewards, dones
+
+
+def make_vector_env(cfg, n_envs: int, seed_offset: int = 0):
+    """Vector env for the actor loop: the learnable cue task when
+    cfg.game == 'cue', else the throughput-oriented random synthetic env."""
+    if cfg.game.lower() == "cue":
+        return VectorCueEnv(cfg, n_envs, seed_offset)
+    return VectorSyntheticEnv(cfg, n_envs, seed_offset)
+
+
 def make_env(cfg, device: str | torch.device = "cpu"):
     """Return the real ALE wrapper when available, else the synthetic env."""
+    if cfg.game.lower() == "cue":  # learnable synthetic task, never ALE
+        return CueEnv(cfg, device)
     try:  # pragma: no cover - no ALE wheel on this box
         try:
             import atari_py  # noqa: F401

@@ -212,7 +212,7 @@ void env_step(torch::Tensor actions, torch::Tensor env_state,
               torch::Tensor rng_state, torch::Tensor ring, int64_t history,
               int64_t ring_head_new, double reward_clip, double p_done,
               int64_t max_episode_frames, torch::Tensor chunk_frames,
-              torch::Tensor chunk_meta, int64_t chunk_pos) {
+              torch::Tensor chunk_meta, int64_t chunk_pos, int64_t task) {
     CHECK_T(actions, torch::kInt64);
     CHECK_T(ring, torch::kUInt8);
     int n_envs = actions.size(0);
@@ -221,7 +221,8 @@ void env_step(torch::Tensor actions, torch::Tensor env_state,
                   ring.data_ptr<uint8_t>(), (int)history, (int)ring_head_new,
                   n_envs, (float)reward_clip, (float)p_done,
                   (int)max_episode_frames, chunk_frames.data_ptr<uint8_t>(),
-                  chunk_meta.data_ptr<float>(), (int)chunk_pos, cur_stream());
+                  chunk_meta.data_ptr<float>(), (int)chunk_pos, (int)task,
+                  cur_stream());
 }
 
 void stack_gather(torch::Tensor ring, torch::Tensor env_state, int64_t history,
@@ -571,7 +572,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("preprocess", &preprocess);
     m.def("rand_uniform", &rand_uniform);
     m.def("noisy_reset", &noisy_reset);
-    m.def("env_step", &env_step);
+    m.def("env_step", &env_step, py::arg("actions"), py::arg("env_state"),
+          py::arg("rng_state"), py::arg("ring"), py::arg("history"),
+          py::arg("ring_head_new"), py::arg("reward_clip"), py::arg("p_done"),
+          py::arg("max_episode_frames"), py::arg("chunk_frames"),
+          py::arg("chunk_meta"), py::arg("chunk_pos"), py::arg("task") = 0);
     m.def("create_cu_masked_stream", &create_cu_masked_stream);
     m.def("destroy_stream", &destroy_stream);
     m.def("adam_step_mixed", &adam_step_mixed);

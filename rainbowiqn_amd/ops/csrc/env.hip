@@ -29,6 +29,20 @@ DEV uint8_t synth_pixel(int x, int y, int t, int cy, int cx, uint8_t noise) {
     return v | noise;
 }
 
+// Learnable cue task (task=1; env.py::CueEnv): dim background + bright
+// 20x20 block in one of 4 horizontal slots; the matching action pays +1,
+// everything else -1.  Reward is a deterministic function of the frame, so
+// the full fused stack can DEMONSTRABLY learn (policy accuracy -> 1), which
+// the random-reward throughput env cannot show.
+#define CUE_SLOTS 4
+
+DEV uint8_t cue_pixel(int x, int y, int t, int cue, uint8_t noise) {
+    uint8_t v = (uint8_t)((x * 3 + y * 2 + t * 5) & 0x3F);
+    int x0 = cue * (FRAME_W / CUE_SLOTS) + 1;
+    if (y >= 32 && y < 52 && x >= x0 && x < x0 + 19) v = 255;
+    return v | noise;
+}
+
 __global__ void env_bump_kernel(uint64_t* rng_state) { rng_state[1] += 1; }
 
 __global__ void env_step_kernel(
@@ -39,9 +53,9 @@ __global__ void env_step_kernel(
     int n_envs, float reward_clip, float p_done, int max_episode_frames,
     uint8_t* __restrict__ chunk_frames,  // [C][84*84], entry chunk_pos*E+e
     float* __restrict__ chunk_meta,      // [C][3]
-    int chunk_pos) {
+    int chunk_pos, int task) {
     int e = blockIdx.x;
-    __shared__ int t_s, phase_s, cy_s, cx_s;
+    __shared__ int t_s, phase_s, cy_s, cx_s, cue_s;
     __shared__ uint32_t nbase_s;
     if (threadIdx.x == 0) {
         int t = env_state[4 * e + 0];
@@ -54,7 +68,17 @@ __global__ void env_step_kernel(
         int t_next = t + 4;
         int a = (int)actions[e];
         float reward = 0.0f;
-        if (u01 < 0.02f) reward = ((a + t_next) % 3 != 0) ? 1.0f : -1.0f;
+        if (task == 1) {
+            // cue task: pay off against the cue VISIBLE in the acted-on
+            // state, then redraw it for the next frame
+            int cue_old = env_state[4 * e + 3];
+            reward = (a == cue_old) ? 1.0f : -1.0f;
+            int cue_new = (int)(u01 * CUE_SLOTS) & (CUE_SLOTS - 1);
+            env_state[4 * e + 3] = cue_new;
+            cue_s = cue_new;
+        } else {
+            if (u01 < 0.02f) reward = ((a + t_next) % 3 != 0) ? 1.0f : -1.0f;
+        }
         if (reward_clip > 0.0f)
             reward = fmaxf(-reward_clip, fminf(reward_clip, reward));
         bool done = (ud < p_done) || (t_next >= max_episode_frames);
@@ -100,7 +124,8 @@ __global__ void env_step_kernel(
         uint32_t h = (nbase_s ^ (uint32_t)(i * 2654435761u));
         h ^= h >> 15;
         uint8_t noise = (uint8_t)(h & 31);
-        dst[i] = synth_pixel(x, y, t, cy, cx, noise);
+        dst[i] = task == 1 ? cue_pixel(x, y, t, cue_s, noise)
+                           : synth_pixel(x, y, t, cy, cx, noise);
     }
     // NOTE: offset bump rides a trailing 1-thread kernel (launcher):
     // bumping from block 0 here would race late blocks' offset reads when
@@ -158,12 +183,12 @@ void riqn_env_step(const int64_t* actions, int32_t* env_state,
                    uint64_t* rng_state, uint8_t* ring, int history,
                    int ring_head_new, int n_envs, float reward_clip,
                    float p_done, int max_episode_frames, uint8_t* chunk_frames,
-                   float* chunk_meta, int chunk_pos, void* stream) {
+                   float* chunk_meta, int chunk_pos, int task, void* stream) {
     hipLaunchKernelGGL(env_step_kernel, dim3(n_envs), dim3(256), 0,
                        (hipStream_t)stream, actions, env_state, rng_state,
                        ring, history, ring_head_new, n_envs, reward_clip,
                        p_done, max_episode_frames, chunk_frames, chunk_meta,
-                       chunk_pos);
+                       chunk_pos, task);
     hipLaunchKernelGGL(env_bump_kernel, dim3(1), dim3(1), 0,
                        (hipStream_t)stream, rng_state);
 }

@@ -167,7 +167,7 @@ void riqn_env_step(const int64_t* actions, int32_t* env_state,
                    uint64_t* rng_state, uint8_t* ring, int history,
                    int ring_head_new, int n_envs, float reward_clip,
                    float p_done, int max_episode_frames, uint8_t* chunk_frames,
-                   float* chunk_meta, int chunk_pos, void* stream);
+                   float* chunk_meta, int chunk_pos, int task, void* stream);
 void riqn_stack_gather(const uint8_t* ring, const int32_t* env_state,
                        int history, int ring_head, int n_envs, float* out,
                        int out_bf16, void* stream);

@@ -29,7 +29,7 @@ import numpy as np
 import torch
 
 from ..agent import Agent
-from ..env import VectorSyntheticEnv
+from ..env import make_vector_env
 from ..replay import ReplayMemory
 from .comm import Communicator
 
@@ -108,7 +108,7 @@ class ActorState:
         self.agent = agent
         self.n_envs = n_envs
         self.chunk_steps = chunk_steps
-        self.env = VectorSyntheticEnv(cfg, n_envs, seed_offset=comm.rank * 1009)
+        self.env = make_vector_env(cfg, n_envs, seed_offset=comm.rank * 1009)
         self.device = agent.device
         h = cfg.history_length
         self.stack = torch.zeros(n_envs, h, 84, 84, device=self.device)

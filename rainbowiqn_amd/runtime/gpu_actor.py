@@ -48,6 +48,13 @@ class GpuActorLoop:
         e = n_envs
         self.ring = torch.zeros(self.h, e, 84, 84, dtype=torch.uint8, device=d)
         self.env_state = torch.zeros(e, 4, dtype=torch.int32, device=d)
+        # device env task: 1 = learnable cue task (env.py::CueEnv semantics
+        # in the env_step kernel), 0 = random-reward throughput env
+        self.task = 1 if cfg.game.lower() == "cue" else 0
+        if self.task == 1:
+            g = torch.Generator().manual_seed(cfg.seed + seed_offset)
+            self.env_state[:, 3] = torch.randint(0, 4, (e,), generator=g,
+                                                 dtype=torch.int32)
         self.rng_state = torch.tensor(
             [cfg.seed * 48271 + 101 + seed_offset, 0], dtype=torch.int64, device=d
         )
@@ -103,6 +110,7 @@ class GpuActorLoop:
                 cfg.reward_clip if cfg.reward_clip_enabled else 0.0,
                 self.p_done, cfg.max_episode_length,
                 self.chunk_frames.view(-1, 84, 84), self.chunk_meta, s,
+                self.task,
             )
 
     def reset_noise(self) -> None:

@@ -1,0 +1,188 @@
+"""End-to-end LEARNING demonstrations.
+
+The random-reward synthetic env can only prove plumbing; these tests use the
+learnable cue task (env.py::CueEnv -- reward a deterministic function of the
+frame) and assert the full system actually LEARNS: loss falls and the greedy
+policy's accuracy rises far above chance, through the same act/replay/learn
+stack the benchmark runs.  This is the rebuild's stand-in for the reference's
+learning-curve correctness culture.
+"""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from rainbowiqn_amd.config import Config
+
+
+def _small_cfg(**kw):
+    base = dict(game="cue", seed=7, num_tau_samples=3, num_tau_prime_samples=3,
+                num_quantile_samples=8, multi_step=1, lr=1e-3, batch_size=16,
+                memory_capacity=4096, learn_start=256, device="cpu",
+                dtype="fp32", hidden_size=64, evaluation_interval=10**9)
+    base.update(kw)
+    return Config(**base)
+
+
+def _greedy_accuracy(agent, cfg, n=100) -> float:
+    from rainbowiqn_amd.env import CueEnv
+
+    env = CueEnv(cfg)
+    env.eval()
+    state = env.reset()
+    correct = 0
+    for _ in range(n):
+        cue = env._cue
+        a = agent.act(state)
+        correct += int(a == cue)
+        state, _, done = env.step(a)
+        if done:
+            state = env.reset()
+    return correct / n
+
+
+def test_cue_task_learns_offline():
+    """Model + quantile-Huber loss + PER replay learn the cue task from a
+    random-action replay: greedy accuracy chance 0.25 -> >= 0.9."""
+    from rainbowiqn_amd.agent import Agent
+    from rainbowiqn_amd.env import CueEnv
+    from rainbowiqn_amd.replay import ReplayMemory
+
+    cfg = _small_cfg()
+    torch.manual_seed(0)
+    env = CueEnv(cfg)
+    agent = Agent(cfg, env.action_space(), "cpu")
+    mem = ReplayMemory(cfg)
+    rng = np.random.default_rng(0)
+    state = env.reset()
+    for _ in range(1500):
+        a = int(rng.integers(0, 4))
+        ns, r, done = env.step(a)
+        mem.append(state, a, r, done)
+        state = env.reset() if done else ns
+    losses = []
+    for i in range(300):
+        agent.reset_noise()
+        losses.append(agent.learn(mem))
+        if (i + 1) % 100 == 0:
+            agent.update_target_net()
+    assert np.mean(losses[-30:]) < np.mean(losses[:30])
+    acc = _greedy_accuracy(agent, cfg)
+    assert acc >= 0.9, acc
+
+
+def test_train_single_learns_cue():
+    """The full single-process trainer (BASELINE config 1 loop: act ->
+    replay -> learn -> target sync) learns the cue task end to end."""
+    from rainbowiqn_amd.train_single import train
+
+    cfg = _small_cfg(seed=3, learn_start=300, replay_frequency=2,
+                     target_update=200, T_max=10**9)
+    agent, mem, losses = train(cfg, max_steps=1300)
+    assert np.mean(losses[-20:]) < np.mean(losses[:20])
+    acc = _greedy_accuracy(agent, cfg)
+    assert acc >= 0.9, acc
+
+
+def _find_free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _worker_apex_cue(rank, world, port, results):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    torch.manual_seed(rank)
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    cfg = _small_cfg(seed=5, memory_capacity=8192, learn_per_round=4,
+                     target_update=400)
+    comm = Communicator(backend="gloo", device=torch.device("cpu"))
+    runner = ApexRunner(cfg, comm, n_envs=8, chunk_steps=8)
+    rews = []
+    for _ in range(45):
+        runner.run_round()
+        if comm.is_actor:
+            rews.append(float(runner.actor.chunk_meta[:, 1].mean()))
+    out = {"rank": rank}
+    if comm.is_actor:
+        out["early"] = float(np.mean(rews[:5]))
+        out["late"] = float(np.mean(rews[-5:]))
+    results[rank] = out
+    comm.close()
+
+
+def test_train_apex_gloo_learns_cue():
+    """Distributed Ape-X (1 learner + 1 actor over gloo, the RCCL code
+    path): the ACTOR's average reward climbs from chance (-0.5) toward +1,
+    proving experience flow, learning and weight broadcast all work."""
+    world = 2
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_apex_cue, args=(world, port, results), nprocs=world,
+             join=True)
+    actor = dict(results)[1]
+    assert actor["early"] < -0.2, actor  # starts near chance
+    assert actor["late"] > 0.5, actor  # learned policy flows back via bcast
+
+
+def test_cue_env_reward_semantics():
+    """CueEnv pays +1 iff the action matches the cue visible in the state
+    acted on (single and vector variants agree on the contract)."""
+    from rainbowiqn_amd.env import CueEnv, VectorCueEnv
+
+    cfg = _small_cfg()
+    env = CueEnv(cfg)
+    env.reset()
+    for _ in range(20):
+        cue = env._cue
+        _, r_good, done = env.step(cue)
+        assert r_good == 1.0
+        if done:
+            env.reset()
+        cue = env._cue
+        wrong = (cue + 1) % 4
+        _, r_bad, done = env.step(wrong)
+        assert r_bad == -1.0
+        if done:
+            env.reset()
+
+    venv = VectorCueEnv(cfg, 16)
+    venv.reset_all()
+    for _ in range(10):
+        cues = venv._cue.copy()
+        actions = cues.copy()
+        actions[:8] = (actions[:8] + 1) % 4  # first half wrong on purpose
+        _, rewards, _ = venv.step(actions)
+        assert (rewards[:8] == -1.0).all()
+        assert (rewards[8:] == 1.0).all()
+
+
+def test_cue_frame_shows_cue():
+    """The cue block is visually present in the slot the reward keys on."""
+    from rainbowiqn_amd.env import CueEnv
+
+    cfg = _small_cfg()
+    env = CueEnv(cfg)
+    state = env.reset()
+    for _ in range(8):
+        cue = env._cue
+        newest = state[-1]  # [84, 84] float
+        w = 84 // 4
+        slot_means = [float(newest[32:52, s * w + 1 : s * w + 20].mean())
+                      for s in range(4)]
+        assert int(np.argmax(slot_means)) == cue
+        assert slot_means[cue] > 0.95  # 255-valued block
+        state, _, done = env.step(cue)
+        if done:
+            state = env.reset()
