@@ -64,6 +64,8 @@ class Config:
     model: Optional[str] = None  # path to load
     checkpoint_interval: int = 0  # 0 disables periodic checkpoints
     memory_save_path: Optional[str] = None  # replay checkpoint (reference: redis RDB)
+    recorded_env: Optional[str] = None  # .npz trajectory tape: run the real
+    # AtariEnv wrapper over recorded frames (env_recorded.py; no ALE needed)
 
     # -- Ape-X topology (reference: redis flags; here: RCCL over xGMI) -----
     nb_actor: int = 1
@@ -143,6 +145,8 @@ def _add_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--model", type=str, default=None, help="checkpoint to load")
     p.add_argument("--checkpoint-interval", type=int, default=d.checkpoint_interval)
     p.add_argument("--memory-save-path", type=str, default=None)
+    p.add_argument("--recorded-env", type=str, default=None,
+                   help=".npz trajectory tape for the recorded-frame env")
     p.add_argument("--nb-actor", type=int, default=d.nb_actor)
     p.add_argument("--id-actor", type=int, default=d.id_actor)
     p.add_argument("--transition-chunk-size", type=int, default=d.transition_chunk_size)

@@ -280,6 +280,10 @@ def make_vector_env(cfg, n_envs: int, seed_offset: int = 0):
 
 def make_env(cfg, device: str | torch.device = "cpu"):
     """Return the real ALE wrapper when available, else the synthetic env."""
+    if getattr(cfg, "recorded_env", None):
+        from .env_recorded import make_recorded_env
+
+        return make_recorded_env(cfg, cfg.recorded_env, device)
     if cfg.game.lower() == "cue":  # learnable synthetic task, never ALE
         return CueEnv(cfg, device)
     try:  # pragma: no cover - no ALE wheel on this box

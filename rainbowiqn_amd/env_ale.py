@@ -37,23 +37,32 @@ def _resize_84(gray: np.ndarray) -> np.ndarray:
 
 
 class AtariEnv:
-    def __init__(self, cfg, device: str | torch.device = "cpu"):
-        try:
-            import atari_py as ale_mod
+    def __init__(self, cfg, device: str | torch.device = "cpu", ale=None):
+        # `ale` injection: any object with the ALEInterface surface this
+        # class uses (act/reset_game/game_over/lives/getScreenGrayscale/
+        # getMinimalActionSet).  env_recorded.FakeALE drives this wrapper
+        # from a recorded trajectory on disk, so the REAL env code path
+        # (resize, max-pool, life-loss pseudo-terminals, no-op starts) is
+        # exercised end-to-end on boxes without an ALE wheel.
+        if ale is not None:
+            self.ale = ale
+        else:
+            try:
+                import atari_py as ale_mod
 
-            self.ale = ale_mod.ALEInterface()
-            rom = ale_mod.get_game_path(cfg.game.lower().split("noframeskip")[0])
-        except ImportError:  # ale_py fallback
-            import ale_py
+                self.ale = ale_mod.ALEInterface()
+                rom = ale_mod.get_game_path(cfg.game.lower().split("noframeskip")[0])
+            except ImportError:  # ale_py fallback
+                import ale_py
 
-            self.ale = ale_py.ALEInterface()
-            rom = ale_py.roms.get_rom_path(cfg.game)
-        self.ale.setInt("random_seed", cfg.seed)
-        self.ale.setInt("max_num_frames_per_episode", cfg.max_episode_length)
-        self.ale.setFloat("repeat_action_probability", 0.0)
-        self.ale.setInt("frame_skip", 0)
-        self.ale.setBool("color_averaging", False)
-        self.ale.loadROM(rom)
+                self.ale = ale_py.ALEInterface()
+                rom = ale_py.roms.get_rom_path(cfg.game)
+            self.ale.setInt("random_seed", cfg.seed)
+            self.ale.setInt("max_num_frames_per_episode", cfg.max_episode_length)
+            self.ale.setFloat("repeat_action_probability", 0.0)
+            self.ale.setInt("frame_skip", 0)
+            self.ale.setBool("color_averaging", False)
+            self.ale.loadROM(rom)
 
         actions = self.ale.getMinimalActionSet()
         self.actions = {i: int(a) for i, a in enumerate(actions)}
