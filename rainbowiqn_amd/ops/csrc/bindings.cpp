@@ -547,8 +547,24 @@ void quant_fp8(torch::Tensor src, torch::Tensor dst, torch::Tensor scales) {
                    cur_stream());
 }
 
+void tau_embed_fp8(torch::Tensor taus, torch::Tensor wiqnf,
+                   torch::Tensor bias, torch::Tensor feat, torch::Tensor hq,
+                   torch::Tensor hsc, int64_t nq) {
+    CHECK_T(taus, torch::kFloat32);
+    CHECK_BF16(wiqnf); CHECK_BF16(bias); CHECK_BF16(feat);
+    CHECK_T(hq, torch::kUInt8); CHECK_T(hsc, torch::kUInt8);
+    int R = (int)taus.numel();
+    TORCH_CHECK(hq.size(0) == R && hq.size(1) == 3200);
+    TORCH_CHECK(hsc.size(0) == R && hsc.size(1) == 100);
+    TORCH_CHECK(R % (int)nq == 0);
+    riqn_tau_embed_fp8(taus.data_ptr<float>(), wiqnf.data_ptr(),
+                       bias.data_ptr(), feat.data_ptr(), hq.data_ptr(),
+                       hsc.data_ptr(), R, (int)nq, cur_stream());
+}
+
 void heads_gemm_fp8(torch::Tensor hq, torch::Tensor hsc, torch::Tensor wq,
-                    torch::Tensor wsc, torch::Tensor bcat, torch::Tensor out) {
+                    torch::Tensor wsc, torch::Tensor bcat, torch::Tensor out,
+                    int64_t variant) {
     CHECK_T(hq, torch::kUInt8);
     CHECK_T(wq, torch::kUInt8);
     CHECK_BF16(bcat); CHECK_BF16(out);
@@ -557,7 +573,7 @@ void heads_gemm_fp8(torch::Tensor hq, torch::Tensor hsc, torch::Tensor wq,
     TORCH_CHECK(out.numel() == (int64_t)R * 1024);
     riqn_heads_gemm_fp8(hq.data_ptr(), hsc.data_ptr(), wq.data_ptr(),
                         wsc.data_ptr(), bcat.data_ptr(), out.data_ptr(), R,
-                        cur_stream());
+                        (int)variant, cur_stream());
 }
 
 }  // namespace
@@ -585,7 +601,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("stack_gather", &stack_gather);
     m.def("pack_fwd_frag", &pack_fwd_frag);
     m.def("quant_fp8", &quant_fp8);
-    m.def("heads_gemm_fp8", &heads_gemm_fp8);
+    m.def("heads_gemm_fp8", &heads_gemm_fp8, py::arg("hq"), py::arg("hsc"),
+          py::arg("wq"), py::arg("wsc"), py::arg("bcat"), py::arg("out"),
+          py::arg("variant") = -1);
+    m.def("tau_embed_fp8", &tau_embed_fp8);
     m.def("conv_encode", &conv_encode, py::arg("x"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"), py::arg("c1g") = c10::nullopt, py::arg("c2g") = c10::nullopt);
     m.def("conv_encode_gather", &conv_encode_gather, py::arg("ring"), py::arg("env_state"), py::arg("history"), py::arg("ring_head"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"));
     m.def("heads_gemm_tau", &heads_gemm_tau, py::arg("taus"), py::arg("wiqnf"), py::arg("biqn"), py::arg("feat"), py::arg("wcat"), py::arg("bcat"), py::arg("out"), py::arg("nq"));

@@ -25,6 +25,12 @@ typedef __attribute__((ext_vector_type(4))) int i32x4;
 
 DEV float bf(const unsigned short u) { return bf16_to_f32(u); }
 
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+DEV f32x4 mfma16f(bf16x8_t a, bf16x8_t b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+DEV bf16x8_t as_bf16x8f(u16x8 v) { return __builtin_bit_cast(bf16x8_t, v); }
+
 // one thread = one 32-element k-block of one row
 __global__ void quant_rows_fp8_kernel(
     const unsigned short* __restrict__ src,  // [M,3136] bf16
@@ -62,6 +68,133 @@ __global__ void quant_rows_fp8_kernel(
         *(unsigned short*)(out + i) = (unsigned short)(packed & 0xFFFF);
     }
     scales[(int64_t)row * 100 + blk] = (unsigned char)(e + 127);
+}
+
+// tau-embed with the MX-fp8 row quantization FUSED into the epilogue
+// (VERDICT round 1 item 4): h[r,k] = relu(cos_r . Wiqn[k,:] + b[k]) *
+// feat[r/nq, k] is quantized per 32-col block IN REGISTERS (cross-lane amax
+// over the 16-lane row group + the paired n-tile) and stored straight as
+// e4m3 + e8m0 -- the bf16 h tensor and the standalone quant pass (the
+// ~310 MB/step of traffic that made --fp8-heads end-to-end negative) both
+// disappear from the actor path.  Geometry: strips of 28 n-tiles (448
+// cols = 14 aligned 32-col blocks), waves own n-tile PAIRS so each lane
+// group sees its block's full 32 columns.  Cos basis via the Chebyshev
+// recurrence exactly as tau_embed_kernel (model_fwd.hip).
+__global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
+    const float* __restrict__ taus,         // [R]
+    const unsigned short* __restrict__ w,   // [3136,64] bf16 frag-major pack
+    const unsigned short* __restrict__ bias,// [3136] bf16
+    const unsigned short* __restrict__ feat,// [B,3136] bf16
+    unsigned char* __restrict__ hq,         // [R,3200] fp8 out
+    unsigned char* __restrict__ hsc,        // [R,100] e8m0 out
+    int R, int nq) {
+    __shared__ unsigned short ftile[4 * 448];
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wave = t >> 6;
+    const int l15 = lane & 15;
+    const int l4 = lane >> 4;
+    const int mt0 = blockIdx.x * 2;
+    const int nt0 = blockIdx.y * 28;  // 28 tiles = 448 cols = 14 k-blocks
+
+    const int s0 = (mt0 * 16) / nq;
+    const int span = min((mt0 * 16 + 31) / nq - s0 + 1, 4);
+    #pragma unroll 4
+    for (int i = t * 4; i < span * 448; i += 256 * 4) {
+        int si = i / 448, cc = i % 448;
+        int64_t src = (int64_t)min(s0 + si, (R - 1) / nq) * 3136 + nt0 * 16 + cc;
+        *(u16x4*)&ftile[si * 448 + cc] = *(const u16x4*)(feat + src);
+    }
+
+    bf16x8_t afrag[2][2];
+    #pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+        const int r_a = (mt0 + mi) * 16 + l15;
+        const float tau = (r_a < R) ? taus[r_a] : 0.0f;
+        const float th = 3.14159265358979f * tau;
+        const int kbase = l4 * 8;
+        float cm1 = cosf((float)kbase * th);
+        float c0 = cosf((float)(kbase + 1) * th);
+        const float two_ct = 2.0f * cosf(th);
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+            u16x8 av;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                if (j == 0) {
+                    av[j] = f32_to_bf16(cm1);
+                } else if (j == 1) {
+                    av[j] = f32_to_bf16(c0);
+                } else {
+                    float c1 = two_ct * c0 - cm1;
+                    cm1 = c0;
+                    c0 = c1;
+                    av[j] = f32_to_bf16(c1);
+                }
+            }
+            if (ks == 0) {
+                cm1 = cosf((float)(kbase + 32) * th);
+                c0 = two_ct * cm1 - cosf((float)(kbase + 31) * th);
+            }
+            afrag[mi][ks] = as_bf16x8f(av);
+        }
+    }
+    __syncthreads();
+
+    for (int pp = wave; pp < 14; pp += 4) {
+        const int nt_a = nt0 + 2 * pp;  // the pair = one 32-col k-block
+        f32x4 acc[2][2];  // [tile][mi]
+        #pragma unroll
+        for (int ti = 0; ti < 2; ++ti) {
+            const int nt = nt_a + ti;
+            acc[ti][0] = f32x4{0, 0, 0, 0};
+            acc[ti][1] = f32x4{0, 0, 0, 0};
+            #pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                bf16x8_t bfr = as_bf16x8f(
+                    *(const u16x8*)(w + ((nt * 2 + ks) << 9) + lane * 8));
+                acc[ti][0] = mfma16f(afrag[0][ks], bfr, acc[ti][0]);
+                acc[ti][1] = mfma16f(afrag[1][ks], bfr, acc[ti][1]);
+            }
+        }
+        const float b_a = bf(bias[nt_a * 16 + l15]);
+        const float b_b = bf(bias[(nt_a + 1) * 16 + l15]);
+        const int col_a = (nt_a - nt0) * 16 + l15;
+        const int blk = blockIdx.y * 14 + pp;
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int rr = l4 * 4 + j;
+                const int r = (mt0 + mi) * 16 + rr;
+                const int fo = min(r / nq - s0, span - 1) * 448;
+                // bf16-round after relu, then after the feat multiply:
+                // matches tau_embed + quant_rows numerics on the bf16 value
+                float va = bf(f32_to_bf16(fmaxf(acc[0][mi][j] + b_a, 0.0f)));
+                va = bf(f32_to_bf16(va * bf(ftile[fo + col_a])));
+                float vb = bf(f32_to_bf16(fmaxf(acc[1][mi][j] + b_b, 0.0f)));
+                vb = bf(f32_to_bf16(vb * bf(ftile[fo + col_a + 16])));
+                // cross-lane block amax over the 16-lane row group x 2 tiles
+                float amax = fmaxf(fabsf(va), fabsf(vb));
+                #pragma unroll
+                for (int m = 8; m >= 1; m >>= 1)
+                    amax = fmaxf(amax, __shfl_xor(amax, m, 16));
+                int e = 0;
+                if (amax > 0.0f) e = max(ilogbf(amax) - 7, -127);
+                const float inv = exp2f((float)-e);
+                int packed = __builtin_amdgcn_cvt_pk_fp8_f32(
+                    va * inv, vb * inv, 0, false);
+                if (r < R) {
+                    unsigned char* row = hq + (int64_t)r * 3200;
+                    row[nt_a * 16 + l15] = (unsigned char)(packed & 0xFF);
+                    row[(nt_a + 1) * 16 + l15] =
+                        (unsigned char)((packed >> 8) & 0xFF);
+                    if (l15 == 0)
+                        hsc[(int64_t)r * 100 + blk] = (unsigned char)(e + 127);
+                }
+            }
+        }
+    }
 }
 
 // 128x128 tile, BK=128 (one scaled MFMA per fragment pair), 2 glds buffers.
@@ -173,6 +306,130 @@ __global__ __launch_bounds__(256) void heads_gemm_fp8_kernel(
     }
 }
 
+// 256x128 tile, BK=128, NBUF LDS buffers with counted-vmcnt pipelining --
+// the fp8 mirror of model_fwd.hip::heads_gemm_big2_kernel.  Per k-tile a WG
+// stages the same 48 KB as the bf16 kernel (128 B/row either way) but
+// covers TWICE the k-depth, so LDS-DMA issue cost and L2 traffic per FLOP
+// halve while the MX matrix core runs at ~1.9x the bf16 rate: this is the
+// kernel that makes --fp8-heads (BASELINE config 5) faster than bf16, not
+// just parity.  Fragment layout (two 16 B pieces at l4 and 4+l4) probed in
+// scripts/mx_probe.hip; XOR source swizzle as in the bf16 kernel.
+template <int NBUF>
+__global__ __launch_bounds__(512) void heads_gemm_fp8_big_kernel(
+    const unsigned char* __restrict__ hq,   // [R,3200] fp8
+    const unsigned char* __restrict__ hsc,  // [R,100] e8m0
+    const unsigned char* __restrict__ wq,   // [1024,3200] fp8
+    const unsigned char* __restrict__ wsc,  // [1024,100] e8m0
+    const unsigned short* __restrict__ bcat,// [1024] bf16
+    unsigned short* __restrict__ out,       // [R,1024] bf16
+    int R) {
+    __shared__ unsigned char smem[NBUF * (256 + 128) * 128];
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;   // 8 waves
+    const int wr = wave >> 1, wc = wave & 1;  // 4x2 grid of 64x64 quadrants
+    const int l15 = lane & 15, l4 = lane >> 4;
+    int bxi = blockIdx.x, byi = blockIdx.y;
+    if ((gridDim.x & 7) == 0) {
+        const int id = blockIdx.x + blockIdx.y * gridDim.x;
+        const int xcd = id & 7, s = id >> 3;
+        byi = s % gridDim.y;
+        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
+    }
+    const int bm = bxi * 256, bn = byi * 128;
+
+    // 48 glds insts per buffer (A 32 + B 16): wave w issues A [4w,4w+4),
+    // B [2w,2w+2); each inst = 8 rows x 128 B, XOR source swizzle.
+    auto stage = [&](int buf, int k0) {
+        unsigned char* base = smem + buf * ((256 + 128) * 128);
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int inst = wave * 4 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            const int gr = min(bm + row, R - 1);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (hq + (int64_t)gr * 3200 + k0 + piece * 16);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + inst * 1024);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+        #pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            const int inst = wave * 2 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (wq + (int64_t)(bn + row) * 3200 + k0 + piece * 16);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + 256 * 128 + inst * 1024);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][4];
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stage(0, 0);
+    if (NBUF == 3) stage(1, 128);
+    for (int kt = 0; kt < 25; ++kt) {
+        if (NBUF == 3 && kt + 1 < 25)
+            asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + NBUF - 1 < 25) stage((kt + NBUF - 1) % NBUF,
+                                      (kt + NBUF - 1) * 128);
+        const unsigned char* As = smem + (kt % NBUF) * ((256 + 128) * 128);
+        const unsigned char* Bs = As + 256 * 128;
+        // B fragments + scales are fm-invariant: load once, reuse 4x
+        i32x8 bfr[4];
+        int sb[4];
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+            const int nrow = wc * 64 + fn * 16 + l15;
+            const int pb0 = l4 ^ (nrow & 7), pb1 = (4 + l4) ^ (nrow & 7);
+            i32x4 blo = *(const i32x4*)&Bs[nrow * 128 + pb0 * 16];
+            i32x4 bhi = *(const i32x4*)&Bs[nrow * 128 + pb1 * 16];
+            bfr[fn] = i32x8{blo[0], blo[1], blo[2], blo[3],
+                            bhi[0], bhi[1], bhi[2], bhi[3]};
+            sb[fn] = wsc[(int64_t)(bn + nrow) * 100 + kt * 4 + l4];
+        }
+        #pragma unroll
+        for (int fm = 0; fm < 4; ++fm) {
+            const int row = wr * 64 + fm * 16 + l15;
+            const int pa0 = l4 ^ (row & 7), pa1 = (4 + l4) ^ (row & 7);
+            i32x4 alo = *(const i32x4*)&As[row * 128 + pa0 * 16];
+            i32x4 ahi = *(const i32x4*)&As[row * 128 + pa1 * 16];
+            i32x8 a = {alo[0], alo[1], alo[2], alo[3],
+                       ahi[0], ahi[1], ahi[2], ahi[3]};
+            const int sa =
+                hsc[(int64_t)min(bm + row, R - 1) * 100 + kt * 4 + l4];
+            #pragma unroll
+            for (int fn = 0; fn < 4; ++fn)
+                acc[fm][fn] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                    a, bfr[fn], acc[fm][fn], 0, 0, 0, sa, 0, sb[fn]);
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+            const int n = bn + wc * 64 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * 64 + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
 }  // namespace
 
 extern "C" {
@@ -186,9 +443,42 @@ void riqn_quant_fp8(const void* src, void* dst, void* scales, int M,
                        (unsigned char*)dst, (unsigned char*)scales, M);
 }
 
+void riqn_tau_embed_fp8(const float* taus, const void* wiqnf,
+                        const void* bias, const void* feat, void* hq,
+                        void* hsc, int R, int nq, void* stream) {
+    dim3 grid((R + 31) / 32, 7);  // 2 m-tiles of 16 x 28-tile n-strips
+    hipLaunchKernelGGL(tau_embed_fp8_kernel, grid, dim3(256), 0,
+                       (hipStream_t)stream, taus,
+                       (const unsigned short*)wiqnf,
+                       (const unsigned short*)bias,
+                       (const unsigned short*)feat, (unsigned char*)hq,
+                       (unsigned char*)hsc, R, nq);
+}
+
 void riqn_heads_gemm_fp8(const void* hq, const void* hsc, const void* wq,
                          const void* wsc, const void* bcat, void* out, int R,
-                         void* stream) {
+                         int variant, void* stream) {
+    // variant: 0 = 128x128 2-buf (round-1 kernel), 1 = 256x128 3-buf
+    // pipelined, -1 = default (env RIQN_FP8_GEMM override, else 1).
+    if (variant < 0) {
+        static int env_variant = -1;
+        if (env_variant < 0) {
+            const char* e = getenv("RIQN_FP8_GEMM");
+            env_variant = e ? atoi(e) : 1;
+        }
+        variant = env_variant;
+    }
+    if (variant == 1 && R >= 1024) {
+        dim3 grid((R + 255) / 256, 1024 / 128);
+        hipLaunchKernelGGL(heads_gemm_fp8_big_kernel<3>, grid, dim3(512), 0,
+                           (hipStream_t)stream, (const unsigned char*)hq,
+                           (const unsigned char*)hsc,
+                           (const unsigned char*)wq,
+                           (const unsigned char*)wsc,
+                           (const unsigned short*)bcat, (unsigned short*)out,
+                           R);
+        return;
+    }
     dim3 grid((R + 127) / 128, 1024 / 128);
     hipLaunchKernelGGL(heads_gemm_fp8_kernel, grid, dim3(256), 0,
                        (hipStream_t)stream, (const unsigned char*)hq,

@@ -147,9 +147,12 @@ void riqn_pack_bwd_weights(const void* w3, const void* w2, void* w3tp,
 // ---- fp8 MX heads path (model_fp8.hip; BASELINE config 5) ------------------
 void riqn_quant_fp8(const void* src, void* dst, void* scales, int M,
                     void* stream);
+void riqn_tau_embed_fp8(const float* taus, const void* wiqnf,
+                        const void* bias, const void* feat, void* hq,
+                        void* hsc, int R, int nq, void* stream);
 void riqn_heads_gemm_fp8(const void* hq, const void* hsc, const void* wq,
                          const void* wsc, const void* bcat, void* out, int R,
-                         void* stream);
+                         int variant, void* stream);
 
 // batched 4-layer NoisyLinear regeneration (rng.hip)
 struct NoisyDesc {

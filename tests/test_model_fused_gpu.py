@@ -255,3 +255,57 @@ def test_fp8_heads_parity():
     expect = q.mean(dim=1).argmax(dim=1)
     agree = (actions == expect).float().mean().item()
     assert agree == 1.0, agree
+
+
+def test_tau_embed_fp8_fused_parity():
+    """Fused tau-embed+quant (tau_embed_fp8) must produce bit-identical
+    e4m3 payloads and e8m0 scales to the unfused pair (tau_embed DIRECT ->
+    quant_rows) -- the fp8 actor fast path vs its reference decomposition."""
+    from rainbowiqn_amd.ops import get_extension
+    from rainbowiqn_amd.runtime.fused_forward import FusedForward
+
+    ext = get_extension()
+    A = 6
+    _, _, net16 = make_nets(A, seed=31)
+    fw = FusedForward(net16, "cuda", fp8_heads=True)
+    for B, nq in [(64, 32), (32, 8), (13, 32)]:
+        R = B * nq
+        feat = (torch.rand(B, 3136, device="cuda") - 0.3).to(torch.bfloat16)
+        taus = torch.rand(R, device="cuda")
+        h = torch.empty(R, 3136, dtype=torch.bfloat16, device="cuda")
+        hq_ref = torch.zeros(R, 3200, dtype=torch.uint8, device="cuda")
+        sc_ref = torch.zeros(R, 100, dtype=torch.uint8, device="cuda")
+        hq = torch.zeros(R, 3200, dtype=torch.uint8, device="cuda")
+        sc = torch.zeros(R, 100, dtype=torch.uint8, device="cuda")
+        ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq, direct=1)
+        ext.quant_fp8(h, hq_ref, sc_ref)
+        ext.tau_embed_fp8(taus, fw.wiqn, fw.biqn, feat, hq, sc, nq)
+        torch.cuda.synchronize()
+        assert torch.equal(sc[:, :98], sc_ref[:, :98]), (B, nq)
+        assert torch.equal(hq[:, :3136], hq_ref[:, :3136]), (B, nq)
+        # padding stays zero (zero-init + kernel never touches k >= 3136)
+        assert int(hq[:, 3136:].abs().sum()) == 0
+        assert int(sc[:, 98:].abs().sum()) == 0
+
+
+def test_fp8_gemm_big_variant_parity():
+    """256x128 pipelined fp8 GEMM == 128x128 reference kernel bit-for-bit
+    (identical per-element k-order MFMA accumulation)."""
+    from rainbowiqn_amd.ops import get_extension
+    from rainbowiqn_amd.runtime.fused_forward import FusedForward
+
+    ext = get_extension()
+    A = 6
+    _, _, net16 = make_nets(A, seed=32)
+    fw = FusedForward(net16, "cuda", fp8_heads=True)
+    for R in [2048, 1024 + 96]:
+        h = (torch.rand(R, 3136, device="cuda") - 0.3).to(torch.bfloat16)
+        hq = torch.zeros(R, 3200, dtype=torch.uint8, device="cuda")
+        sc = torch.zeros(R, 100, dtype=torch.uint8, device="cuda")
+        ext.quant_fp8(h, hq, sc)
+        out0 = torch.empty(R, 1024, dtype=torch.bfloat16, device="cuda")
+        out1 = torch.empty(R, 1024, dtype=torch.bfloat16, device="cuda")
+        ext.heads_gemm_fp8(hq, sc, fw.wq, fw.wsc, fw.bcat, out0, variant=0)
+        ext.heads_gemm_fp8(hq, sc, fw.wq, fw.wsc, fw.bcat, out1, variant=1)
+        torch.cuda.synchronize()
+        assert torch.equal(out0, out1), R
