@@ -41,7 +41,8 @@ def main():
     rng = torch.tensor([123, 0], dtype=torch.int64, device="cuda")
 
     print(f"{'kernel':<28} {'shape':<18} {'us':>9}")
-    for B, nq, tag in [(256, 32, "act E=256"), (64, 32, "act E=64"),
+    for B, nq, tag in [(1024, 32, "act E=1024"), (256, 32, "act E=256"),
+                       (64, 32, "act E=64"),
                        (32, 32, "learn sel"), (32, 8, "learn tgt")]:
         R = B * nq
         x = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
@@ -79,16 +80,23 @@ def main():
     # fp8 MX heads path
     fw8 = FusedForward(RainbowIQN(cfg, A).cuda().to(torch.bfloat16).train(),
                        "cuda", fp8_heads=True)
-    for B, nq in [(256, 32), (32, 32)]:
+    for B, nq in [(1024, 32), (256, 32), (32, 32)]:
         R = B * nq
-        _, h8, hh8, hq, hsc = fw8._bufs(B, nq)
+        bufs8 = fw8._bufs(B, nq)
+        feat8, h8, hh8, hq, hsc = bufs8
+        taus8 = torch.rand(R, device="cuda")
         us = timeit(lambda: ext.quant_fp8(h8, hq, hsc), iters)
         print(f"{'quant_fp8':<28} {f'R={R}':<18} {us:9.2f}")
-        us = timeit(lambda: ext.heads_gemm_fp8(hq, hsc, fw8.wq, fw8.wsc,
-                                               fw8.bcat, hh8), iters)
+        us = timeit(lambda: ext.tau_embed_fp8(taus8, fw8.wiqn, fw8.biqn,
+                                              feat8, hq, hsc, nq), iters)
+        print(f"{'tau_embed_fp8 (fused q)':<28} {f'R={R}':<18} {us:9.2f}")
         gf = R * 1024 * 3200 * 2 / 1e9
-        print(f"{'heads_gemm_fp8':<28} {f'R={R}':<18} {us:9.2f}  "
-              f"({gf / us * 1e6 / 1e3:.2f} TF)")
+        for var, name in [(0, "fp8 gemm 128x128"), (1, "fp8 gemm 256x128p")]:
+            us = timeit(lambda: ext.heads_gemm_fp8(hq, hsc, fw8.wq, fw8.wsc,
+                                                   fw8.bcat, hh8,
+                                                   variant=var), iters)
+            print(f"{name:<28} {f'R={R}':<18} {us:9.2f}  "
+                  f"({gf / us * 1e6 / 1e3:.2f} TF)")
 
     us = timeit(lambda: ext.noisy_material4(pack, rng), iters)
     print(f"{'noisy_material4':<28} {'4 layers':<18} {us:9.2f}")
