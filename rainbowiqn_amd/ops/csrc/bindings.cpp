@@ -381,13 +381,13 @@ void tau_embed(torch::Tensor taus, torch::Tensor w, torch::Tensor bias,
 }
 
 void heads_gemm(torch::Tensor h, torch::Tensor wcat, torch::Tensor bcat,
-                torch::Tensor out) {
+                torch::Tensor out, int64_t variant) {
     CHECK_BF16(h); CHECK_BF16(wcat); CHECK_BF16(bcat); CHECK_BF16(out);
     int R = (int)h.size(0);
     TORCH_CHECK(h.size(1) == 3136 && wcat.numel() == 1024 * 3136);
     TORCH_CHECK(out.numel() == (int64_t)R * 1024);
     riqn_heads_gemm(h.data_ptr(), wcat.data_ptr(), bcat.data_ptr(),
-                    out.data_ptr(), R, cur_stream());
+                    out.data_ptr(), R, (int)variant, cur_stream());
 }
 
 // mode 0: q_out [R,A] f32; mode 1: act_io out [R/nq]; mode 2: q_out [R] f32,
@@ -609,7 +609,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_encode_gather", &conv_encode_gather, py::arg("ring"), py::arg("env_state"), py::arg("history"), py::arg("ring_head"), py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("w3p"), py::arg("b3"), py::arg("feat"));
     m.def("heads_gemm_tau", &heads_gemm_tau, py::arg("taus"), py::arg("wiqnf"), py::arg("biqn"), py::arg("feat"), py::arg("wcat"), py::arg("bcat"), py::arg("out"), py::arg("nq"));
     m.def("tau_embed", &tau_embed, py::arg("taus"), py::arg("w"), py::arg("bias"), py::arg("feat"), py::arg("h"), py::arg("nq"), py::arg("phi_out") = c10::nullopt, py::arg("cosb_out") = c10::nullopt, py::arg("direct") = -1);
-    m.def("heads_gemm", &heads_gemm);
+    m.def("heads_gemm", &heads_gemm, py::arg("h"), py::arg("wcat"),
+          py::arg("bcat"), py::arg("out"), py::arg("variant") = -1);
     m.def("heads_finish", &heads_finish);
     m.def("finish_bwd_dhh", &finish_bwd_dhh);
     m.def("zgrad", &zgrad);

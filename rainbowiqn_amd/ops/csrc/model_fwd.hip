@@ -1159,7 +1159,27 @@ void riqn_heads_gemm_tau(const float* taus, const void* wiqnf,
 }
 
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
-                     void* out, int R, void* stream) {
+                     void* out, int R, int variant, void* stream) {
+    // variant: 1 = 256x128 NBUF3 (shipped), 2 = 256x256 NBUF2,
+    // 3 = 256x256 NBUF3 (needs 192KB -- rejected at launch; kept for doc),
+    // -1 = default (env RIQN_GEMM override, else 1)
+    if (variant < 0) {
+        static int env_variant = -1;
+        if (env_variant < 0) {
+            const char* e = getenv("RIQN_GEMM");
+            env_variant = e ? atoi(e) : 1;
+        }
+        variant = env_variant;
+    }
+    if (variant == 2 && R >= 8192) {
+        dim3 gridb((R + 255) / 256, 1024 / 256);
+        hipLaunchKernelGGL(heads_gemm_256n_kernel<2>, gridb, dim3(512), 0,
+                           (hipStream_t)stream, (const unsigned short*)h,
+                           (const unsigned short*)wcat,
+                           (const unsigned short*)bcat, (unsigned short*)out,
+                           R);
+        return;
+    }
     if (R >= 8192) {
         dim3 gridb((R + 255) / 256, 1024 / 128);
         hipLaunchKernelGGL(heads_gemm_big2_kernel<3>, gridb, dim3(512), 0,

@@ -99,7 +99,7 @@ void riqn_tau_embed(const float* taus, const void* w, const void* bias,
                     const void* feat, void* h, void* phi_out, void* cosb_out,
                     int R, int nq, int direct, void* stream);
 void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
-                     void* out, int R, void* stream);
+                     void* out, int R, int variant, void* stream);
 void riqn_heads_gemm_tau(const float* taus, const void* wiqnf,
                          const void* biqn, const void* feat, const void* wcat,
                          const void* bcat, void* out, int R, int nq,
