@@ -61,10 +61,13 @@ def main():
         us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq,
                                           direct=True), iters)
         print(f"{'tau_embed direct':<28} {f'R={R}':<18} {us:9.2f}")
-        us = timeit(lambda: ext.heads_gemm(h, fw.wcat, fw.bcat, hh), iters)
         gf = R * 1024 * 3136 * 2 / 1e9
-        print(f"{'heads_gemm':<28} {f'R={R}':<18} {us:9.2f}  "
-              f"({gf / us * 1e6 / 1e3:.2f} TF)")
+        for var, name in ([(1, "heads_gemm 256x128"), (2, "heads_gemm 256x256")]
+                          if R >= 8192 else [(1, "heads_gemm")]):
+            us = timeit(lambda: ext.heads_gemm(h, fw.wcat, fw.bcat, hh,
+                                               variant=var), iters)
+            print(f"{name:<28} {f'R={R}':<18} {us:9.2f}  "
+                  f"({gf / us * 1e6 / 1e3:.2f} TF)")
         us = timeit(lambda: ext.heads_finish(hh, fw.wz, fw.bz, A, nq, 1,
                                              torch.zeros(1, device='cuda'),
                                              acts), iters)

@@ -309,3 +309,22 @@ def test_fp8_gemm_big_variant_parity():
         ext.heads_gemm_fp8(hq, sc, fw.wq, fw.wsc, fw.bcat, out1, variant=1)
         torch.cuda.synchronize()
         assert torch.equal(out0, out1), R
+
+
+def test_bf16_gemm_256n_variant_parity():
+    """256x256 NBUF2 bf16 GEMM == shipped 256x128 NBUF3 kernel bit-for-bit
+    (identical per-element k-order accumulation)."""
+    from rainbowiqn_amd.ops import get_extension
+    from rainbowiqn_amd.runtime.fused_forward import FusedForward
+
+    ext = get_extension()
+    _, _, net16 = make_nets(6, seed=33)
+    fw = FusedForward(net16, "cuda")
+    for R in [8192, 8192 + 160]:
+        h = (torch.rand(R, 3136, device="cuda") - 0.3).to(torch.bfloat16)
+        out1 = torch.empty(R, 1024, dtype=torch.bfloat16, device="cuda")
+        out2 = torch.empty(R, 1024, dtype=torch.bfloat16, device="cuda")
+        ext.heads_gemm(h, fw.wcat, fw.bcat, out1, variant=1)
+        ext.heads_gemm(h, fw.wcat, fw.bcat, out2, variant=2)
+        torch.cuda.synchronize()
+        assert torch.equal(out1, out2), R
