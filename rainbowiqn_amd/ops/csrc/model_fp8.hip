@@ -31,6 +31,29 @@ DEV f32x4 mfma16f(bf16x8_t a, bf16x8_t b, f32x4 c) {
 }
 DEV bf16x8_t as_bf16x8f(u16x8 v) { return __builtin_bit_cast(bf16x8_t, v); }
 
+// e4m3 block exponent: e = ilogb(amax) - 7 for normal amax, clamped to
+// -127; bit-extract instead of ilogbf (which lowers through an f64 frexp
+// chain on this toolchain).  amax == 0 keeps e = 0 (zero payload either
+// way) to stay bit-compatible across the quant kernels.
+DEV int block_exp(float amax) {
+    if (amax <= 0.0f) return 0;
+    int e = (int)((__float_as_uint(amax) >> 23) & 0xFF) - 127 - 7;
+    return max(e, -127);
+}
+
+// xor-butterfly max over 16-lane groups via ds_swizzle (one instruction per
+// step; __shfl_xor's bpermute lowering adds index setup + cndmask per step)
+DEV float group16_max(float v) {
+    #pragma unroll
+    for (int m = 8; m >= 1; m >>= 1) {
+        int pat = (m << 10) | 31;  // full-data mode: xor=m, or=0, and=31
+        float o = __int_as_float(
+            __builtin_amdgcn_ds_swizzle(__float_as_int(v), pat));
+        v = fmaxf(v, o);
+    }
+    return v;
+}
+
 // one thread = one 32-element k-block of one row
 __global__ void quant_rows_fp8_kernel(
     const unsigned short* __restrict__ src,  // [M,3136] bf16
@@ -57,8 +80,7 @@ __global__ void quant_rows_fp8_kernel(
         amax = fmaxf(amax, fmaxf(fmaxf(fabsf(v[i]), fabsf(v[i + 1])),
                                  fmaxf(fabsf(v[i + 2]), fabsf(v[i + 3]))));
     }
-    int e = 0;
-    if (amax > 0.0f) e = max(ilogbf(amax) - 7, -127);
+    const int e = block_exp(amax);
     const float inv = exp2f((float)-e);
     #pragma unroll
     for (int i = 0; i < 32; i += 2) {
@@ -88,22 +110,32 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
     unsigned char* __restrict__ hq,         // [R,3200] fp8 out
     unsigned char* __restrict__ hsc,        // [R,100] e8m0 out
     int R, int nq) {
-    __shared__ unsigned short ftile[4 * 448];
+    __shared__ unsigned short ftile[4 * 800];
     const int t = threadIdx.x;
     const int lane = t & 63;
     const int wave = t >> 6;
     const int l15 = lane & 15;
     const int l4 = lane >> 4;
     const int mt0 = blockIdx.x * 2;
-    const int nt0 = blockIdx.y * 28;  // 28 tiles = 448 cols = 14 k-blocks
+    // 4 strips of 25/25/24/24 k-block PAIRS (50/50/48/48 tiles): the same
+    // WG count as the bf16 tau_embed (the afrag cos build is per-WG work),
+    // every strip start 32-col aligned
+    const int by = blockIdx.y;
+    const int pair0 = by * 25 - max(0, by - 2);
+    const int pairs = by < 2 ? 25 : 24;
+    const int nt0 = pair0 * 2;
+    const int ncols = pairs * 32;
 
     const int s0 = (mt0 * 16) / nq;
     const int span = min((mt0 * 16 + 31) / nq - s0 + 1, 4);
     #pragma unroll 4
-    for (int i = t * 4; i < span * 448; i += 256 * 4) {
-        int si = i / 448, cc = i % 448;
-        int64_t src = (int64_t)min(s0 + si, (R - 1) / nq) * 3136 + nt0 * 16 + cc;
-        *(u16x4*)&ftile[si * 448 + cc] = *(const u16x4*)(feat + src);
+    for (int i = t * 4; i < span * 800; i += 256 * 4) {
+        int si = i / 800, cc = i % 800;
+        if (cc < ncols) {
+            int64_t src =
+                (int64_t)min(s0 + si, (R - 1) / nq) * 3136 + nt0 * 16 + cc;
+            *(u16x4*)&ftile[si * 800 + cc] = *(const u16x4*)(feat + src);
+        }
     }
 
     bf16x8_t afrag[2][2];
@@ -141,7 +173,7 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
     }
     __syncthreads();
 
-    for (int pp = wave; pp < 14; pp += 4) {
+    for (int pp = wave; pp < pairs; pp += 4) {
         const int nt_a = nt0 + 2 * pp;  // the pair = one 32-col k-block
         f32x4 acc[2][2];  // [tile][mi]
         #pragma unroll
@@ -160,14 +192,14 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
         const float b_a = bf(bias[nt_a * 16 + l15]);
         const float b_b = bf(bias[(nt_a + 1) * 16 + l15]);
         const int col_a = (nt_a - nt0) * 16 + l15;
-        const int blk = blockIdx.y * 14 + pp;
+        const int blk = pair0 + pp;
         #pragma unroll
         for (int mi = 0; mi < 2; ++mi) {
             #pragma unroll
             for (int j = 0; j < 4; ++j) {
                 const int rr = l4 * 4 + j;
                 const int r = (mt0 + mi) * 16 + rr;
-                const int fo = min(r / nq - s0, span - 1) * 448;
+                const int fo = min(r / nq - s0, span - 1) * 800;
                 // bf16-round after relu, then after the feat multiply:
                 // matches tau_embed + quant_rows numerics on the bf16 value
                 float va = bf(f32_to_bf16(fmaxf(acc[0][mi][j] + b_a, 0.0f)));
@@ -175,12 +207,9 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
                 float vb = bf(f32_to_bf16(fmaxf(acc[1][mi][j] + b_b, 0.0f)));
                 vb = bf(f32_to_bf16(vb * bf(ftile[fo + col_a + 16])));
                 // cross-lane block amax over the 16-lane row group x 2 tiles
-                float amax = fmaxf(fabsf(va), fabsf(vb));
-                #pragma unroll
-                for (int m = 8; m >= 1; m >>= 1)
-                    amax = fmaxf(amax, __shfl_xor(amax, m, 16));
-                int e = 0;
-                if (amax > 0.0f) e = max(ilogbf(amax) - 7, -127);
+                const float amax =
+                    group16_max(fmaxf(fabsf(va), fabsf(vb)));
+                const int e = block_exp(amax);
                 const float inv = exp2f((float)-e);
                 int packed = __builtin_amdgcn_cvt_pk_fp8_f32(
                     va * inv, vb * inv, 0, false);
@@ -446,7 +475,7 @@ void riqn_quant_fp8(const void* src, void* dst, void* scales, int M,
 void riqn_tau_embed_fp8(const float* taus, const void* wiqnf,
                         const void* bias, const void* feat, void* hq,
                         void* hsc, int R, int nq, void* stream) {
-    dim3 grid((R + 31) / 32, 7);  // 2 m-tiles of 16 x 28-tile n-strips
+    dim3 grid((R + 31) / 32, 4);  // 2 m-tiles of 16 x uneven pair-strips
     hipLaunchKernelGGL(tau_embed_fp8_kernel, grid, dim3(256), 0,
                        (hipStream_t)stream, taus,
                        (const unsigned short*)wiqnf,
