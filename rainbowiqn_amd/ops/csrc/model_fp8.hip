@@ -43,15 +43,20 @@ DEV int block_exp(float amax) {
 
 // xor-butterfly max over 16-lane groups via ds_swizzle (one instruction per
 // step; __shfl_xor's bpermute lowering adds index setup + cndmask per step)
+template <int M>
+DEV float swz_max(float v) {
+    // full-data mode pattern: xor=M, or=0, and=31 (swizzle needs an
+    // immediate, hence the template)
+    float o = __int_as_float(
+        __builtin_amdgcn_ds_swizzle(__float_as_int(v), (M << 10) | 31));
+    return fmaxf(v, o);
+}
+
 DEV float group16_max(float v) {
-    #pragma unroll
-    for (int m = 8; m >= 1; m >>= 1) {
-        int pat = (m << 10) | 31;  // full-data mode: xor=m, or=0, and=31
-        float o = __int_as_float(
-            __builtin_amdgcn_ds_swizzle(__float_as_int(v), pat));
-        v = fmaxf(v, o);
-    }
-    return v;
+    v = swz_max<8>(v);
+    v = swz_max<4>(v);
+    v = swz_max<2>(v);
+    return swz_max<1>(v);
 }
 
 // one thread = one 32-element k-block of one row
