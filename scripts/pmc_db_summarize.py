@@ -60,9 +60,14 @@ def main():
     """
     agg = defaultdict(lambda: [0.0, 0])
     times = defaultdict(lambda: [0.0, 0])
-    seen = set()
+    def clean(n):
+        n = n.replace("(anonymous namespace)::", "")
+        return n.split("(")[0].split("<")[0]
+
+    nrows = 0
     for name, cname, val, dt in cur.execute(q):
-        name = name.split("(")[0]
+        nrows += 1
+        name = clean(name)
         if "rocclr" in name:
             continue
         a = agg[(name, cname)]
@@ -76,8 +81,17 @@ def main():
              FROM {kd} d JOIN {ks} s ON d.{kd_kid}=s.id
              GROUP BY s.display_name"""
     for name, n, avg in cur.execute(q2):
-        name = name.split("(")[0]
-        times[name] = [avg / 1e3, n]
+        name = clean(name)
+        t = times[name]
+        # merge groups that clean() collapses together
+        tot = t[0] * t[1] + (avg / 1e3) * n
+        t[1] += n
+        t[0] = tot / t[1]
+    if nrows == 0:
+        print("NO pmc rows; counters recorded in rocpd_info_pmc:")
+        for r in cur.execute(f"SELECT * FROM {pi} LIMIT 40"):
+            print(" ", r)
+        return
     kernels = sorted({k for (k, _) in agg})
     for k in kernels:
         us, n = times.get(k, [0, 0])
