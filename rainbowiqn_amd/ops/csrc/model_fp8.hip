@@ -357,7 +357,13 @@ __global__ __launch_bounds__(512) void heads_gemm_fp8_big_kernel(
     const unsigned short* __restrict__ bcat,// [1024] bf16
     unsigned short* __restrict__ out,       // [R,1024] bf16
     int R) {
-    __shared__ unsigned char smem[NBUF * (256 + 128) * 128];
+    // per-buffer: A 32 KB | B 16 KB | scales 1.5 KB (384 rows x 4 bytes
+    // for this kt's four 32-blocks).  Scales ride the SAME pipelined
+    // LDS-DMA as the operands: the per-lane scattered byte loads they
+    // replace touched ~16 cache lines per instruction and dominated the
+    // kernel's memory traffic (r02 PMC: fp8 GEMM was fill-bound at HALF
+    // the bf16 kernel's effective rate).
+    __shared__ unsigned char smem[NBUF * ((256 + 128) * 128 + 384 * 4)];
     const int t = threadIdx.x;
     const int lane = t & 63, wave = t >> 6;   // 8 waves
     const int wr = wave >> 1, wc = wave & 1;  // 4x2 grid of 64x64 quadrants
@@ -371,10 +377,28 @@ __global__ __launch_bounds__(512) void heads_gemm_fp8_big_kernel(
     }
     const int bm = bxi * 256, bn = byi * 128;
 
-    // 48 glds insts per buffer (A 32 + B 16): wave w issues A [4w,4w+4),
-    // B [2w,2w+2); each inst = 8 rows x 128 B, XOR source swizzle.
+    // 56 glds insts per buffer (A 32 + B 16 + scales 8): wave w issues A
+    // [4w,4w+4), B [2w,2w+2) and one 48-lane scale-dword piece; each
+    // operand inst = 8 rows x 128 B, XOR source swizzle.
     auto stage = [&](int buf, int k0) {
-        unsigned char* base = smem + buf * ((256 + 128) * 128);
+        unsigned char* base = smem + buf * ((256 + 128) * 128 + 384 * 4);
+        {
+            // scale dwords: rows 0..255 = A (hsc), 256..383 = B (wsc);
+            // row*100 is 4-aligned (100 = 4*25) and k0/32 is a multiple
+            // of 4, so the dword loads are aligned
+            const int srow = wave * 48 + lane;
+            if (lane < 48) {
+                const unsigned char* sp =
+                    srow < 256
+                        ? hsc + (int64_t)min(bm + srow, R - 1) * 100 + k0 / 32
+                        : wsc + (int64_t)(bn + srow - 256) * 100 + k0 / 32;
+                auto* gp =
+                    (const __attribute__((address_space(1))) uint32_t*)sp;
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + 384 * 128 + srow * 4);
+                __builtin_amdgcn_global_load_lds(gp, lp, 4, 0, 0);
+            }
+        }
         #pragma unroll
         for (int i = 0; i < 4; ++i) {
             const int inst = wave * 4 + i;
@@ -408,15 +432,20 @@ __global__ __launch_bounds__(512) void heads_gemm_fp8_big_kernel(
     stage(0, 0);
     if (NBUF == 3) stage(1, 128);
     for (int kt = 0; kt < 25; ++kt) {
+        // newest stage keeps up to 7 loads in flight per wave (4 A + 2 B
+        // + 1 scale piece; lanes >= 48 issue no scale load but 7 is a
+        // per-wave UPPER bound either way)
         if (NBUF == 3 && kt + 1 < 25)
-            asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            asm volatile("s_waitcnt vmcnt(7)" ::: "memory");
         else
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         if (kt + NBUF - 1 < 25) stage((kt + NBUF - 1) % NBUF,
                                       (kt + NBUF - 1) * 128);
-        const unsigned char* As = smem + (kt % NBUF) * ((256 + 128) * 128);
+        const unsigned char* As =
+            smem + (kt % NBUF) * ((256 + 128) * 128 + 384 * 4);
         const unsigned char* Bs = As + 256 * 128;
+        const unsigned char* Ss = As + 384 * 128;
         // B fragments + scales are fm-invariant: load once, reuse 4x
         i32x8 bfr[4];
         int sb[4];
@@ -428,7 +457,7 @@ __global__ __launch_bounds__(512) void heads_gemm_fp8_big_kernel(
             i32x4 bhi = *(const i32x4*)&Bs[nrow * 128 + pb1 * 16];
             bfr[fn] = i32x8{blo[0], blo[1], blo[2], blo[3],
                             bhi[0], bhi[1], bhi[2], bhi[3]};
-            sb[fn] = wsc[(int64_t)(bn + nrow) * 100 + kt * 4 + l4];
+            sb[fn] = Ss[(256 + nrow) * 4 + l4];
         }
         #pragma unroll
         for (int fm = 0; fm < 4; ++fm) {
@@ -438,8 +467,7 @@ __global__ __launch_bounds__(512) void heads_gemm_fp8_big_kernel(
             i32x4 ahi = *(const i32x4*)&As[row * 128 + pa1 * 16];
             i32x8 a = {alo[0], alo[1], alo[2], alo[3],
                        ahi[0], ahi[1], ahi[2], ahi[3]};
-            const int sa =
-                hsc[(int64_t)min(bm + row, R - 1) * 100 + kt * 4 + l4];
+            const int sa = Ss[row * 4 + l4];
             #pragma unroll
             for (int fn = 0; fn < 4; ++fn)
                 acc[fm][fn] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(

@@ -1171,7 +1171,11 @@ void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
         }
         variant = env_variant;
     }
-    if (variant == 2 && R >= 8192) {
+    if ((variant == 2 || (variant == 1 && R >= 16384)) && R >= 8192) {
+        // 256x256 wins once the grid fills the chip (R/256 x 4 >= 256 WGs):
+        // 193 -> 157.9 us (1090 -> 1333 TF) at the actor shape R=32768; at
+        // R=8192 its 128-WG grid half-fills the chip and loses (707 TF) --
+        // kbench r02, fill-rate-bound analysis in gpurun_out/r02_pmc_sq.txt
         dim3 gridb((R + 255) / 256, 1024 / 256);
         hipLaunchKernelGGL(heads_gemm_256n_kernel<2>, gridb, dim3(512), 0,
                            (hipStream_t)stream, (const unsigned short*)h,
