@@ -81,8 +81,10 @@ def build_runner(cfg, comm, n_envs, chunk_steps):
     # way).  Reference fill point: learn_start transitions in a 1M ring.
     if comm.is_learner:
         n_act = n_actions_for(cfg.game)
-        per_ring = cfg.learn_start // comm.n_actors + 256
-        for a in range(comm.n_actors):
+        rings = getattr(runner.replay, "n_rings",
+                        getattr(runner.replay, "n_actors", comm.n_actors))
+        per_ring = cfg.learn_start // rings + 256
+        for a in range(rings):
             prefill_ring(runner.replay, a, per_ring, n_act, cfg.device,
                          seed=a)
     return runner
@@ -187,7 +189,12 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=200)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--n-envs", type=int, default=1024, help="env streams per actor rank")
+    p.add_argument("--n-envs", type=int, default=1024,
+                   help="env streams per actor loop")
+    p.add_argument("--nb-actor", type=int, default=1,
+                   help="concurrent local actor loops (colocated overlap "
+                        "runs each on its own stream; total env streams = "
+                        "nb_actor x n_envs)")
     p.add_argument("--chunk-steps", type=int, default=8, help="env steps per round")
     p.add_argument("--learn-per-round", type=int, default=8)
     p.add_argument("--device", type=str, default=None)
@@ -230,6 +237,7 @@ def main():
         dtype="bf16" if on_gpu else "fp32",
         device="cuda" if on_gpu else "cpu",
         seed=int(os.environ.get("RANK", 0)) + 1,
+        nb_actor=args.nb_actor,
         fp8_heads=args.fp8_heads,
         hipgraph=args.hipgraph,
         fused_forward=args.fused_forward,

@@ -308,12 +308,15 @@ class ApexRunner:
                     actor_quota(cfg.batch_size, comm.n_actors,
                                 comm.actor_index()),
                     cfg.learn_per_round, dev, fast=True)
-        # colocated overlap: actor round runs on its own HIP stream,
-        # concurrent with the learner's grad steps (one-round bounded
-        # staleness -- the same contract as the multi-GPU split).  The
-        # outgoing chunk is double-buffered so the learner appends round
-        # k-1's chunk while round k records.
-        self.overlap = (comm.world_size == 1 and M == 1 and self.gactors
+        # colocated overlap: each actor round runs on its own HIP stream,
+        # concurrent with the learner's grad steps AND with the other local
+        # actor loops (one-round bounded staleness -- the same contract as
+        # the multi-GPU split).  With --nb-actor M > 1 the M half-size act
+        # chains interleave on M queues: the act-path kernels are ~70%
+        # latency waits (r02 PMC), so co-resident independent chains fill
+        # the stall slots.  Outgoing chunks are double-buffered so the
+        # learner appends round k-1's chunks while round k records.
+        self.overlap = (comm.world_size == 1 and bool(self.gactors)
                         and self.glearner is not None
                         and getattr(cfg, "overlap_colocated", True))
         if self.overlap:
@@ -321,7 +324,7 @@ class ApexRunner:
 
             d = self.gactors[0].device
             c = self.n_envs * self.chunk_steps
-            self._astream = torch.cuda.Stream(device=d)
+            self._astreams = [torch.cuda.Stream(device=d) for _ in range(M)]
             # RIQN_LPRIO=1: learner work on a high-priority HSA queue so its
             # small grids (batch 32) dispatch ahead of the actor's queued
             # chip-filling WGs instead of waiting behind them (opt-in until
@@ -353,22 +356,29 @@ class ApexRunner:
                         return w
 
                     lp = ext.create_cu_masked_stream(_mask(0, cus))
-                    ap = ext.create_cu_masked_stream(_mask(cus, total))
                     self._lstream = torch.cuda.ExternalStream(lp, device=d)
-                    self._astream = torch.cuda.ExternalStream(ap, device=d)
                     # destroy at exit: a leaked HSA queue crashes profiler
                     # teardown (__cxa_finalize segfault under rocprofv3)
                     import atexit
 
                     atexit.register(ext.destroy_stream, lp)
-                    atexit.register(ext.destroy_stream, ap)
-                    # unmasked stream for the pre-learn_start warmup phase
-                    self._astream_full = torch.cuda.Stream(device=d)
+                    # all actor queues share the actor CU partition
+                    self._astreams = []
+                    for _ in range(M):
+                        ap = ext.create_cu_masked_stream(_mask(cus, total))
+                        atexit.register(ext.destroy_stream, ap)
+                        self._astreams.append(
+                            torch.cuda.ExternalStream(ap, device=d))
+                    # unmasked streams for the pre-learn_start warmup phase
+                    self._astreams_full = [torch.cuda.Stream(device=d)
+                                           for _ in range(M)]
                 except Exception:  # no CU-mask support: plain streams
                     pass
-            self._staging_f = [torch.zeros(c, 84, 84, dtype=torch.uint8,
-                                           device=d) for _ in range(2)]
-            self._staging_m = [torch.zeros(c, 3, device=d) for _ in range(2)]
+            self._staging_f = [[torch.zeros(c, 84, 84, dtype=torch.uint8,
+                                            device=d) for _ in range(2)]
+                               for _ in range(M)]
+            self._staging_m = [[torch.zeros(c, 3, device=d) for _ in range(2)]
+                               for _ in range(M)]
             self._round_idx = 0
 
     # -- one Ape-X round ----------------------------------------------------
@@ -536,18 +546,19 @@ class ApexRunner:
                 comm.broadcast_weights(self.flat_weights, src=0)
 
     def _run_round_overlap(self, learn: bool) -> None:
-        """Colocated round with actor/learner stream concurrency."""
+        """Colocated round: M actor streams + learner stream, concurrent."""
         cfg = self.cfg
         gl = self.glearner
-        loop = self.gactors[0]
+        M = self.n_local_actors
         j = self._round_idx % 2
         prev = (self._round_idx - 1) % 2
         outer = torch.cuda.current_stream()
         # round-boundary weight snapshot: the previous round's learner work
-        # is complete w.r.t. the outer stream here, so the actor's flat copy
-        # sees consistent (one-round-stale) weights -- never torn mid-step
-        if self._actor_flats[0] is not None:
-            self._actor_flats[0].copy_(gl.flat_online)
+        # is complete w.r.t. the outer stream here, so the actors' flat
+        # copies see consistent (one-round-stale) weights -- never torn
+        for flat in self._actor_flats:
+            if flat is not None:
+                flat.copy_(gl.flat_online)
         # learner work optionally rides a high-priority queue (RIQN_LPRIO):
         # its small grids then dispatch ahead of the actor's queued WGs
         from contextlib import nullcontext
@@ -556,27 +567,31 @@ class ApexRunner:
             if self._lstream is not None else nullcontext
         if self._lstream is not None:
             self._lstream.wait_stream(outer)
-        # learner stream: append LAST round's chunk, then grad steps; these
+        # learner stream: append LAST round's chunks, then grad steps; these
         # small-grid kernels fill the CUs the actor's big grids leave idle
         with lctx():
             if self._round_idx > 0:
                 with self._sp("append"):
-                    gl.append_actor_chunk(0, self._staging_f[prev],
-                                          meta=self._staging_m[prev])
-        # actor stream: round + chunk stage-out, concurrent with the
-        # learner.  Phase-aware partition: before learn_start the learner
-        # queue is idle, so the actor runs on the UNMASKED stream (all 256
-        # CUs) instead of its 160-CU share.
-        astream = self._astream
-        if (getattr(self, "_astream_full", None) is not None
+                    for a in range(M):
+                        gl.append_actor_chunk(a, self._staging_f[a][prev],
+                                              meta=self._staging_m[a][prev])
+        # actor streams: each loop's round + chunk stage-out, concurrent
+        # with the learner and each other.  Phase-aware partition: before
+        # learn_start the learner queue is idle, so actors run on UNMASKED
+        # streams (all 256 CUs) instead of their partition share.
+        astreams = self._astreams
+        if (getattr(self, "_astreams_full", None) is not None
                 and len(gl.replay) < cfg.learn_start):
-            astream = self._astream_full
-        astream.wait_stream(outer)
-        with torch.cuda.stream(astream):
-            with self._sp("actor"):
-                loop.run_round()
-                self._staging_f[j].copy_(loop.chunk_frames, non_blocking=True)
-                self._staging_m[j].copy_(loop.chunk_meta, non_blocking=True)
+            astreams = self._astreams_full
+        for a, loop in enumerate(self.gactors):
+            astreams[a].wait_stream(outer)
+            with torch.cuda.stream(astreams[a]):
+                with self._sp(f"actor{a}" if M > 1 else "actor"):
+                    loop.run_round()
+                    self._staging_f[a][j].copy_(loop.chunk_frames,
+                                                non_blocking=True)
+                    self._staging_m[a][j].copy_(loop.chunk_meta,
+                                                non_blocking=True)
         with lctx():
             if learn and len(gl.replay) >= cfg.learn_start:
                 with self._sp("learn"):
@@ -590,7 +605,8 @@ class ApexRunner:
                     gl.maintenance()
         if self._lstream is not None:
             outer.wait_stream(self._lstream)
-        outer.wait_stream(astream)
+        for a in range(M):
+            outer.wait_stream(astreams[a])
         self._round_idx += 1
 
     def _run_round_ref(self, learn: bool) -> None:

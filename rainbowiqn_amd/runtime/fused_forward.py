@@ -168,7 +168,19 @@ class FusedForward:
 
     def _tau_heads_fp8(self, feat, taus, nq, bufs, hh):
         """tau-embed -> MX-fp8 heads GEMM with the row quantization fused
-        into tau-embed's stores (h never exists in bf16 on this path)."""
+        into tau-embed's stores (h never exists in bf16 on this path).
+
+        Small R falls back to the bf16 pair: below ~2048 rows the fp8
+        GEMM's grid can't fill the chip and loses to bf16 (kbench r02:
+        R=256 fp8 ~30 us vs bf16 16.5), and the learner's target forward
+        (R = B*N' = 256) sits on the critical path 8x per round."""
+        R = hh.shape[0] if hh.dim() == 2 else taus.numel()
+        if R < 2048:
+            h = bufs[1]
+            self.ext.tau_embed(taus.view(-1), self.wiqn, self.biqn, feat, h,
+                               nq, direct=TAU_DIRECT)
+            self.ext.heads_gemm(h, self.wcat, self.bcat, hh)
+            return hh
         hq, hsc = bufs[3], bufs[4]
         if FP8_FUSE_QUANT:
             self.ext.tau_embed_fp8(taus.view(-1), self.wiqn, self.biqn,
