@@ -178,35 +178,18 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
     }
     __syncthreads();
 
-    // W-fragment double buffer (see tau_embed_kernel): prefetch the next
-    // pair's four 1 KB wave-loads behind the current pair's MFMA + quant
-    u16x8 wbuf[2][4];
-    if (wave < pairs) {
-        const int nt = nt0 + 2 * wave;
-        #pragma unroll
-        for (int q = 0; q < 4; ++q)
-            wbuf[0][q] =
-                *(const u16x8*)(w + ((nt * 2 + q) << 9) + lane * 8);
-    }
-    int it = 0;
-    for (int pp = wave; pp < pairs; pp += 4, ++it) {
-        const int cur = it & 1;
-        if (pp + 4 < pairs) {
-            const int nt = nt0 + 2 * (pp + 4);
-            #pragma unroll
-            for (int q = 0; q < 4; ++q)
-                wbuf[cur ^ 1][q] =
-                    *(const u16x8*)(w + ((nt * 2 + q) << 9) + lane * 8);
-        }
+    for (int pp = wave; pp < pairs; pp += 4) {
         const int nt_a = nt0 + 2 * pp;  // the pair = one 32-col k-block
         f32x4 acc[2][2];  // [tile][mi]
         #pragma unroll
         for (int ti = 0; ti < 2; ++ti) {
+            const int nt = nt_a + ti;
             acc[ti][0] = f32x4{0, 0, 0, 0};
             acc[ti][1] = f32x4{0, 0, 0, 0};
             #pragma unroll
             for (int ks = 0; ks < 2; ++ks) {
-                bf16x8_t bfr = as_bf16x8f(wbuf[cur][ti * 2 + ks]);
+                bf16x8_t bfr = as_bf16x8f(
+                    *(const u16x8*)(w + ((nt * 2 + ks) << 9) + lane * 8));
                 acc[ti][0] = mfma16f(afrag[0][ks], bfr, acc[ti][0]);
                 acc[ti][1] = mfma16f(afrag[1][ks], bfr, acc[ti][1]);
             }

@@ -410,31 +410,13 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
     }
     __syncthreads();
 
-    // W-fragment double buffer: the next tile's two 1 KB wave-loads are
-    // issued while the current tile's MFMAs + epilogue run -- the loop was
-    // 50% parked on W-load latency before (r02 PMC, WAIT_ANY 50.6%)
-    u16x8 wbuf[2][2];
-    {
-        const int nt = nt0 + wave;
-        if (nt < nt0 + 49) {
-            wbuf[0][0] = *(const u16x8*)(w + ((nt * 2 + 0) << 9) + lane * 8);
-            wbuf[0][1] = *(const u16x8*)(w + ((nt * 2 + 1) << 9) + lane * 8);
-        }
-    }
-    int it = 0;
-    for (int nt = nt0 + wave; nt < nt0 + 49; nt += 4, ++it) {
-        const int cur = it & 1;
-        if (nt + 4 < nt0 + 49) {
-            wbuf[cur ^ 1][0] =
-                *(const u16x8*)(w + (((nt + 4) * 2 + 0) << 9) + lane * 8);
-            wbuf[cur ^ 1][1] =
-                *(const u16x8*)(w + (((nt + 4) * 2 + 1) << 9) + lane * 8);
-        }
+    for (int nt = nt0 + wave; nt < nt0 + 49; nt += 4) {
         const int n = nt * 16 + l15;
         f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-            bf16x8 bfr = as_bf16x8(wbuf[cur][ks]);
+            bf16x8 bfr = as_bf16x8(
+                *(const u16x8*)(w + ((nt * 2 + ks) << 9) + lane * 8));
             acc[0] = mfma16(afrag[0][ks], bfr, acc[0]);
             acc[1] = mfma16(afrag[1][ks], bfr, acc[1]);
         }
