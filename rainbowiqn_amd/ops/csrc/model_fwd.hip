@@ -410,16 +410,17 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
     }
     __syncthreads();
 
-    for (int nt = nt0 + wave; nt < nt0 + 49; nt += 4) {
+    // per-tile body; the caller software-pipelines the two W-fragment
+    // loads (wv0/wv1) one tile ahead with STATICALLY-named registers -- a
+    // dynamically indexed buffer array lowered to ~1300 v_cndmask selects
+    // and ran 10x slower (PERF_LOG r02 rejected row)
+    auto tile_body = [&](int nt, u16x8 wv0, u16x8 wv1) {
         const int n = nt * 16 + l15;
         f32x4 acc[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
-        #pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-            bf16x8 bfr = as_bf16x8(
-                *(const u16x8*)(w + ((nt * 2 + ks) << 9) + lane * 8));
-            acc[0] = mfma16(afrag[0][ks], bfr, acc[0]);
-            acc[1] = mfma16(afrag[1][ks], bfr, acc[1]);
-        }
+        acc[0] = mfma16(afrag[0][0], as_bf16x8(wv0), acc[0]);
+        acc[1] = mfma16(afrag[1][0], as_bf16x8(wv0), acc[1]);
+        acc[0] = mfma16(afrag[0][1], as_bf16x8(wv1), acc[0]);
+        acc[1] = mfma16(afrag[1][1], as_bf16x8(wv1), acc[1]);
         const float bb = bf(bias[n]);
         const int col = (nt - nt0) * 16 + l15;
         #pragma unroll
@@ -440,6 +441,28 @@ __global__ __launch_bounds__(256) void tau_embed_kernel(
                 }
                 if (phi_out != nullptr && r < R)
                     phi_out[(int64_t)r * 3136 + nt * 16 + l15] = ph;
+            }
+        }
+    };
+    auto wload = [&](int nt, int ks) {
+        return *(const u16x8*)(w + ((nt * 2 + ks) << 9) + lane * 8);
+    };
+    {
+        const int end = nt0 + 49;
+        int nt = nt0 + wave;
+        if (nt < end) {
+            u16x8 a0 = wload(nt, 0), a1 = wload(nt, 1);
+            while (true) {
+                const int ntB = nt + 4;
+                u16x8 b0, b1;
+                if (ntB < end) { b0 = wload(ntB, 0); b1 = wload(ntB, 1); }
+                tile_body(nt, a0, a1);
+                if (ntB >= end) break;
+                const int ntC = ntB + 4;
+                if (ntC < end) { a0 = wload(ntC, 0); a1 = wload(ntC, 1); }
+                tile_body(ntB, b0, b1);
+                if (ntC >= end) break;
+                nt = ntC;
             }
         }
     }
