@@ -175,20 +175,19 @@ __global__ __launch_bounds__(512) void conv_encode_kernel(
         const int m = mt * 16 + l15;
         const int oy = m / 9, ox = m % 9;
         const bool mv = m < 81;
-        // gather addresses are fixed per lane except the channel plane,
-        // which advances +2 per k-step (+2*20*C1_STRIDE ushorts) -- the
-        // per-element index arithmetic dominated this loop otherwise
-        int ga[8];
+        // A-taps via ALIGNED dword LDS reads: each lane's 8 taps are two
+        // rows of 4 consecutive x positions, and (c*20 + y)*C1_STRIDE +
+        // ox*2 is always even (C1_STRIDE = 24), so each row is two aligned
+        // u32 reads (ds_read2-able) instead of four scalar u16 reads --
+        // the scalar gather's 8 LDS issues per MFMA dominated this loop.
+        // Addresses advance +2 channel planes per k-step (stays even).
+        int ga0, ga4;
         {
             const int c0 = l4 >> 1;
             const int ky0 = (l4 & 1) * 2;  // t16 = (l4*8)&15 in {0,8}
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int row = ky0 + (j >> 2), kx = j & 3;
-                ga[j] = mv ? LDS_C1 + (c0 * 20 + oy * 2 + row) * C1_STRIDE +
-                                 ox * 2 + kx
-                           : LDS_C1 + c0 * 20 * C1_STRIDE;
-            }
+            ga0 = mv ? LDS_C1 + (c0 * 20 + oy * 2 + ky0) * C1_STRIDE + ox * 2
+                     : LDS_C1 + c0 * 20 * C1_STRIDE;
+            ga4 = mv ? ga0 + C1_STRIDE : LDS_C1 + c0 * 20 * C1_STRIDE;
         }
         bf16x8 bbuf[2];
         bbuf[0] = as_bf16x8(*(const u16x8*)(w2 + (nq2 << 9) + lane * 8));
@@ -198,11 +197,16 @@ __global__ __launch_bounds__(512) void conv_encode_kernel(
             if (ks + 1 < 16)
                 bbuf[cur ^ 1] = as_bf16x8(*(const u16x8*)(
                     w2 + (((ks + 1) * 4 + nq2) << 9) + lane * 8));
+            const uint32_t* p0 = (const uint32_t*)&lds[ga0];
+            const uint32_t* p1 = (const uint32_t*)&lds[ga4];
+            const uint32_t d0 = p0[0], d1 = p0[1], d2 = p1[0], d3 = p1[1];
             u16x8 av;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) av[j] = lds[ga[j]];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) ga[j] += 2 * 20 * C1_STRIDE;
+            av[0] = (unsigned short)d0; av[1] = (unsigned short)(d0 >> 16);
+            av[2] = (unsigned short)d1; av[3] = (unsigned short)(d1 >> 16);
+            av[4] = (unsigned short)d2; av[5] = (unsigned short)(d2 >> 16);
+            av[6] = (unsigned short)d3; av[7] = (unsigned short)(d3 >> 16);
+            ga0 += 2 * 20 * C1_STRIDE;
+            ga4 += 2 * 20 * C1_STRIDE;
             acc1 = mfma16(as_bf16x8(av), bbuf[cur], acc1);
         }
         {
@@ -230,18 +234,30 @@ __global__ __launch_bounds__(512) void conv_encode_kernel(
         const int m = mt * 16 + l15;
         const int oy = m / 7, ox = m % 7;
         const bool mv = m < 49;
-        int ga[8];
-        unsigned short okm[8];
+        // A-taps via aligned dword superset reads: a lane's 8 taps are two
+        // rows of (kx 0..2 + a masked kx=3 slot); each row's 3 ushorts sit
+        // inside the two aligned dwords at (base & ~1), selected by the
+        // lane-invariant base parity -- 4 u32 LDS reads + a few VALU per
+        // MFMA instead of 8 scalar u16 reads (same rewrite as conv2).
+        // Invalid rows (ky = 3 at t16 = 8, or m >= 49) read a safe dummy
+        // and are fully zero-masked, matching the scalar gather exactly.
+        int gb[2];
+        unsigned int rm[2];
+        bool codd;
         {
             const int c0 = l4 >> 1;
             const int t16 = (l4 & 1) * 8;
+            const int ky0 = t16 >> 2;  // 0 or 2
+            codd = ((ox & 1) != 0);
             #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int tt = t16 + j, ky = tt >> 2, kx = tt & 3;
-                const bool ok = mv && ky < 3 && kx < 3;
-                ga[j] = LDS_C2 + (c0 * 9 + (ok ? oy + ky : 0)) * C2_STRIDE +
-                        (ok ? ox + kx : 0);
-                okm[j] = ok ? 0xFFFFu : 0;
+            for (int rr2 = 0; rr2 < 2; ++rr2) {
+                const int ky = ky0 + rr2;
+                const bool ok = mv && ky < 3;
+                const int base = LDS_C2 +
+                    (c0 * 9 + (ok ? oy + ky : 0)) * C2_STRIDE +
+                    (ok ? ox : 0);
+                gb[rr2] = base & ~1;
+                rm[rr2] = ok ? 0xFFFFFFFFu : 0u;
             }
         }
         bf16x8 bbuf[2][2];
@@ -260,10 +276,20 @@ __global__ __launch_bounds__(512) void conv_encode_kernel(
             }
             u16x8 av;
             #pragma unroll
-            for (int j = 0; j < 8; ++j)
-                av[j] = lds[ga[j]] & okm[j];
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) ga[j] += 2 * 9 * C2_STRIDE;
+            for (int rr2 = 0; rr2 < 2; ++rr2) {
+                const uint32_t* p = (const uint32_t*)&lds[gb[rr2]];
+                const uint32_t d0 = p[0] & rm[rr2], d1 = p[1] & rm[rr2];
+                // kx 0..2 from the superset, kx 3 slot always zero
+                const uint32_t w01 =
+                    codd ? __builtin_amdgcn_alignbit(d1, d0, 16) : d0;
+                const unsigned short w2v = (unsigned short)(
+                    codd ? (d1 >> 16) : d1);
+                av[rr2 * 4 + 0] = (unsigned short)w01;
+                av[rr2 * 4 + 1] = (unsigned short)(w01 >> 16);
+                av[rr2 * 4 + 2] = w2v;
+                av[rr2 * 4 + 3] = 0;
+                gb[rr2] += 2 * 9 * C2_STRIDE;
+            }
             bf16x8 a = as_bf16x8(av);
             acc[0] = mfma16(a, bbuf[cur][0], acc[0]);
             acc[1] = mfma16(a, bbuf[cur][1], acc[1]);

@@ -78,6 +78,13 @@ class GpuLearner:
         self.rng_state = torch.tensor(
             [cfg.seed * 7919 + 13, 0], dtype=torch.int64, device=self.device
         )
+        # target-net noise gets its OWN philox state so its regeneration can
+        # ride the side stream concurrently with the online net's (two
+        # streams bumping one offset would race; the reference's target net
+        # draws independent noise anyway)
+        self.rng_state_tgt = torch.tensor(
+            [cfg.seed * 7919 + 7001, 0], dtype=torch.int64,
+            device=self.device)
         b = self.B
         # one contiguous buffer -> ONE rand_uniform launch per step
         self._taus_all = torch.empty(b * (self.N + self.Np + self.K), 1,
@@ -136,6 +143,12 @@ class GpuLearner:
         self.dtheta = z(B, N, dt=torch.float32)
         self.loss_buf = z(1, dt=torch.float32)
         self.per_sample = z(B, dt=torch.float32)
+        # side stream for the learn step's independent chains (double-Q +
+        # target forward; z/heads weight grads).  The grids are tiny (16-64
+        # WGs), so the captured graph replays the two chains CONCURRENTLY
+        # on the learner's CU partition instead of serializing ~30 small
+        # kernels.  Fork/join edges via wait_stream during capture.
+        self._bstream = torch.cuda.Stream(device=self.device)
         # grad views (slices of the flat bf16 grad buffer, written in-place)
         net = self.online
         g = lambda p: p.grad
@@ -170,10 +183,15 @@ class GpuLearner:
         b, N, A = self.B, self.N, self.action_space
         fo, ft = self.fused_online, self.fused_target
         ext.rand_uniform(self._taus_all, self.rng_state)
+        # target-net noise + packs on the side stream (own philox state),
+        # concurrent with the online net's
+        main0 = torch.cuda.current_stream()
+        self._bstream.wait_stream(main0)
+        with torch.cuda.stream(self._bstream):
+            self._reset_noise_device(self.target, self.rng_state_tgt)
+            ft.pack_weights()
         self._reset_noise_device(self.online)
-        self._reset_noise_device(self.target)
         fo.pack_weights()
-        ft.pack_weights()
         ext.pack_bwd_weights(fo.w3raw, fo.w2raw, self.w3tp, self.w2tp)
         idx, s, a, r, s2, nt, w = self.replay.sample()
         # ---- online conv over [s; s'] in ONE call (activations saved for
@@ -183,17 +201,24 @@ class GpuLearner:
                         fo.w3p, fo.b3, self.feat_cat, self.c1g, self.c2g)
         feat_g = self.feat_cat[:b]
         from .fused_forward import TAU_DIRECT
+        main = torch.cuda.current_stream()
+        sb = self._bstream
+        # fork: double-Q select + target forward (side) run concurrently
+        # with the online theta chain (main); both need only the conv
+        # features and the pre-drawn taus
+        sb.wait_stream(main)
+        with torch.cuda.stream(sb):
+            fo.act_from_feat(self.feat_cat[b:], self.taus_select.view(-1),
+                             self.K, self.a_star)
+            ft.theta(s2, self.taus_target.view(-1), self.Np, self.a_star,
+                     self.theta_p_buf)
         ext.tau_embed(self.taus_online.view(-1), fo.wiqn, fo.biqn, feat_g,
                       h_g, N, phi_out=self.phi, cosb_out=self.cosb,
                       direct=TAU_DIRECT)
         ext.heads_gemm(h_g, fo.wcat, fo.bcat, hh_g)
         ext.heads_finish(hh_g, fo.wz, fo.bz, A, N, 2,
                          self.theta_buf.view(-1), a)
-        # ---- double-Q select (online, from the s' conv half) + target ----
-        fo.act_from_feat(self.feat_cat[b:], self.taus_select.view(-1),
-                         self.K, self.a_star)
-        ft.theta(s2, self.taus_target.view(-1), self.Np, self.a_star,
-                 self.theta_p_buf)
+        main.wait_stream(sb)  # join: loss needs theta AND theta'
         # ---- fused loss: loss, per-sample priorities, dtheta -------------
         ext.quantile_huber(self.theta_buf, self.theta_p_buf,
                            self.taus_online.view(b, N), r, nt, w,
@@ -202,14 +227,18 @@ class GpuLearner:
         # ---- backward chain ----------------------------------------------
         dt_flat = self.dtheta.view(-1)
         ext.finish_bwd_dhh(dt_flat, a, hh_g, fo.wz, self.dhh, A, N)
-        ext.zgrad(dt_flat, a, hh_g, A, N, self._z_eps, self._z_grads)
+        # fork: the z-layer / heads weight-grad family only needs dhh and
+        # saved activations -- it runs beside the dx -> conv backward chain
+        sb.wait_stream(main)
+        with torch.cuda.stream(sb):
+            ext.zgrad(dt_flat, a, hh_g, A, N, self._z_eps, self._z_grads)
+            ext.gemm_dwcat(self.dhh, h_g, self._h_weps[0], self._h_weps[1],
+                           self._h_wgrads)
+            ext.bias_grad_heads(self.dhh, self._h_beps[0], self._h_beps[1],
+                                self._h_bgrads)
         self.dfeat.zero_()
         ext.gemm_dx(self.dhh, fo.wcat, feat_g, self.phi, self.dphi,
                     self.dfeat, N)
-        ext.gemm_dwcat(self.dhh, h_g, self._h_weps[0], self._h_weps[1],
-                       self._h_wgrads)
-        ext.bias_grad_heads(self.dhh, self._h_beps[0], self._h_beps[1],
-                            self._h_bgrads)
         ext.tau_dw(self.dphi, self.cosb, self._iqn_wgrad)
         ext.bias_grad_cols(self.dphi, self._iqn_bgrad)
         ext.conv_bwd_data(self.dfeat, feat_g, self.c1g, self.c2g, self.w3tp,
@@ -219,13 +248,14 @@ class GpuLearner:
         ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 12)
         ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 32)
         ext.conv_finalize(self._conv_scr, self._conv_grads)
+        main.wait_stream(sb)  # join: Adam reads the whole flat grad
         # ---- optimizer + priorities --------------------------------------
         self.opt.step()
         self.replay.update_priorities(idx, self.per_sample)
         self.loss_static.copy_(self.loss_buf)
 
     # -- pieces -------------------------------------------------------------
-    def _reset_noise_device(self, net: RainbowIQN) -> None:
+    def _reset_noise_device(self, net: RainbowIQN, rng=None) -> None:
         # batched: all 4 NoisyLinear layers in 2 kernels (K6/K7 fused).
         # Packs are built lazily AFTER FusedForward re-points weight_noisy.
         packs = getattr(self, "_noisy_packs", None)
@@ -236,7 +266,8 @@ class GpuLearner:
             from .fused_forward import noisy_pack
 
             packs[key] = noisy_pack(net, self.device)
-        self.ext.noisy_material4(packs[key], self.rng_state)
+        self.ext.noisy_material4(packs[key],
+                                 rng if rng is not None else self.rng_state)
 
     def _step_body(self):
         if self.manual_bwd:
