@@ -74,17 +74,30 @@ def test_cue_task_learns_offline():
     assert acc >= 0.9, acc
 
 
-def test_train_single_learns_cue():
+def test_train_single_learns_cue(tmp_path):
     """The full single-process trainer (BASELINE config 1 loop: act ->
-    replay -> learn -> target sync) learns the cue task end to end."""
+    replay -> learn -> target sync) learns the cue task end to end, and
+    the EVALUATOR (reference eval protocol: E greedy episodes, best-model
+    checkpoint, reward curve) reports the improvement -- on the cue task
+    eval scores are meaningful, not plumbing noise."""
+    from rainbowiqn_amd.eval import Evaluator
     from rainbowiqn_amd.train_single import train
 
     cfg = _small_cfg(seed=3, learn_start=300, replay_frequency=2,
-                     target_update=200, T_max=10**9)
+                     target_update=200, T_max=10**9, evaluation_episodes=5)
     agent, mem, losses = train(cfg, max_steps=1300)
     assert np.mean(losses[-20:]) < np.mean(losses[:20])
     acc = _greedy_accuracy(agent, cfg)
     assert acc >= 0.9, acc
+    ev = Evaluator(cfg, device="cpu")
+    avg, std = ev.evaluate(agent, T=1300, results_dir=str(tmp_path))
+    # random policy averages ~ -0.5/step; a learned one nets > 0 per
+    # episode with high margin (most steps pay +1)
+    assert avg > 0.0, (avg, std)
+    import os
+
+    assert os.path.exists(tmp_path / "best_model.pth")
+    assert os.path.exists(tmp_path / "rewards.html")
 
 
 def _find_free_port():
