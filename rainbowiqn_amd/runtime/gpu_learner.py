@@ -239,16 +239,23 @@ class GpuLearner:
         self.dfeat.zero_()
         ext.gemm_dx(self.dhh, fo.wcat, feat_g, self.phi, self.dphi,
                     self.dfeat, N)
-        ext.tau_dw(self.dphi, self.cosb, self._iqn_wgrad)
-        ext.bias_grad_cols(self.dphi, self._iqn_bgrad)
+        # the dphi consumers join the side queue behind the z/heads grads
+        sb.wait_stream(main)
+        with torch.cuda.stream(sb):
+            ext.tau_dw(self.dphi, self.cosb, self._iqn_wgrad)
+            ext.bias_grad_cols(self.dphi, self._iqn_bgrad)
         ext.conv_bwd_data(self.dfeat, feat_g, self.c1g, self.c2g, self.w3tp,
                           self.w2tp, self.dpre3g, self.dc2g, self.dc1g,
                           self.db1s, self.db2s, self.db3s)
+        # the three conv dW kernels are mutually independent: split them
+        # across the two queues behind conv_bwd_data
+        sb.wait_stream(main)
+        with torch.cuda.stream(sb):
+            ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 12)
         ext.conv_bwd_dw(3, self.dpre3g, self.c2g, self.dW3s, b, 8)
-        ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 12)
         ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 32)
+        main.wait_stream(sb)  # join before finalize + Adam
         ext.conv_finalize(self._conv_scr, self._conv_grads)
-        main.wait_stream(sb)  # join: Adam reads the whole flat grad
         # ---- optimizer + priorities --------------------------------------
         self.opt.step()
         self.replay.update_priorities(idx, self.per_sample)
