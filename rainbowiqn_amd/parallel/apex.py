@@ -520,6 +520,14 @@ class ApexRunner:
                         meta_list.extend(ml)
         else:
             frames_list, meta_list = local_f, local_m
+        # weight flow BEFORE the learn phase: every rank broadcasts the
+        # learner's current (post-last-round) weights, then actor ranks run
+        # their NEXT round concurrently with the learner's grad steps --
+        # the same one-round bounded staleness as the colocated overlap.
+        # Collective order (gather -> broadcast) is identical on all ranks.
+        if comm.world_size > 1:
+            with self._sp("broadcast"):
+                comm.broadcast_weights(self.flat_weights, src=0)
         if comm.is_learner:
             # ring id = actor_rank * M + local actor (interleaved per j-gather:
             # frames_list order is [j0: ranks 1..W-1][j1: ...] when W > 1)
@@ -540,10 +548,6 @@ class ApexRunner:
                 if self.grad_steps % tgt < cfg.learn_per_round:
                     self.glearner.update_target()
                     self.glearner.maintenance()
-
-        if comm.world_size > 1:
-            with self._sp("broadcast"):
-                comm.broadcast_weights(self.flat_weights, src=0)
 
     def _run_round_overlap(self, learn: bool) -> None:
         """Colocated round: M actor streams + learner stream, concurrent."""
