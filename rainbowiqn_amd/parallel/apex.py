@@ -259,9 +259,14 @@ class ApexRunner:
                 dtt = torch.bfloat16 if cfg.dtype == "bf16" else torch.float32
                 ext_replay = RemoteReplay(cfg, comm, cfg.learn_per_round, dev,
                                           dtype=dtt)
+            # colocated (world 1): the learner shares the chip with the
+            # actor streams under a CU partition -- shallow fork (level 1);
+            # a dedicated learner GPU (world > 1) takes the full fork
             self.glearner = GpuLearner(cfg, n_act, n_rings=comm.n_actors * M,
                                        device=dev, use_graph=cfg.hipgraph,
-                                       replay=ext_replay)
+                                       replay=ext_replay,
+                                       fork_level=1 if comm.world_size == 1
+                                       else 2)
             self.replay = self.glearner.replay
             self.flat_weights = self.glearner.flat_online
         if comm.is_actor:

@@ -35,7 +35,16 @@ def _noisy_layers(net: RainbowIQN):
 class GpuLearner:
     def __init__(self, cfg, action_space: int, n_rings: int = 1,
                  device: str = "cuda", use_graph: bool = True,
-                 replay=None):
+                 replay=None, fork_level: int = 2):
+        # fork_level: how much of the learn step rides the side queue.
+        #   2 = full (theta/double-Q chain, z-grads, dphi consumers, dw2) --
+        #       best when the learner owns the whole GPU (learner-only
+        #       mode 1995 -> 2102 grad/s; multi-GPU learner rank);
+        #   1 = theta chain + z-grad family only -- the colocated CU-
+        #       partitioned mode: the deeper fork's extra cross-queue sync
+        #       nodes in the 8-step-unrolled graph cost far more than the
+        #       overlap there (5.74 -> 3.4M fr/s measured).
+        self.fork_level = fork_level
         self.cfg = cfg
         self.ext = get_extension()
         self.device = torch.device(device)
@@ -239,22 +248,33 @@ class GpuLearner:
         self.dfeat.zero_()
         ext.gemm_dx(self.dhh, fo.wcat, feat_g, self.phi, self.dphi,
                     self.dfeat, N)
-        # the dphi consumers join the side queue behind the z/heads grads
-        sb.wait_stream(main)
-        with torch.cuda.stream(sb):
+        if self.fork_level >= 2:
+            # dphi consumers join the side queue behind the z/heads grads
+            sb.wait_stream(main)
+            with torch.cuda.stream(sb):
+                ext.tau_dw(self.dphi, self.cosb, self._iqn_wgrad)
+                ext.bias_grad_cols(self.dphi, self._iqn_bgrad)
+            ext.conv_bwd_data(self.dfeat, feat_g, self.c1g, self.c2g,
+                              self.w3tp, self.w2tp, self.dpre3g, self.dc2g,
+                              self.dc1g, self.db1s, self.db2s, self.db3s)
+            # the three conv dW kernels are mutually independent: split
+            # them across the two queues behind conv_bwd_data
+            sb.wait_stream(main)
+            with torch.cuda.stream(sb):
+                ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 12)
+            ext.conv_bwd_dw(3, self.dpre3g, self.c2g, self.dW3s, b, 8)
+            ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 32)
+            main.wait_stream(sb)  # join before finalize + Adam
+        else:
             ext.tau_dw(self.dphi, self.cosb, self._iqn_wgrad)
             ext.bias_grad_cols(self.dphi, self._iqn_bgrad)
-        ext.conv_bwd_data(self.dfeat, feat_g, self.c1g, self.c2g, self.w3tp,
-                          self.w2tp, self.dpre3g, self.dc2g, self.dc1g,
-                          self.db1s, self.db2s, self.db3s)
-        # the three conv dW kernels are mutually independent: split them
-        # across the two queues behind conv_bwd_data
-        sb.wait_stream(main)
-        with torch.cuda.stream(sb):
+            ext.conv_bwd_data(self.dfeat, feat_g, self.c1g, self.c2g,
+                              self.w3tp, self.w2tp, self.dpre3g, self.dc2g,
+                              self.dc1g, self.db1s, self.db2s, self.db3s)
+            ext.conv_bwd_dw(3, self.dpre3g, self.c2g, self.dW3s, b, 8)
             ext.conv_bwd_dw(2, self.dc2g, self.c1g, self.dW2s, b, 12)
-        ext.conv_bwd_dw(3, self.dpre3g, self.c2g, self.dW3s, b, 8)
-        ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 32)
-        main.wait_stream(sb)  # join before finalize + Adam
+            ext.conv_bwd_dw(1, self.dc1g, s, self.dW1s, b, 32)
+            main.wait_stream(sb)  # join the z-grad family before Adam
         ext.conv_finalize(self._conv_scr, self._conv_grads)
         # ---- optimizer + priorities --------------------------------------
         self.opt.step()
