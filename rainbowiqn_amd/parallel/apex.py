@@ -367,13 +367,6 @@ class ApexRunner:
                     import atexit
 
                     atexit.register(ext.destroy_stream, lp)
-                    # the learn step's intra-graph side queue must share
-                    # the LEARNER's CU mask -- an unmasked side stream
-                    # tramples the actor partition (measured 5.7 -> 4.0M)
-                    lp2 = ext.create_cu_masked_stream(_mask(0, cus))
-                    atexit.register(ext.destroy_stream, lp2)
-                    self.glearner._bstream = torch.cuda.ExternalStream(
-                        lp2, device=d)
                     # all actor queues share the actor CU partition
                     self._astreams = []
                     for _ in range(M):
