@@ -842,6 +842,107 @@ __global__ __launch_bounds__(512) void heads_gemm_256n_kernel(
     }
 }
 
+// 320x256 tile (NBUF=2, 144 KB LDS): 72 KB staged per 64-k tile for
+// 320x256x64x2 = 10.5 MFLOP = 142 FLOP/B, +9% arithmetic intensity over
+// 256x256 against the fill-rate bound.  Wave tile 80x128 (acc[5][8], 160
+// VGPRs -- the 384-row version spilled at 256).  Experimental variant 3.
+template <int NBUF>
+__global__ __launch_bounds__(512) void heads_gemm_320_kernel(
+    const unsigned short* __restrict__ hin,  // [R,3136]
+    const unsigned short* __restrict__ wcat, // [1024,3136]
+    const unsigned short* __restrict__ bcat, // [1024]
+    unsigned short* __restrict__ out,        // [R,1024]
+    int R) {
+    __shared__ unsigned short smem[NBUF * (320 + 256) * 64];
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;   // 8 waves
+    const int wr = wave >> 1, wc = wave & 1;  // 4x2 grid of 80x128 tiles
+    const int l15 = lane & 15, l4 = lane >> 4;
+    int bxi = blockIdx.x, byi = blockIdx.y;
+    if ((gridDim.x & 7) == 0) {
+        const int id = blockIdx.x + blockIdx.y * gridDim.x;
+        const int xcd = id & 7, s = id >> 3;
+        byi = s % gridDim.y;
+        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
+    }
+    const int bm = bxi * 320, bn = byi * 256;
+
+    // 72 glds insts per buffer (A 40 + B 32): wave w issues 9.
+    auto stage = [&](int buf, int k0) {
+        unsigned short* base = smem + buf * ((320 + 256) * 64);
+        #pragma unroll
+        for (int i = 0; i < 5; ++i) {
+            const int inst = wave * 5 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            const int gr = min(bm + row, R - 1);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (hin + (int64_t)gr * 3136 + k0 + piece * 8);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + inst * 512);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int inst = wave * 4 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (wcat + (int64_t)(bn + row) * 3136 + k0 + piece * 8);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + 320 * 64 + inst * 512);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[5][8];
+    #pragma unroll
+    for (int fm = 0; fm < 5; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 8; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stage(0, 0);
+    for (int kt = 0; kt < 3136 / 64; ++kt) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + NBUF - 1 < 3136 / 64)
+            stage((kt + NBUF - 1) % NBUF, (kt + NBUF - 1) * 64);
+        const unsigned short* As = smem + (kt % NBUF) * ((320 + 256) * 64);
+        const unsigned short* Bs = As + 320 * 64;
+        #pragma unroll
+        for (int ksub = 0; ksub < 2; ++ksub) {
+            #pragma unroll
+            for (int fm = 0; fm < 5; ++fm) {
+                const int row = wr * 80 + fm * 16 + l15;
+                const int piece = (ksub * 4 + l4) ^ (row & 7);
+                bf16x8 a = as_bf16x8(*(const u16x8*)&As[row * 64 + piece * 8]);
+                #pragma unroll
+                for (int fn = 0; fn < 8; ++fn) {
+                    const int nrow = wc * 128 + fn * 16 + l15;
+                    const int bp = (ksub * 4 + l4) ^ (nrow & 7);
+                    bf16x8 b = as_bf16x8(*(const u16x8*)&Bs[nrow * 64 + bp * 8]);
+                    acc[fm][fn] = mfma16(a, b, acc[fm][fn]);
+                }
+            }
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    #pragma unroll
+    for (int fm = 0; fm < 5; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 8; ++fn) {
+            const int n = bn + wc * 128 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * 80 + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
 // Actor-path fusion of tau_embed INTO the heads GEMM (K4-K6 as one kernel):
 // h[r,k] = relu(cos_r . Wiqn[k,:] + b_iqn[k]) * feat[r/nq, k] is produced
 // per 256x64 k-tile straight into the LDS A-image the main MFMA loop reads
@@ -1219,6 +1320,15 @@ void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
             env_variant = e ? atoi(e) : 1;
         }
         variant = env_variant;
+    }
+    if (variant == 3 && R >= 8192) {
+        dim3 gridb((R + 319) / 320, 1024 / 256);
+        hipLaunchKernelGGL(heads_gemm_320_kernel<2>, gridb, dim3(512), 0,
+                           (hipStream_t)stream, (const unsigned short*)h,
+                           (const unsigned short*)wcat,
+                           (const unsigned short*)bcat, (unsigned short*)out,
+                           R);
+        return;
     }
     if ((variant == 2 || (variant == 1 && R >= 16384)) && R >= 8192) {
         // 256x256 wins once the grid fills the chip (R/256 x 4 >= 256 WGs):
