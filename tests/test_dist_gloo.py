@@ -332,3 +332,121 @@ def test_ring_id_assignment_world8():
     learner = res[0]
     for a, vals in enumerate(learner["ring_rewards"]):
         assert vals == {float(a + 1)}, (a, vals)
+
+
+def _worker_fastflow(rank, world, port, results):
+    """Drive the REAL _run_round_fast control flow (the RCCL multi-rank
+    path: gather -> broadcast -> learn) over gloo with mocked GPU pieces --
+    validates collective order, ring-id assignment and the broadcast-
+    before-learn staleness reorder without GPU hardware."""
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+    )
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    cfg = Config(device="cpu", learn_start=64, memory_capacity=4096,
+                 learn_per_round=2, batch_size=8, hidden_size=32,
+                 num_tau_samples=2, num_tau_prime_samples=2,
+                 num_quantile_samples=3)
+    comm = Communicator(backend="gloo", device=torch.device("cpu"))
+    runner = ApexRunner.__new__(ApexRunner)
+    runner.cfg, runner.comm = cfg, comm
+    runner.n_envs, runner.chunk_steps = 4, 4
+    runner.n_local_actors = 1
+    runner.grad_steps, runner.losses = 0, []
+    runner.tracer = None
+    runner.sharded = False
+    runner.overlap = False
+    runner.fast = True
+    runner.gactors = []
+    runner.glearner = None
+    runner.flat_weights = torch.zeros(8)
+    c = runner.n_envs * runner.chunk_steps
+
+    class _FakeLoop:
+        def __init__(self):
+            self.chunk_frames = torch.full((c, 84, 84), float(rank),
+                                           dtype=torch.uint8)
+            self.chunk_meta = torch.full((c, 3), float(rank))
+            self.total_steps = 0
+
+        def run_round(self):
+            self.total_steps += c
+
+    class _FakeReplay(dict):
+        def __len__(self):
+            return 10**6  # past learn_start
+
+    class _FakeLearner:
+        def __init__(self):
+            self.replay = _FakeReplay()
+            self.appended = []
+            self.grad_steps = 0
+            self.flat_online = torch.zeros(8)
+
+        def append_actor_chunk(self, ring, f, meta=None):
+            self.appended.append((ring, float(f.float().mean()),
+                                  float(meta.mean())))
+
+        def learn_steps(self, k):
+            self.grad_steps += k
+            # weights change AFTER the broadcast already happened
+            self.flat_online += 1.0
+
+        def last_loss(self):
+            return 0.5
+
+        def update_target(self):
+            pass
+
+        def maintenance(self):
+            pass
+
+    if comm.is_learner:
+        runner.glearner = _FakeLearner()
+        runner.flat_weights = runner.glearner.flat_online
+        runner.actor = None
+    else:
+        loop = _FakeLoop()
+        runner.gactors = [loop]
+        runner._actor_flats = [runner.flat_weights]
+        from rainbowiqn_amd.parallel.apex import _ActorSteps
+
+        runner.actor = _ActorSteps([loop])
+        runner.gactor = loop
+    out = {"rank": rank, "bcast": []}
+    for rnd in range(3):
+        runner._run_round_fast(learn=True)
+        out["bcast"].append(float(runner.flat_weights[0]))
+    if comm.is_learner:
+        out["appended"] = runner.glearner.appended
+        out["grad_steps"] = runner.glearner.grad_steps
+    results[rank] = out
+    comm.close()
+
+
+def test_fast_multirank_flow_gloo():
+    """The fast-path multi-rank round (the path the 8-GPU RCCL run takes):
+    chunks land in per-actor rings, and actors receive the PRE-learn
+    weights each round (bounded one-round staleness: broadcast precedes
+    the learn phase)."""
+    world = 3
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_fastflow, args=(world, port, results), nprocs=world,
+             join=True)
+    res = dict(results)
+    learner = res[0]
+    # ring id == actor index; payload identifies the sender rank (= a+1)
+    rings = sorted({(ring, int(fmean)) for ring, fmean, _ in
+                    learner["appended"]})
+    assert rings == [(0, 1), (1, 2)], rings
+    assert learner["grad_steps"] == 6
+    # actors saw weights 0, 1, 2 at rounds 0, 1, 2: each round's broadcast
+    # carries the weights BEFORE that round's learn (one-round stale)
+    for r in (1, 2):
+        assert res[r]["bcast"] == [0.0, 1.0, 2.0], res[r]["bcast"]
