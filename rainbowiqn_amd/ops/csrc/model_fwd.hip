@@ -842,6 +842,108 @@ __global__ __launch_bounds__(512) void heads_gemm_256n_kernel(
     }
 }
 
+// 256x256 tile with REGISTER-DIRECT B (variant 4): only A rides the
+// LDS-DMA fill path (32 KB/kt, half the 256x256 kernel's); each wave reads
+// its B fragments straight from L2 into VGPRs (B is L2-resident -- 6.4 MB
+// re-read by every m-band of the XCD).  Splits the per-kt traffic across
+// the two independent load paths AND halves LDS to 64 KB -> 2 WGs/CU for
+// latency hiding.  Same per-element k-order as the other variants
+// (bit-exact).
+template <int NBUF>
+__global__ __launch_bounds__(512) void heads_gemm_rb_kernel(
+    const unsigned short* __restrict__ hin,  // [R,3136]
+    const unsigned short* __restrict__ wcat, // [1024,3136]
+    const unsigned short* __restrict__ bcat, // [1024]
+    unsigned short* __restrict__ out,        // [R,1024]
+    int R) {
+    __shared__ unsigned short smem[NBUF * 256 * 64];
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;   // 8 waves
+    const int wr = wave >> 1, wc = wave & 1;  // 4x2 grid of 64x128 tiles
+    const int l15 = lane & 15, l4 = lane >> 4;
+    int bxi = blockIdx.x, byi = blockIdx.y;
+    if ((gridDim.x & 7) == 0) {
+        const int id = blockIdx.x + blockIdx.y * gridDim.x;
+        const int xcd = id & 7, s = id >> 3;
+        byi = s % gridDim.y;
+        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
+    }
+    const int bm = bxi * 256, bn = byi * 256;
+
+    // A only: 32 glds insts per buffer, wave w issues 4
+    auto stage = [&](int buf, int k0) {
+        unsigned short* base = smem + buf * (256 * 64);
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int inst = wave * 4 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            const int gr = min(bm + row, R - 1);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (hin + (int64_t)gr * 3136 + k0 + piece * 8);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + inst * 512);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[4][8];
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 8; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stage(0, 0);
+    if (NBUF == 3) stage(1, 64);
+    for (int kt = 0; kt < 3136 / 64; ++kt) {
+        const int k0 = kt * 64;
+        if (NBUF == 3 && kt + 1 < 3136 / 64)
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + NBUF - 1 < 3136 / 64)
+            stage((kt + NBUF - 1) % NBUF, (kt + NBUF - 1) * 64);
+        const unsigned short* As = smem + (kt % NBUF) * (256 * 64);
+        #pragma unroll
+        for (int ksub = 0; ksub < 2; ++ksub) {
+            // B fragments straight from global (fm-invariant, loaded once)
+            bf16x8 bfr[8];
+            #pragma unroll
+            for (int fn = 0; fn < 8; ++fn) {
+                const int nrow = wc * 128 + fn * 16 + l15;
+                bfr[fn] = as_bf16x8(*(const u16x8*)(
+                    wcat + (int64_t)(bn + nrow) * 3136 + k0 +
+                    (ksub * 4 + l4) * 8));
+            }
+            #pragma unroll
+            for (int fm = 0; fm < 4; ++fm) {
+                const int row = wr * 64 + fm * 16 + l15;
+                const int piece = (ksub * 4 + l4) ^ (row & 7);
+                bf16x8 a = as_bf16x8(*(const u16x8*)&As[row * 64 + piece * 8]);
+                #pragma unroll
+                for (int fn = 0; fn < 8; ++fn)
+                    acc[fm][fn] = mfma16(a, bfr[fn], acc[fm][fn]);
+            }
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 8; ++fn) {
+            const int n = bn + wc * 128 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * 64 + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
 // 320x256 tile (NBUF=2, 144 KB LDS): 72 KB staged per 64-k tile for
 // 320x256x64x2 = 10.5 MFLOP = 142 FLOP/B, +9% arithmetic intensity over
 // 256x256 against the fill-rate bound.  Wave tile 80x128 (acc[5][8], 160
@@ -1320,6 +1422,15 @@ void riqn_heads_gemm(const void* h, const void* wcat, const void* bcat,
             env_variant = e ? atoi(e) : 1;
         }
         variant = env_variant;
+    }
+    if (variant == 4 && R >= 8192) {
+        dim3 gridb((R + 255) / 256, 1024 / 256);
+        hipLaunchKernelGGL(heads_gemm_rb_kernel<3>, gridb, dim3(512), 0,
+                           (hipStream_t)stream, (const unsigned short*)h,
+                           (const unsigned short*)wcat,
+                           (const unsigned short*)bcat, (unsigned short*)out,
+                           R);
+        return;
     }
     if (variant == 3 && R >= 8192) {
         dim3 gridb((R + 319) / 320, 1024 / 256);

@@ -63,7 +63,7 @@ def main():
         print(f"{'tau_embed direct':<28} {f'R={R}':<18} {us:9.2f}")
         gf = R * 1024 * 3136 * 2 / 1e9
         for var, name in ([(1, "heads_gemm 256x128"), (2, "heads_gemm 256x256"),
-                           (3, "heads_gemm 320x256")]
+                           (3, "heads_gemm 320x256"), (4, "heads_gemm 256rb")]
                           if R >= 8192 else [(1, "heads_gemm")]):
             us = timeit(lambda: ext.heads_gemm(h, fw.wcat, fw.bcat, hh,
                                                variant=var), iters)

@@ -331,3 +331,6 @@ def test_bf16_gemm_256n_variant_parity():
         ext.heads_gemm(h, fw.wcat, fw.bcat, out2, variant=3)
         torch.cuda.synchronize()
         assert torch.equal(out1, out2), (R, "320x256")
+        ext.heads_gemm(h, fw.wcat, fw.bcat, out2, variant=4)
+        torch.cuda.synchronize()
+        assert torch.equal(out1, out2), (R, "256rb")
