@@ -215,3 +215,37 @@ def test_fused_stack_learns_cue():
     assert early < -0.2, (early, late)  # random policy near chance
     assert late > 0.5, (early, late)  # fused stack learned the task
     assert runner.grad_steps > 1000
+
+
+def test_manual_backward_over_remote_replay():
+    """ADVICE r01 item 1 regression: --sharded-replay + the default fused
+    backward crashed because RemoteReplay lacked states_cat.  Drive the
+    GPU learner's manual fused learn step (graph-captured) over a
+    RemoteReplay facade with locally filled batch buffers (no dist): the
+    states_cat contract must feed the one-call [s; s'] conv encode."""
+    from rainbowiqn_amd.parallel.sharded import RemoteReplay
+    from rainbowiqn_amd.runtime.gpu_learner import GpuLearner
+
+    class _Comm:
+        n_actors = 2
+
+    cfg = Config(device="cuda", dtype="bf16", learn_per_round=2,
+                 learn_start=16, batch_size=32)
+    rep = RemoteReplay(cfg, _Comm(), steps=cfg.learn_per_round,
+                       device="cuda", dtype=torch.bfloat16)
+    torch.manual_seed(0)
+    rep.bst.uniform_(0, 1)
+    rep.bs2.uniform_(0, 1)
+    rep.baa.random_(0, 6)
+    rep.brr.uniform_(-1, 1)
+    rep.bnt.fill_(1.0)
+    rep.bww.fill_(1.0)
+    rep.set_len(10**6)
+    lrn = GpuLearner(cfg, 6, device="cuda", use_graph=True, replay=rep)
+    assert lrn.manual_bwd
+    for _ in range(3):
+        lrn.learn_steps(cfg.learn_per_round)
+    torch.cuda.synchronize()
+    assert lrn.grad_steps >= 6
+    assert torch.isfinite(lrn.loss_static).all()
+    assert torch.isfinite(rep.bpri).all() and float(rep.bpri.max()) > 0
