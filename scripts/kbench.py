@@ -55,6 +55,15 @@ def main():
         us = timeit(lambda: ext.conv_encode(x, fw.w1, fw.b1, fw.w2, fw.b2,
                                             fw.w3p, fw.b3, feat), iters)
         print(f"{'conv_encode':<28} {f'B={B}':<18} {us:9.2f}")
+        if B >= 256:  # actor path: conv1 gathers from the u8 frame ring
+            ring = torch.randint(0, 255, (4, B, 7056), dtype=torch.uint8,
+                                 device="cuda")
+            est = torch.zeros(B, 4, dtype=torch.int32, device="cuda")
+            est[:, 1] = 100  # steps past history: all frames valid
+            us = timeit(lambda: ext.conv_encode_gather(
+                ring, est.view(-1), 4, 3, fw.w1, fw.b1, fw.w2, fw.b2,
+                fw.w3p, fw.b3, feat), iters)
+            print(f"{'conv_encode_gather':<28} {f'E={B}':<18} {us:9.2f}")
         us = timeit(lambda: ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h,
                                           nq, direct=0), iters)
         print(f"{'tau_embed staged':<28} {f'R={R}':<18} {us:9.2f}")
