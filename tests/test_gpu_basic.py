@@ -434,3 +434,25 @@ def test_replay_at_reference_scale():
     assert torch.isfinite(w).all() and float(w.max()) <= 1.0 + 1e-5
     assert torch.isfinite(r).all()
     rep.check_invariants()
+
+
+@pytest.mark.gpu
+def test_single_env_actor_reference_shape():
+    """n_envs=1: the reference's one-env-per-actor shape
+    must run through the same fused kernels (grid-of-1 edge cases)."""
+    from rainbowiqn_amd.config import Config
+    from rainbowiqn_amd.parallel.apex import ApexRunner
+    from rainbowiqn_amd.parallel.comm import Communicator
+
+    cfg = Config(device="cuda", dtype="bf16", learn_start=64,
+                 memory_capacity=4096, learn_per_round=1)
+    runner = ApexRunner(cfg, Communicator(backend="auto"), n_envs=1,
+                        chunk_steps=8)
+    assert runner.fast
+    for _ in range(16):
+        runner.run_round()
+    torch.cuda.synchronize()
+    assert runner.grad_steps > 0
+    assert all(l == l for l in runner.losses)  # finite (not NaN)
+    acts = runner.gactor.actions
+    assert 0 <= int(acts[0]) < 6
