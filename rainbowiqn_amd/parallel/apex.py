@@ -360,13 +360,27 @@ class ApexRunner:
                             w[cu // 32] |= 1 << (cu % 32)
                         return w
 
-                    lp = ext.create_cu_masked_stream(_mask(0, cus))
-                    self._lstream = torch.cuda.ExternalStream(lp, device=d)
+                    # RIQN_LFREE=1: leave the learner on free high-
+                    # priority queues (no mask) while the actor keeps its
+                    # CU share -- the learn step's small grids fit in far
+                    # fewer CUs than the partition grants, so the mask may
+                    # cost more latency than the isolation buys (A/B knob).
+                    lfree = _os.environ.get("RIQN_LFREE", "0") == "1"
+                    if lfree:
+                        self._lstream = torch.cuda.Stream(device=d,
+                                                          priority=-1)
+                        self.glearner._bstream = torch.cuda.Stream(
+                            device=d, priority=-1)
+                    else:
+                        lp = ext.create_cu_masked_stream(_mask(0, cus))
+                        self._lstream = torch.cuda.ExternalStream(lp,
+                                                                  device=d)
                     # destroy at exit: a leaked HSA queue crashes profiler
                     # teardown (__cxa_finalize segfault under rocprofv3)
                     import atexit
 
-                    atexit.register(ext.destroy_stream, lp)
+                    if not lfree:
+                        atexit.register(ext.destroy_stream, lp)
                     # all actor queues share the actor CU partition
                     self._astreams = []
                     for _ in range(M):
