@@ -42,3 +42,22 @@ def test_bench_learner_only_contract():
     assert d["unit"] == "grad-steps/s" and d["value"] > 0
     assert d["env_steps_per_sec"] == 0.0
     assert "learner-only" in d["config"]["parallelism"]
+
+
+def test_bench_torchrun_world2_contract():
+    """The driver's exact multi-rank launch shape (torchrun --nnodes=1):
+    rank 0 emits ONE JSON line aggregating both ranks' work."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29653", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--n-envs", "4",
+         "--chunk-steps", "4", "--learn-per-round", "1", "--device", "cpu"],
+        capture_output=True, text=True, timeout=900, check=True,
+    )
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert "1 learner + 1 actor" in d["config"]["parallelism"]
+    assert d["value"] > 0 and d["learner_grad_steps_per_sec"] > 0
