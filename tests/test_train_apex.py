@@ -102,3 +102,35 @@ def test_metrics_exporter(tmp_path):
         "http://127.0.0.1:18123/metrics", timeout=5).read().decode()
     assert "riqn_grad_steps_total" in body
     assert "riqn_env_steps_per_sec" in body
+
+
+def test_multiseed_launcher_smoke(tmp_path):
+    """scripts/launch_multiseed.py (reference multi-seed run scripts,
+    SURVEY 3.1 plotting/multi-seed row): two sequential tiny seeds run and
+    produce per-seed results dirs + a collation line."""
+    import json
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "scripts/launch_multiseed.py", "--seeds", "2",
+         "--game", "cue", "--id", "mstest-tmp", "--",
+         "--max-rounds", "2", "--n-envs", "4", "--chunk-steps", "4",
+         "--hidden-size", "32", "--memory-capacity", "2048",
+         "--learn-start", "64", "--learn-per-round", "1",
+         "--num-tau-samples", "2", "--num-tau-prime-samples", "2",
+         "--num-quantile-samples", "3",
+         "--evaluation-interval", "1000000000",
+         "--device", "cpu", "--dtype", "fp32"],
+        capture_output=True, text=True, timeout=600, check=True,
+    )
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert lines, out.stdout
+    runs = json.loads(lines[-1])["runs"]
+    assert len(runs) == 2
+    import os
+    import shutil
+
+    for r in runs:
+        assert os.path.isdir(r)
+        shutil.rmtree(r, ignore_errors=True)
