@@ -107,6 +107,7 @@ __global__ void quant_rows_fp8_kernel(
 // cols = 14 aligned 32-col blocks), waves own n-tile PAIRS so each lane
 // group sees its block's full 32 columns.  Cos basis via the Chebyshev
 // recurrence exactly as tau_embed_kernel (model_fwd.hip).
+template <int V2>
 __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
     const float* __restrict__ taus,         // [R]
     const unsigned short* __restrict__ w,   // [3136,64] bf16 frag-major pack
@@ -116,6 +117,12 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
     unsigned char* __restrict__ hsc,        // [R,100] e8m0 out
     int R, int nq) {
     __shared__ unsigned short ftile[4 * 800];
+    // V2 epilogue: wave-private 32x32 bf16 staging tile per pair -- the
+    // computed h values transpose through LDS so each LANE quantizes one
+    // row-half (16 values) with a single block_exp and ONE dwordx4 store,
+    // replacing the per-value swizzle-amax chains and byte stores of V1
+    // (per-pair instruction count ~120 -> ~80; stores 32x1B -> 4x16B).
+    __shared__ unsigned short qtile[V2 ? 4 * 32 * 32 : 1];
     const int t = threadIdx.x;
     const int lane = t & 63;
     const int wave = t >> 6;
@@ -198,6 +205,69 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
         const float b_b = bf(bias[(nt_a + 1) * 16 + l15]);
         const int col_a = (nt_a - nt0) * 16 + l15;
         const int blk = pair0 + pp;
+        if (V2) {
+            unsigned short* qt = qtile + wave * (32 * 32);
+            #pragma unroll
+            for (int mi = 0; mi < 2; ++mi) {
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int rr = mi * 16 + l4 * 4 + j;  // 0..31
+                    const int r = (mt0 + mi) * 16 + l4 * 4 + j;
+                    const int fo = min(r / nq - s0, span - 1) * 800;
+                    unsigned short pa =
+                        f32_to_bf16(fmaxf(acc[0][mi][j] + b_a, 0.0f));
+                    pa = f32_to_bf16(bf(pa) * bf(ftile[fo + col_a]));
+                    unsigned short pb =
+                        f32_to_bf16(fmaxf(acc[1][mi][j] + b_b, 0.0f));
+                    pb = f32_to_bf16(bf(pb) * bf(ftile[fo + col_a + 16]));
+                    qt[rr * 32 + l15] = pa;
+                    qt[rr * 32 + 16 + l15] = pb;
+                }
+            }
+            // quant phase: lane = (row 0..31, half 0..1); one 16-byte
+            // store per lane covers the pair's 32x32 block
+            {
+                const int qrow = lane >> 1, half = lane & 1;
+                const unsigned short* qr = qt + qrow * 32 + half * 16;
+                float v[16];
+                float am_h = 0.0f;
+                #pragma unroll
+                for (int i = 0; i < 16; ++i) {
+                    v[i] = bf(qr[i]);
+                    am_h = fmaxf(am_h, fabsf(v[i]));
+                }
+                // combine the two halves' amax (xor-1 swizzle pairs lanes)
+                float o = __int_as_float(__builtin_amdgcn_ds_swizzle(
+                    __float_as_int(am_h), (1 << 10) | 31));
+                const int e = block_exp(fmaxf(am_h, o));
+                const float inv = exp2f((float)-e);
+                int w0 = 0, w1 = 0, w2 = 0, w3 = 0;
+                w0 = __builtin_amdgcn_cvt_pk_fp8_f32(v[0] * inv, v[1] * inv,
+                                                     w0, false);
+                w0 = __builtin_amdgcn_cvt_pk_fp8_f32(v[2] * inv, v[3] * inv,
+                                                     w0, true);
+                w1 = __builtin_amdgcn_cvt_pk_fp8_f32(v[4] * inv, v[5] * inv,
+                                                     w1, false);
+                w1 = __builtin_amdgcn_cvt_pk_fp8_f32(v[6] * inv, v[7] * inv,
+                                                     w1, true);
+                w2 = __builtin_amdgcn_cvt_pk_fp8_f32(v[8] * inv, v[9] * inv,
+                                                     w2, false);
+                w2 = __builtin_amdgcn_cvt_pk_fp8_f32(v[10] * inv, v[11] * inv,
+                                                     w2, true);
+                w3 = __builtin_amdgcn_cvt_pk_fp8_f32(v[12] * inv, v[13] * inv,
+                                                     w3, false);
+                w3 = __builtin_amdgcn_cvt_pk_fp8_f32(v[14] * inv, v[15] * inv,
+                                                     w3, true);
+                const int r = mt0 * 16 + qrow;
+                if (r < R) {
+                    i32x4 pk = {w0, w1, w2, w3};
+                    *(i32x4*)(hq + (int64_t)r * 3200 + nt_a * 16 +
+                              half * 16) = pk;
+                    if (half == 0)
+                        hsc[(int64_t)r * 100 + blk] = (unsigned char)(e + 127);
+                }
+            }
+        } else {
         #pragma unroll
         for (int mi = 0; mi < 2; ++mi) {
             #pragma unroll
@@ -227,6 +297,7 @@ __global__ __launch_bounds__(256) void tau_embed_fp8_kernel(
                         hsc[(int64_t)r * 100 + blk] = (unsigned char)(e + 127);
                 }
             }
+        }
         }
     }
 }
@@ -509,12 +580,27 @@ void riqn_tau_embed_fp8(const float* taus, const void* wiqnf,
                         const void* bias, const void* feat, void* hq,
                         void* hsc, int R, int nq, void* stream) {
     dim3 grid((R + 31) / 32, 4);  // 2 m-tiles of 16 x uneven pair-strips
-    hipLaunchKernelGGL(tau_embed_fp8_kernel, grid, dim3(256), 0,
-                       (hipStream_t)stream, taus,
-                       (const unsigned short*)wiqnf,
-                       (const unsigned short*)bias,
-                       (const unsigned short*)feat, (unsigned char*)hq,
-                       (unsigned char*)hsc, R, nq);
+    // RIQN_TAUQ8_V2: 1 (default) = LDS-transpose quant epilogue,
+    // 0 = per-value swizzle-amax epilogue (A/B reference)
+    static int v2 = -1;
+    if (v2 < 0) {
+        const char* e = getenv("RIQN_TAUQ8_V2");
+        v2 = e ? atoi(e) : 1;
+    }
+    if (v2)
+        hipLaunchKernelGGL(tau_embed_fp8_kernel<1>, grid, dim3(256), 0,
+                           (hipStream_t)stream, taus,
+                           (const unsigned short*)wiqnf,
+                           (const unsigned short*)bias,
+                           (const unsigned short*)feat, (unsigned char*)hq,
+                           (unsigned char*)hsc, R, nq);
+    else
+        hipLaunchKernelGGL(tau_embed_fp8_kernel<0>, grid, dim3(256), 0,
+                           (hipStream_t)stream, taus,
+                           (const unsigned short*)wiqnf,
+                           (const unsigned short*)bias,
+                           (const unsigned short*)feat, (unsigned char*)hq,
+                           (unsigned char*)hsc, R, nq);
 }
 
 void riqn_heads_gemm_fp8(const void* hq, const void* hsc, const void* wq,
