@@ -563,6 +563,137 @@ __global__ __launch_bounds__(512) void heads_gemm_fp8_big_kernel(
     }
 }
 
+// 256x256 MX-fp8 tile (NBUF=2, 132 KB LDS): 66 KB staged per 128-k tile
+// for 256x256x128x2 = 16.8 MFLOP = 254 FLOP/B -- 2.25x the 256x128
+// kernel's intensity against the same ~10 TB/s LDS-DMA fill ceiling.
+// Wave tile 64x128 (acc[4][8]); scales ride the stage as in the 256x128
+// kernel.  Same per-element k-order: bit-exact vs the other variants.
+template <int NBUF, int FM>  // row block = FM*64 (FM=4: 256, FM=2: 128)
+__global__ __launch_bounds__(512) void heads_gemm_fp8_256n_kernel(
+    const unsigned char* __restrict__ hq,   // [R,3200] fp8
+    const unsigned char* __restrict__ hsc,  // [R,100] e8m0
+    const unsigned char* __restrict__ wq,   // [1024,3200] fp8
+    const unsigned char* __restrict__ wsc,  // [1024,100] e8m0
+    const unsigned short* __restrict__ bcat,// [1024] bf16
+    unsigned short* __restrict__ out,       // [R,1024] bf16
+    int R) {
+    __shared__ unsigned char smem[NBUF * ((FM * 64 + 256) * 128
+                                          + (FM * 64 + 256) * 4)];
+    const int t = threadIdx.x;
+    const int lane = t & 63, wave = t >> 6;   // 8 waves
+    const int wr = wave >> 1, wc = wave & 1;  // 4x2 grid of 64x128 tiles
+    const int l15 = lane & 15, l4 = lane >> 4;
+    int bxi = blockIdx.x, byi = blockIdx.y;
+    if ((gridDim.x & 7) == 0) {
+        const int id = blockIdx.x + blockIdx.y * gridDim.x;
+        const int xcd = id & 7, s = id >> 3;
+        byi = s % gridDim.y;
+        bxi = xcd * (gridDim.x >> 3) + s / gridDim.y;
+    }
+    const int AROWS = FM * 64;
+    const int bm = bxi * AROWS, bn = byi * 256;
+
+    // glds per buffer: A AROWS/8 + B 32 + scale (AROWS+256)/64
+    auto stage = [&](int buf, int k0) {
+        unsigned char* base =
+            smem + buf * ((AROWS + 256) * 128 + (AROWS + 256) * 4);
+        {
+            // scale dwords: rows 0..AROWS-1 = A (hsc), rest = B (wsc)
+            const int chunk = (AROWS + 256) / 8;
+            const int srow = wave * chunk + lane;
+            if (lane < chunk) {
+                const unsigned char* sp =
+                    srow < AROWS
+                        ? hsc + (int64_t)min(bm + srow, R - 1) * 100 + k0 / 32
+                        : wsc + (int64_t)(bn + srow - AROWS) * 100 + k0 / 32;
+                auto* gp =
+                    (const __attribute__((address_space(1))) uint32_t*)sp;
+                auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                    (base + (AROWS + 256) * 128 + srow * 4);
+                __builtin_amdgcn_global_load_lds(gp, lp, 4, 0, 0);
+            }
+        }
+        #pragma unroll
+        for (int i = 0; i < FM; ++i) {
+            const int inst = wave * FM + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            const int gr = min(bm + row, R - 1);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (hq + (int64_t)gr * 3200 + k0 + piece * 16);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + inst * 1024);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int inst = wave * 4 + i;
+            const int row = inst * 8 + (lane >> 3);
+            const int piece = (lane & 7) ^ (row & 7);
+            auto* gp = (const __attribute__((address_space(1))) uint32_t*)
+                (wq + (int64_t)(bn + row) * 3200 + k0 + piece * 16);
+            auto* lp = (__attribute__((address_space(3))) uint32_t*)
+                (base + AROWS * 128 + inst * 1024);
+            __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+    };
+
+    f32x4 acc[FM][8];
+    #pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+        #pragma unroll
+        for (int fn = 0; fn < 8; ++fn) acc[fm][fn] = f32x4{0, 0, 0, 0};
+    stage(0, 0);
+    for (int kt = 0; kt < 25; ++kt) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        if (kt + NBUF - 1 < 25) stage((kt + NBUF - 1) % NBUF,
+                                      (kt + NBUF - 1) * 128);
+        const unsigned char* As =
+            smem + (kt % NBUF) * ((AROWS + 256) * 128 + (AROWS + 256) * 4);
+        const unsigned char* Bs = As + AROWS * 128;
+        const unsigned char* Ss = As + (AROWS + 256) * 128;
+        #pragma unroll
+        for (int fm = 0; fm < FM; ++fm) {
+            const int row = wr * (FM * 16) + fm * 16 + l15;
+            const int pa0 = l4 ^ (row & 7), pa1 = (4 + l4) ^ (row & 7);
+            i32x4 alo = *(const i32x4*)&As[row * 128 + pa0 * 16];
+            i32x4 ahi = *(const i32x4*)&As[row * 128 + pa1 * 16];
+            i32x8 a = {alo[0], alo[1], alo[2], alo[3],
+                       ahi[0], ahi[1], ahi[2], ahi[3]};
+            const int sa = Ss[row * 4 + l4];
+            #pragma unroll
+            for (int fn = 0; fn < 8; ++fn) {
+                const int nrow = wc * 128 + fn * 16 + l15;
+                const int pb0 = l4 ^ (nrow & 7), pb1 = (4 + l4) ^ (nrow & 7);
+                i32x4 blo = *(const i32x4*)&Bs[nrow * 128 + pb0 * 16];
+                i32x4 bhi = *(const i32x4*)&Bs[nrow * 128 + pb1 * 16];
+                i32x8 b = {blo[0], blo[1], blo[2], blo[3],
+                           bhi[0], bhi[1], bhi[2], bhi[3]};
+                const int sb = Ss[(AROWS + nrow) * 4 + l4];
+                acc[fm][fn] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+                    a, b, acc[fm][fn], 0, 0, 0, sa, 0, sb);
+            }
+        }
+    }
+    __builtin_amdgcn_s_barrier();
+    #pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+        #pragma unroll
+        for (int fn = 0; fn < 8; ++fn) {
+            const int n = bn + wc * 128 + fn * 16 + l15;
+            const float bias = bf(bcat[n]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int r = bm + wr * (FM * 16) + fm * 16 + l4 * 4 + j;
+                if (r < R)
+                    out[(int64_t)r * 1024 + n] =
+                        f32_to_bf16(fmaxf(acc[fm][fn][j] + bias, 0.0f));
+            }
+        }
+    }
+}
+
 }  // namespace
 
 extern "C" {
@@ -615,6 +746,30 @@ void riqn_heads_gemm_fp8(const void* hq, const void* hsc, const void* wq,
             env_variant = e ? atoi(e) : 1;
         }
         variant = env_variant;
+    }
+    if (variant == 2 && R >= 1024) {  // 256x256 (A/B: spills 40 B/lane)
+        dim3 grid((R + 255) / 256, 1024 / 256);
+        hipLaunchKernelGGL((heads_gemm_fp8_256n_kernel<2, 4>), grid,
+                           dim3(512), 0, (hipStream_t)stream,
+                           (const unsigned char*)hq,
+                           (const unsigned char*)hsc,
+                           (const unsigned char*)wq,
+                           (const unsigned char*)wsc,
+                           (const unsigned short*)bcat,
+                           (unsigned short*)out, R);
+        return;
+    }
+    if (variant == 3 && R >= 1024) {  // 128x256 (spill-free, 168 FLOP/B)
+        dim3 grid((R + 127) / 128, 1024 / 256);
+        hipLaunchKernelGGL((heads_gemm_fp8_256n_kernel<2, 2>), grid,
+                           dim3(512), 0, (hipStream_t)stream,
+                           (const unsigned char*)hq,
+                           (const unsigned char*)hsc,
+                           (const unsigned char*)wq,
+                           (const unsigned char*)wsc,
+                           (const unsigned short*)bcat,
+                           (unsigned short*)out, R);
+        return;
     }
     if (variant == 1 && R >= 1024) {
         dim3 grid((R + 255) / 256, 1024 / 128);

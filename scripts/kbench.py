@@ -104,7 +104,8 @@ def main():
                                               feat8, hq, hsc, nq), iters)
         print(f"{'tau_embed_fp8 (fused q)':<28} {f'R={R}':<18} {us:9.2f}")
         gf = R * 1024 * 3200 * 2 / 1e9
-        for var, name in [(0, "fp8 gemm 128x128"), (1, "fp8 gemm 256x128p")]:
+        for var, name in [(0, "fp8 gemm 128x128"), (1, "fp8 gemm 256x128p"),
+                          (2, "fp8 gemm 256x256"), (3, "fp8 gemm 128x256")]:
             us = timeit(lambda: ext.heads_gemm_fp8(hq, hsc, fw8.wq, fw8.wsc,
                                                    fw8.bcat, hh8,
                                                    variant=var), iters)

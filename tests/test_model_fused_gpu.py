@@ -309,6 +309,11 @@ def test_fp8_gemm_big_variant_parity():
         ext.heads_gemm_fp8(hq, sc, fw.wq, fw.wsc, fw.bcat, out1, variant=1)
         torch.cuda.synchronize()
         assert torch.equal(out0, out1), R
+        for v in (2, 3):
+            ext.heads_gemm_fp8(hq, sc, fw.wq, fw.wsc, fw.bcat, out1,
+                               variant=v)
+            torch.cuda.synchronize()
+            assert torch.equal(out0, out1), (R, v)
 
 
 def test_bf16_gemm_256n_variant_parity():
