@@ -118,14 +118,23 @@ __global__ void env_step_kernel(
     // synthesize the new frame
     int t = t_s, cy = cy_s, cx = cx_s;
     uint8_t* dst = ring + ((int64_t)ring_head_new * n_envs + e) * FRAME_PX;
-    for (int i = threadIdx.x; i < FRAME_PX; i += blockDim.x) {
-        int x = i % FRAME_W, y = i / FRAME_W;
-        // cheap per-pixel noise hash (5 bits, matches CPU env's 0..31 range)
-        uint32_t h = (nbase_s ^ (uint32_t)(i * 2654435761u));
-        h ^= h >> 15;
-        uint8_t noise = (uint8_t)(h & 31);
-        dst[i] = task == 1 ? cue_pixel(x, y, t, cue_s, noise)
-                           : synth_pixel(x, y, t, cy, cx, noise);
+    // four pixels per dword store: the per-byte store loop was store-issue
+    // bound (FRAME_PX byte insts per WG); FRAME_PX = 7056 = 4 * 1764
+    for (int i4 = threadIdx.x; i4 < FRAME_PX / 4; i4 += blockDim.x) {
+        uint32_t word = 0;
+        #pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+            const int i = i4 * 4 + jj;
+            int x = i % FRAME_W, y = i / FRAME_W;
+            // cheap per-pixel noise hash (5 bits, matches CPU env's 0..31)
+            uint32_t h = (nbase_s ^ (uint32_t)(i * 2654435761u));
+            h ^= h >> 15;
+            uint8_t noise = (uint8_t)(h & 31);
+            uint8_t px = task == 1 ? cue_pixel(x, y, t, cue_s, noise)
+                                   : synth_pixel(x, y, t, cy, cx, noise);
+            word |= (uint32_t)px << (8 * jj);
+        }
+        ((uint32_t*)dst)[i4] = word;
     }
     // NOTE: offset bump rides a trailing 1-thread kernel (launcher):
     // bumping from block 0 here would race late blocks' offset reads when
