@@ -44,8 +44,38 @@ __global__ void adam_mixed_kernel(
     float bc1 = 1.0f - __powf(beta1, (float)t);
     float bc2 = 1.0f - __powf(beta2, (float)t);
     float step_size = lr / bc1;
-    for (int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x; i < numel;
-         i += (int64_t)gridDim.x * blockDim.x) {
+    // four params per iteration (float4 state, u16x4 grad/param): the
+    // scalar loop was memory-ISSUE bound (8 insts per element; VALU is
+    // trivial next to it).  Flat buffers are 16B-aligned allocations.
+    typedef __attribute__((ext_vector_type(4))) float f4;
+    typedef __attribute__((ext_vector_type(4))) unsigned short u16x4a;
+    const int64_t n4 = numel / 4;
+    for (int64_t i4 = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+         i4 < n4; i4 += (int64_t)gridDim.x * blockDim.x) {
+        u16x4a gv = ((const u16x4a*)grad_bf16)[i4];
+        f4 mv = ((const f4*)m)[i4];
+        f4 vv = ((const f4*)v)[i4];
+        f4 pv = ((const f4*)master)[i4];
+        u16x4a ov;
+        #pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+            float g = bf16_to_f32(gv[jj]);
+            float mi = beta1 * mv[jj] + (1.0f - beta1) * g;
+            float vi = beta2 * vv[jj] + (1.0f - beta2) * g * g;
+            mv[jj] = mi;
+            vv[jj] = vi;
+            float p = pv[jj] - step_size * mi / (sqrtf(vi / bc2) + eps);
+            pv[jj] = p;
+            ov[jj] = f32_to_bf16(p);
+        }
+        ((f4*)m)[i4] = mv;
+        ((f4*)v)[i4] = vv;
+        ((f4*)master)[i4] = pv;
+        ((u16x4a*)param_bf16)[i4] = ov;
+    }
+    // tail
+    for (int64_t i = n4 * 4 + threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
+         i < numel; i += (int64_t)gridDim.x * blockDim.x) {
         float g = bf16_to_f32(grad_bf16[i]);
         float mi = beta1 * m[i] + (1.0f - beta1) * g;
         float vi = beta2 * v[i] + (1.0f - beta2) * g * g;

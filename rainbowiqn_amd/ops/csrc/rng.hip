@@ -163,10 +163,38 @@ __global__ void noisy_material4_kernel(NoisyPack4 p, uint64_t* rng_state) {
     if (threadIdx.x == 0 && blockIdx.x == 0) rng_state[1] += 1;
     int64_t i = threadIdx.x + (int64_t)blockIdx.x * blockDim.x;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    // weight segments
+    // weight segments: TWO adjacent elements per iteration on the bf16
+    // path (n_in is even for every layer, so a pair never straddles a
+    // row: one div, one u32 load/store pair instead of two of each)
     for (int L = 0; L < 4; ++L) {
         const NoisyDesc& d = p.l[L];
         const int64_t tot = d.n_in * d.n_out;
+        const bool aligned4 =
+            ((((uintptr_t)d.w_mu) | ((uintptr_t)d.w_sigma) |
+              ((uintptr_t)d.w_eps) | ((uintptr_t)d.w_noisy)) & 3) == 0;
+        if (BF16 && (d.n_in & 1) == 0 && aligned4) {
+            const int64_t tot2 = tot / 2;
+            for (int64_t k2 = i; k2 < tot2; k2 += stride) {
+                const int64_t k = k2 * 2;
+                const float eo = d.eps_out[k / d.n_in];
+                const int64_t col = k % d.n_in;
+                const float e0 = eo * d.eps_in[col];
+                const float e1 = eo * d.eps_in[col + 1];
+                const uint32_t muw = ((const uint32_t*)d.w_mu)[k2];
+                const uint32_t sgw = ((const uint32_t*)d.w_sigma)[k2];
+                const float mu0 = bf16_to_f32((unsigned short)muw);
+                const float mu1 = bf16_to_f32((unsigned short)(muw >> 16));
+                const float sg0 = bf16_to_f32((unsigned short)sgw);
+                const float sg1 = bf16_to_f32((unsigned short)(sgw >> 16));
+                ((uint32_t*)d.w_eps)[k2] =
+                    (uint32_t)f32_to_bf16(e0) |
+                    ((uint32_t)f32_to_bf16(e1) << 16);
+                ((uint32_t*)d.w_noisy)[k2] =
+                    (uint32_t)f32_to_bf16(mu0 + sg0 * e0) |
+                    ((uint32_t)f32_to_bf16(mu1 + sg1 * e1) << 16);
+            }
+            continue;
+        }
         for (int64_t k = i; k < tot; k += stride) {
             float e = d.eps_out[k / d.n_in] * d.eps_in[k % d.n_in];
             if (BF16) {
