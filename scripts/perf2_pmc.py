@@ -51,5 +51,26 @@ for _ in range(iters):
     ext.tau_embed(taus, fw.wiqn, fw.biqn, feat, h, nq, direct=1)
 for _ in range(iters):
     ext.tau_embed_fp8(taus, fw8.wiqn, fw8.biqn, f8, hq, hsc, nq)
+# actor-path remainder: conv (both variants), finish, env step
+x = torch.rand(B, 4, 84, 84, device="cuda").to(torch.bfloat16)
+ring = torch.randint(0, 255, (4, B, 7056), dtype=torch.uint8, device="cuda")
+est = torch.zeros(B, 4, dtype=torch.int32, device="cuda")
+est[:, 1] = 100
+rng = torch.tensor([7, 0], dtype=torch.int64, device="cuda")
+acts = torch.zeros(B, dtype=torch.int64, device="cuda")
+cf = torch.zeros(B * 8, 84, 84, dtype=torch.uint8, device="cuda")
+cm = torch.zeros(B * 8, 3, device="cuda")
+for _ in range(iters):
+    ext.conv_encode(x, fw.w1, fw.b1, fw.w2, fw.b2, fw.w3p, fw.b3,
+                    fw._bufs(B, nq)[0])
+for _ in range(iters):
+    ext.conv_encode_gather(ring, est.view(-1), 4, 3, fw.w1, fw.b1, fw.w2,
+                           fw.b2, fw.w3p, fw.b3, fw._bufs(B, nq)[0])
+for _ in range(iters):
+    ext.heads_finish(hh, fw.wz, fw.bz, 6, nq, 1,
+                     torch.zeros(1, device="cuda"), acts)
+for _ in range(iters):
+    ext.env_step(acts, est.view(-1), rng, ring.view(4, -1), 4, 0, 1.0,
+                 4.0 / 512.0, 108000, cf, cm, 0, 0)
 torch.cuda.synchronize()
 print("done", iters)
