@@ -23,8 +23,9 @@ from rainbowiqn_amd.parallel.comm import Communicator
 def main():
     rounds = int(sys.argv[1]) if len(sys.argv) > 1 else 2000
     out = sys.argv[2] if len(sys.argv) > 2 else "gpurun_out/cue_curve.csv"
+    lr = float(sys.argv[3]) if len(sys.argv) > 3 else 1e-3
     cfg = Config(game="cue", device="cuda", dtype="bf16", seed=11,
-                 lr=1e-3, multi_step=1, learn_start=4096,
+                 lr=lr, multi_step=1, learn_start=4096,
                  memory_capacity=262144, learn_per_round=4,
                  target_update=8192)
     runner = ApexRunner(cfg, Communicator(backend="auto"), n_envs=256,
