@@ -249,3 +249,39 @@ def test_manual_backward_over_remote_replay():
     assert lrn.grad_steps >= 6
     assert torch.isfinite(lrn.loss_static).all()
     assert torch.isfinite(rep.bpri).all() and float(rep.bpri.max()) > 0
+
+
+def test_actor_shard_host_fast_path():
+    """Sharded-replay ACTOR side on hardware (config 5): the GpuReplay-
+    backed shard samples its batch quota into the static send buffers
+    (frame-stack + n-step assembly local to the shard) and applies
+    returned priorities -- everything except the P2P wire, which the gloo
+    suite covers."""
+    from rainbowiqn_amd.parallel.sharded import ActorShardHost
+
+    class _Comm:
+        n_actors = 2
+        rank = 1
+
+    cfg = Config(device="cuda", dtype="bf16", learn_per_round=2,
+                 batch_size=32, memory_capacity=65536)
+    host = ActorShardHost(cfg, _Comm(), quota=16, steps=2, device="cuda",
+                          fast=True)
+    g = torch.Generator().manual_seed(0)
+    f = torch.randint(0, 256, (8192, 84, 84), dtype=torch.uint8,
+                      generator=g).cuda()
+    m = torch.rand(8192, 3, generator=g).cuda()
+    host.append_chunk(f, meta=m)
+    host.fill_buffers()
+    torch.cuda.synchronize()
+    assert torch.isfinite(host.ww).all() and float(host.ww.max()) <= 1.01
+    assert torch.isfinite(host.rr).all()
+    assert host.st.abs().sum() > 0  # frame stacks actually assembled
+    # priority return path: update the owning tree from a [S, q] block
+    host.pri.uniform_(0.1, 1.0)
+    for s in range(2):
+        host.replay.update_priorities(host._idxs[s], host.pri[s])
+    torch.cuda.synchronize()
+    host.replay.check_invariants()
+    host.anneal_beta(0.7)
+    assert abs(host.replay.priority_weight - 0.7) < 1e-6
